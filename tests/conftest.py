@@ -1,0 +1,34 @@
+import multiprocessing
+import os
+import sys
+
+import pytest
+
+# Make the in-tree package importable when tests run from the repo root.
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires an AMD GPU (MI355X); skipped on CPU"
+    )
+
+
+def pytest_collection_modifyitems(config, items):
+    try:
+        import torch
+
+        has_gpu = torch.cuda.is_available()
+    except Exception:
+        has_gpu = False
+    if has_gpu:
+        return
+    skip_gpu = pytest.mark.skip(reason="no GPU available")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip_gpu)
+
+
+@pytest.fixture(scope="session")
+def mp_spawn_context():
+    return multiprocessing.get_context("spawn")

@@ -1,0 +1,372 @@
+"""BatchQueue unit tests.
+
+Mirrors the reference's 12-case suite
+(reference: ray_shuffling_data_loader/tests/test_batch_queue.py) on the
+MI355X-native C++ core: FIFO, blocking/timeout/nowait get & put (sync and
+async), concurrent get/put, batched ops + maxsize overflow, qsize tracking,
+shutdown, epoch-window backpressure and an end-to-end streaming pull.
+Ray-actor-specific cases (custom actor resources) have no analog here; the
+added epoch-window tests cover the semantics the reference only exercises
+implicitly through dataset smoke runs.
+"""
+
+import asyncio
+import threading
+import time
+
+import pytest
+
+from ray_shuffling_data_loader_amd.batch_queue import BatchQueue, Empty, Full
+
+
+def make_queue(**kwargs):
+    defaults = dict(num_epochs=1, num_trainers=1, max_concurrent_epochs=1)
+    defaults.update(kwargs)
+    return BatchQueue(**defaults)
+
+
+def test_simple_usage():
+    q = make_queue()
+    items = list(range(10))
+    for item in items:
+        q.put(rank=0, epoch=0, item=item)
+    for item in items:
+        assert item == q.get(rank=0, epoch=0)
+
+
+def test_get():
+    q = make_queue()
+
+    item = 0
+    q.put(rank=0, epoch=0, item=item)
+    assert q.get(rank=0, epoch=0, block=False) == item
+
+    item = 1
+    q.put(rank=0, epoch=0, item=item)
+    assert q.get(rank=0, epoch=0, timeout=0.2) == item
+
+    with pytest.raises(ValueError):
+        q.get(rank=0, epoch=0, timeout=-1)
+
+    with pytest.raises(Empty):
+        q.get_nowait(rank=0, epoch=0)
+
+    with pytest.raises(Empty):
+        q.get(rank=0, epoch=0, timeout=0.2)
+
+
+def test_get_async():
+    asyncio.run(_test_get_async())
+
+
+async def _test_get_async():
+    q = make_queue()
+
+    item = 0
+    await q.put_async(rank=0, epoch=0, item=item)
+    assert await q.get_async(rank=0, epoch=0, block=False) == item
+
+    item = 1
+    await q.put_async(rank=0, epoch=0, item=item)
+    assert await q.get_async(rank=0, epoch=0, timeout=0.2) == item
+
+    with pytest.raises(ValueError):
+        await q.get_async(rank=0, epoch=0, timeout=-1)
+
+    with pytest.raises(Empty):
+        await q.get_async(rank=0, epoch=0, block=False)
+
+    with pytest.raises(Empty):
+        await q.get_async(rank=0, epoch=0, timeout=0.2)
+
+
+def test_put():
+    q = make_queue(maxsize=1)
+
+    item = 0
+    q.put(rank=0, epoch=0, item=item, block=False)
+    assert q.get(rank=0, epoch=0) == item
+
+    item = 1
+    q.put(rank=0, epoch=0, item=item, timeout=0.2)
+    assert q.get(rank=0, epoch=0) == item
+
+    with pytest.raises(ValueError):
+        q.put(rank=0, epoch=0, item=0, timeout=-1)
+
+    q.put(rank=0, epoch=0, item=0)
+    with pytest.raises(Full):
+        q.put_nowait(rank=0, epoch=0, item=1)
+
+    with pytest.raises(Full):
+        q.put(rank=0, epoch=0, item=1, timeout=0.2)
+
+
+def test_put_async():
+    asyncio.run(_test_put_async())
+
+
+async def _test_put_async():
+    q = make_queue(maxsize=1)
+
+    item = 0
+    await q.put_async(rank=0, epoch=0, item=item, block=False)
+    assert await q.get_async(rank=0, epoch=0) == item
+
+    item = 1
+    await q.put_async(rank=0, epoch=0, item=item, timeout=0.2)
+    assert await q.get_async(rank=0, epoch=0) == item
+
+    with pytest.raises(ValueError):
+        await q.put_async(rank=0, epoch=0, item=0, timeout=-1)
+
+    await q.put_async(rank=0, epoch=0, item=0)
+    with pytest.raises(Full):
+        await q.put_async(rank=0, epoch=0, item=1, block=False)
+
+    with pytest.raises(Full):
+        await q.put_async(rank=0, epoch=0, item=1, timeout=0.2)
+
+
+def test_concurrent_get():
+    q = make_queue()
+    result = []
+
+    def getter():
+        result.append(q.get(rank=0, epoch=0))
+
+    t = threading.Thread(target=getter)
+    t.start()
+
+    with pytest.raises(Empty):
+        q.get_nowait(rank=0, epoch=0)
+
+    time.sleep(0.1)
+    assert t.is_alive()  # blocked, not canceled
+
+    q.put(rank=0, epoch=0, item=1)
+    t.join(timeout=5)
+    assert result == [1]
+
+
+def test_concurrent_put():
+    q = make_queue(maxsize=1)
+    q.put(rank=0, epoch=0, item=1)
+
+    t = threading.Thread(target=lambda: q.put(rank=0, epoch=0, item=2))
+    t.start()
+
+    with pytest.raises(Full):
+        q.put_nowait(rank=0, epoch=0, item=3)
+
+    time.sleep(0.1)
+    assert t.is_alive()  # blocked, not canceled
+
+    assert q.get(rank=0, epoch=0) == 1
+    t.join(timeout=5)
+    assert q.get(rank=0, epoch=0) == 2
+
+
+def test_batch():
+    q = make_queue(maxsize=1)
+
+    with pytest.raises(Full):
+        q.put_nowait_batch(rank=0, epoch=0, items=[1, 2])
+
+    with pytest.raises(Empty):
+        q.get_nowait_batch(rank=0, epoch=0, num_items=1)
+
+    big_q = make_queue(maxsize=100)
+    big_q.put_nowait_batch(rank=0, epoch=0, items=list(range(100)))
+    assert big_q.get_nowait_batch(rank=0, epoch=0, num_items=100) == list(
+        range(100)
+    )
+
+
+def test_qsize():
+    q = make_queue()
+    items = list(range(10))
+    size = 0
+    assert q.qsize(rank=0, epoch=0) == size
+    for item in items:
+        q.put(rank=0, epoch=0, item=item)
+        size += 1
+        assert q.qsize(rank=0, epoch=0) == size
+    for item in items:
+        assert q.get(rank=0, epoch=0) == item
+        size -= 1
+        assert q.qsize(rank=0, epoch=0) == size
+
+
+def test_shutdown():
+    q = make_queue()
+    q.shutdown()
+    assert q.core is None
+    with pytest.raises(RuntimeError):
+        q.empty(rank=0, epoch=0)
+
+
+def test_get_batch_blocks_then_drains():
+    q = make_queue()
+    q.put_nowait_batch(rank=0, epoch=0, items=[1, 2, 3])
+    assert q.get_batch(rank=0, epoch=0) == [1, 2, 3]
+
+    result = []
+
+    def getter():
+        result.append(q.get_batch(rank=0, epoch=0))
+
+    t = threading.Thread(target=getter)
+    t.start()
+    time.sleep(0.1)
+    assert t.is_alive()
+    q.put(rank=0, epoch=0, item=42)
+    t.join(timeout=5)
+    assert result == [[42]]
+
+
+def test_epoch_window_backpressure():
+    # max_concurrent_epochs=2: starting epoch 2 must block until epoch 0's
+    # producers are done AND its queues are joined
+    # (reference batch_queue.py:395-418).
+    q = BatchQueue(num_epochs=4, num_trainers=2, max_concurrent_epochs=2)
+    q.new_epoch(0)
+    q.new_epoch(1)
+    for rank in range(2):
+        q.put(rank=rank, epoch=0, item="x")
+        q.producer_done(rank=rank, epoch=0)
+
+    entered = threading.Event()
+
+    def advance():
+        q.new_epoch(2)
+        entered.set()
+
+    t = threading.Thread(target=advance)
+    t.start()
+    time.sleep(0.2)
+    # Producers done, but items (incl. sentinel) not task_done'd yet.
+    assert not entered.is_set()
+
+    for rank in range(2):
+        assert q.get(rank=rank, epoch=0) == "x"
+        assert q.get(rank=rank, epoch=0) is None  # producer-done sentinel
+        q.task_done(rank=rank, epoch=0, num_items=2)
+
+    t.join(timeout=5)
+    assert entered.is_set()
+
+
+def test_wait_until_all_epochs_done():
+    q = BatchQueue(num_epochs=2, num_trainers=1, max_concurrent_epochs=2)
+    q.new_epoch(0)
+    q.new_epoch(1)
+    q.producer_done(rank=0, epoch=0)
+    q.get(rank=0, epoch=0)
+    q.task_done(rank=0, epoch=0)
+
+    finished = threading.Event()
+
+    def waiter():
+        q.wait_until_all_epochs_done()
+        finished.set()
+
+    t = threading.Thread(target=waiter)
+    t.start()
+    time.sleep(0.1)
+    assert not finished.is_set()
+
+    q.producer_done(rank=0, epoch=1)
+    q.get(rank=0, epoch=1)
+    q.task_done(rank=0, epoch=1)
+    t.join(timeout=5)
+    assert finished.is_set()
+
+
+def test_task_done_too_many_times():
+    q = make_queue()
+    q.put(rank=0, epoch=0, item=1)
+    q.get(rank=0, epoch=0)
+    q.task_done(rank=0, epoch=0)
+    with pytest.raises(ValueError):
+        q.task_done(rank=0, epoch=0)
+
+
+def test_named_queue_cross_connection():
+    # Server/client parity with the reference's named actor + connect retry
+    # (reference batch_queue.py:358-380, dataset.py:56-83).
+    name = f"test_q_{time.time_ns()}"
+    server_q = BatchQueue(
+        num_epochs=1,
+        num_trainers=2,
+        max_concurrent_epochs=1,
+        name=name,
+        connect=False,
+    )
+    client_q = BatchQueue(
+        num_epochs=1,
+        num_trainers=2,
+        max_concurrent_epochs=1,
+        name=name,
+        connect=True,
+    )
+    client_q.ready()
+    server_q.put(rank=0, epoch=0, item={"payload": [1, 2, 3]})
+    assert client_q.get(rank=0, epoch=0) == {"payload": [1, 2, 3]}
+    client_q.put(rank=1, epoch=0, item="from-client")
+    assert server_q.get(rank=1, epoch=0) == "from-client"
+    with pytest.raises(Empty):
+        client_q.get_nowait(rank=0, epoch=0)
+    client_q.shutdown()
+    server_q.shutdown()
+
+
+def test_connect_retries_exhausted():
+    with pytest.raises(ValueError):
+        BatchQueue(
+            num_epochs=1,
+            num_trainers=1,
+            max_concurrent_epochs=1,
+            name=f"nonexistent_{time.time_ns()}",
+            connect=True,
+            connect_retries=2,
+        )
+
+
+def test_pull_from_streaming_batch_queue():
+    # End-to-end streaming pull modeling the real consumer: epochs delimited
+    # by None sentinels (reference test :231-288, with plain items instead of
+    # ObjectRefs).
+    num_batches = 5
+    batch_size = 4
+    q = BatchQueue(
+        num_epochs=num_batches, num_trainers=1, max_concurrent_epochs=1
+    )
+    consumed = []
+
+    def consume():
+        epoch = 0
+        is_done = False
+        while not is_done:
+            for item in q.get_batch(rank=0, epoch=epoch):
+                if item is None:
+                    epoch += 1
+                    if epoch >= num_batches:
+                        is_done = True
+                    break
+                consumed.append(item)
+                time.sleep(0.01)
+
+    t = threading.Thread(target=consume)
+    t.start()
+    data = list(range(batch_size * num_batches))
+    for epoch, idx in enumerate(range(0, len(data), batch_size)):
+        time.sleep(0.05)
+        q.put_nowait_batch(
+            rank=0, epoch=epoch, items=data[idx : idx + batch_size]
+        )
+        q.put_nowait(rank=0, epoch=epoch, item=None)
+    t.join(timeout=30)
+    assert not t.is_alive()
+    assert len(consumed) == len(data)
+    assert set(consumed) == set(data)
