@@ -1,0 +1,233 @@
+// Torch bindings for the MI355X shuffle kernels (csrc/shuffle_kernels.hip).
+//
+// Exposes the fused reducer-side ops (gather-permute / unpack+cast+pack /
+// pack) to Python as ray_shuffling_data_loader_amd._rsdl_hip. All ops run on
+// the CALLER's current HIP stream, so the shuffle engine's side-stream
+// discipline (double-buffering against the trainer step) is controlled from
+// Python via torch.cuda.Stream.
+
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <vector>
+
+namespace rsdl {
+
+// Mirror of the definitions in shuffle_kernels.hip.
+enum DType : int32_t {
+  DT_F32 = 0,
+  DT_F64 = 1,
+  DT_I32 = 2,
+  DT_I64 = 3,
+  DT_F16 = 4,
+  DT_BF16 = 5,
+  DT_U8 = 6,
+};
+
+struct ColDesc {
+  int64_t col_ptr;
+  int32_t packed_off;
+  int32_t src_dtype;
+  int32_t dst_dtype;
+  int32_t numel;
+};
+
+struct ColTable {
+  ColDesc cols[128];
+};
+
+void launch_gather_rows(const void* src, void* dst, const int64_t* perm,
+                        int64_t n_rows, int64_t row_bytes, hipStream_t stream);
+void launch_gather_rows_idx32(const void* src, void* dst, const int32_t* perm,
+                              int64_t n_rows, int64_t row_bytes,
+                              hipStream_t stream);
+void launch_unpack_permute(const void* packed, int64_t row_stride,
+                           const int64_t* perm, const ColTable& table,
+                           int32_t num_cols, int64_t n_rows,
+                           hipStream_t stream);
+void launch_pack_columns(void* packed, int64_t row_stride, const int64_t* perm,
+                         const ColTable& table, int32_t num_cols,
+                         int64_t n_rows, hipStream_t stream);
+
+namespace {
+
+int32_t dtype_code(at::ScalarType st) {
+  switch (st) {
+    case at::kFloat: return DT_F32;
+    case at::kDouble: return DT_F64;
+    case at::kInt: return DT_I32;
+    case at::kLong: return DT_I64;
+    case at::kHalf: return DT_F16;
+    case at::kBFloat16: return DT_BF16;
+    case at::kByte: return DT_U8;
+    default:
+      TORCH_CHECK(false, "unsupported dtype for shuffle ops: ", st);
+  }
+}
+
+hipStream_t current_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+int64_t row_bytes_of(const at::Tensor& t) {
+  TORCH_CHECK(t.dim() == 2, "packed tensor must be 2-D [rows, row_bytes]");
+  TORCH_CHECK(t.is_contiguous(), "packed tensor must be contiguous");
+  int64_t rb = t.size(1) * t.element_size();
+  TORCH_CHECK(rb % 16 == 0,
+              "packed row byte width must be a multiple of 16, got ", rb);
+  return rb;
+}
+
+// dst[i,:] = src[perm[i],:]
+at::Tensor gather_rows(const at::Tensor& src, const at::Tensor& perm) {
+  TORCH_CHECK(src.is_cuda() && perm.is_cuda(), "gather_rows: inputs on GPU");
+  TORCH_CHECK(perm.scalar_type() == at::kLong || perm.scalar_type() == at::kInt,
+              "perm must be int64 or int32");
+  TORCH_CHECK(perm.is_contiguous(), "perm must be contiguous");
+  int64_t rb = row_bytes_of(src);
+  int64_t n = perm.numel();
+  auto dst = at::empty({n, src.size(1)}, src.options());
+  if (n == 0) return dst;
+  if (perm.scalar_type() == at::kLong) {
+    launch_gather_rows(src.data_ptr(), dst.data_ptr(),
+                       perm.data_ptr<int64_t>(), n, rb, current_stream());
+  } else {
+    launch_gather_rows_idx32(src.data_ptr(), dst.data_ptr(),
+                             perm.data_ptr<int32_t>(), n, rb,
+                             current_stream());
+  }
+  return dst;
+}
+
+void gather_rows_out(const at::Tensor& src, const at::Tensor& perm,
+                     at::Tensor& dst) {
+  TORCH_CHECK(src.is_cuda() && perm.is_cuda() && dst.is_cuda(),
+              "gather_rows_out: inputs on GPU");
+  int64_t rb = row_bytes_of(src);
+  TORCH_CHECK(row_bytes_of(dst) == rb, "dst row width mismatch");
+  TORCH_CHECK(dst.size(0) >= perm.numel(), "dst too small");
+  TORCH_CHECK(perm.is_contiguous(), "perm must be contiguous");
+  int64_t n = perm.numel();
+  if (n == 0) return;
+  if (perm.scalar_type() == at::kLong) {
+    launch_gather_rows(src.data_ptr(), dst.data_ptr(),
+                       perm.data_ptr<int64_t>(), n, rb, current_stream());
+  } else if (perm.scalar_type() == at::kInt) {
+    launch_gather_rows_idx32(src.data_ptr(), dst.data_ptr(),
+                             perm.data_ptr<int32_t>(), n, rb,
+                             current_stream());
+  } else {
+    TORCH_CHECK(false, "perm must be int64 or int32");
+  }
+}
+
+// outs[c][i*numel+e] = cast(packed[perm[i]*stride + off_c + e*esz])
+void unpack_permute(const at::Tensor& packed,
+                    const c10::optional<at::Tensor>& perm,
+                    const std::vector<at::Tensor>& outs,
+                    const std::vector<int64_t>& packed_offsets,
+                    const std::vector<int64_t>& src_dtype_codes) {
+  TORCH_CHECK(packed.is_cuda(), "packed must be on GPU");
+  TORCH_CHECK(packed.scalar_type() == at::kByte, "packed must be uint8");
+  TORCH_CHECK(outs.size() == packed_offsets.size() &&
+                  outs.size() == src_dtype_codes.size(),
+              "descriptor length mismatch");
+  TORCH_CHECK(outs.size() <= 128, "at most 128 columns per launch");
+  int64_t stride = packed.size(1);
+  int64_t n_rows;
+  const int64_t* perm_ptr = nullptr;
+  if (perm.has_value()) {
+    TORCH_CHECK(perm->is_cuda() && perm->scalar_type() == at::kLong &&
+                    perm->is_contiguous(),
+                "perm must be contiguous int64 on GPU");
+    n_rows = perm->numel();
+    perm_ptr = perm->data_ptr<int64_t>();
+  } else {
+    n_rows = packed.size(0);
+  }
+  if (n_rows == 0 || outs.empty()) return;
+  ColTable table{};
+  for (size_t c = 0; c < outs.size(); ++c) {
+    const auto& o = outs[c];
+    TORCH_CHECK(o.is_cuda() && o.is_contiguous(),
+                "output column must be contiguous GPU tensor");
+    TORCH_CHECK(o.size(0) == n_rows, "output rows mismatch");
+    table.cols[c] = ColDesc{
+        reinterpret_cast<int64_t>(o.data_ptr()),
+        (int32_t)packed_offsets[c],
+        (int32_t)src_dtype_codes[c],
+        dtype_code(o.scalar_type()),
+        (int32_t)(o.numel() / n_rows),
+    };
+  }
+  launch_unpack_permute(packed.data_ptr(), stride, perm_ptr, table,
+                        (int32_t)outs.size(), n_rows, current_stream());
+}
+
+// packed[perm[i]*stride + off_c + e*esz] = cast(cols[c][i*numel+e])
+at::Tensor pack_columns(const std::vector<at::Tensor>& cols,
+                        const std::vector<int64_t>& packed_offsets,
+                        const std::vector<int64_t>& packed_dtype_codes,
+                        int64_t row_stride,
+                        const c10::optional<at::Tensor>& perm) {
+  TORCH_CHECK(!cols.empty(), "need at least one column");
+  TORCH_CHECK(cols.size() <= 128, "at most 128 columns per launch");
+  TORCH_CHECK(row_stride % 16 == 0, "row_stride must be multiple of 16");
+  int64_t n_rows = cols[0].size(0);
+  const int64_t* perm_ptr = nullptr;
+  if (perm.has_value()) {
+    TORCH_CHECK(perm->is_cuda() && perm->scalar_type() == at::kLong &&
+                    perm->is_contiguous(),
+                "perm must be contiguous int64 on GPU");
+    TORCH_CHECK(perm->numel() == n_rows, "perm length mismatch");
+    perm_ptr = perm->data_ptr<int64_t>();
+  }
+  auto packed = at::empty(
+      {n_rows, row_stride},
+      cols[0].options().dtype(at::kByte));
+  ColTable table{};
+  for (size_t c = 0; c < cols.size(); ++c) {
+    const auto& t = cols[c];
+    TORCH_CHECK(t.is_cuda() && t.is_contiguous(),
+                "input column must be contiguous GPU tensor");
+    TORCH_CHECK(t.size(0) == n_rows, "column rows mismatch");
+    table.cols[c] = ColDesc{
+        reinterpret_cast<int64_t>(t.data_ptr()),
+        (int32_t)packed_offsets[c],
+        (int32_t)packed_dtype_codes[c],
+        dtype_code(t.scalar_type()),
+        (int32_t)(t.numel() / n_rows),
+    };
+  }
+  if (n_rows > 0) {
+    launch_pack_columns(packed.data_ptr(), row_stride, perm_ptr, table,
+                        (int32_t)cols.size(), n_rows, current_stream());
+  }
+  return packed;
+}
+
+}  // namespace
+}  // namespace rsdl
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "MI355X (gfx950) fused shuffle kernels";
+  m.def("gather_rows", &rsdl::gather_rows, "row gather of packed rows",
+        py::arg("src"), py::arg("perm"));
+  m.def("gather_rows_out", &rsdl::gather_rows_out, py::arg("src"),
+        py::arg("perm"), py::arg("dst"));
+  m.def("unpack_permute", &rsdl::unpack_permute, py::arg("packed"),
+        py::arg("perm"), py::arg("outs"), py::arg("packed_offsets"),
+        py::arg("src_dtype_codes"));
+  m.def("pack_columns", &rsdl::pack_columns, py::arg("cols"),
+        py::arg("packed_offsets"), py::arg("packed_dtype_codes"),
+        py::arg("row_stride"), py::arg("perm") = c10::nullopt);
+  m.attr("DT_F32") = (int)rsdl::DT_F32;
+  m.attr("DT_F64") = (int)rsdl::DT_F64;
+  m.attr("DT_I32") = (int)rsdl::DT_I32;
+  m.attr("DT_I64") = (int)rsdl::DT_I64;
+  m.attr("DT_F16") = (int)rsdl::DT_F16;
+  m.attr("DT_BF16") = (int)rsdl::DT_BF16;
+  m.attr("DT_U8") = (int)rsdl::DT_U8;
+}

@@ -1,0 +1,354 @@
+"""ShuffleEngine: the per-epoch map -> exchange -> reduce pipeline.
+
+MI355X-native redesign of the reference's shuffle driver + map/reduce Ray
+tasks (reference: ray_shuffling_data_loader/shuffle.py:51-219). Instead of a
+central driver fanning out object-store tasks, every trainer process runs a
+symmetric engine worker:
+
+  map     : read my Parquet shard -> pack rows on GPU (io.read_files_packed);
+            the packed source block can stay RESIDENT IN HBM (288 GB/GPU)
+            so later epochs shuffle at HBM speed with zero disk/host traffic
+            ("in-memory shuffling, loads from disk once" is the reference's
+            own contract, shuffle.py:46-48 comment).
+  exchange: random destination-trainer assignment per row, GPU radix grouping
+            (ops.partition_rows), then ONE RCCL all-to-all over xGMI between
+            the per-GPU ranks (parallel.fabric.exchange_rows). This replaces
+            the reference's object-store all-to-all (shuffle.py:112-123).
+  reduce  : multinomial split of the received rows into this rank's reducer
+            partitions + full random permutation, executed as a single fused
+            gather kernel per partition (ops.gather_rows / unpack_permute —
+            the reference's pd.concat + sample(frac=1) + convert_to_tensor,
+            shuffle.py:192-194, torch_dataset.py:204-236).
+
+Statistical equivalence with the reference's two-stage randomness
+(per-row uniform reducer id, then per-reducer permutation): partitioning a
+uniform random permutation at multinomial(N, 1/R) boundaries yields the same
+joint distribution over (partition membership, within-partition order).
+
+The engine runs in a daemon thread per rank; epochs are pipelined under the
+consumer's max_concurrent_epochs window (BatchQueue.new_epoch gate), and all
+GPU work runs on a dedicated side HIP stream so the trainer's compute stream
+never stalls behind shuffle work.
+"""
+
+import threading
+import time
+from typing import Callable, Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+
+from ray_shuffling_data_loader_amd.io import (
+    fuse_schema,
+    infer_schema,
+    read_files_packed,
+)
+from ray_shuffling_data_loader_amd.ops.shuffle_ops import (
+    gather_rows,
+    partition_rows,
+    unpack_permute,
+)
+from ray_shuffling_data_loader_amd.parallel import fabric
+from ray_shuffling_data_loader_amd.utils.rowblock import RowBlock
+from ray_shuffling_data_loader_amd.utils.schema import (
+    Schema,
+    dtype_bytes,
+    homogeneous_dtype,
+)
+
+
+class ShuffleEngine:
+    def __init__(
+        self,
+        filenames: Sequence[str],
+        consumer,  # BatchConsumer
+        num_epochs: int,
+        num_reducers: int,
+        num_trainers: int,
+        rank: int = 0,
+        device: Optional[torch.device] = None,
+        seed: Optional[int] = None,
+        source_cache: str = "auto",  # "auto" | "device" | "host" | "none"
+        reader_threads: int = 8,
+        group=None,
+        schema: Optional[Schema] = None,
+        feature_matrix: Optional[Tuple[str, List[str]]] = None,
+        output: str = "auto",  # "auto" | "views" | "columns"
+        stats_collector=None,
+    ):
+        self.filenames = list(filenames)
+        self.consumer = consumer
+        self.num_epochs = num_epochs
+        self.num_reducers = num_reducers
+        self.num_trainers = num_trainers
+        self.group = group
+        self.stats = stats_collector
+        self.reader_threads = reader_threads
+        self.output = output
+        self.source_cache = source_cache
+
+        world, dist_rank, initialized = fabric.dist_info()
+        self.distributed = initialized and world > 1
+        if self.distributed:
+            if num_trainers != world:
+                raise ValueError(
+                    f"num_trainers={num_trainers} must equal "
+                    f"torch.distributed world size {world}"
+                )
+            if rank != dist_rank:
+                raise ValueError(
+                    f"rank={rank} disagrees with torch.distributed rank "
+                    f"{dist_rank}"
+                )
+        self.rank = rank
+
+        if device is None:
+            device = torch.device(
+                "cuda" if torch.cuda.is_available() else "cpu"
+            )
+        self.device = torch.device(device)
+
+        # File shard: symmetric striped split in distributed mode; the whole
+        # set in local mode (reference: one map task per file, any node —
+        # shuffle.py:111-114).
+        if self.distributed:
+            self.my_files = self.filenames[self.rank :: num_trainers]
+            self.owned_trainers = [self.rank]
+        else:
+            self.my_files = self.filenames
+            self.owned_trainers = list(range(num_trainers))
+
+        # Reducer -> trainer ownership: np.array_split parity
+        # (reference shuffle.py:125-126).
+        splits = np.array_split(np.arange(num_reducers), num_trainers)
+        self.reducers_per_trainer = [len(s) for s in splits]
+        if min(self.reducers_per_trainer) == 0:
+            raise ValueError(
+                f"num_reducers={num_reducers} < num_trainers={num_trainers}"
+            )
+
+        # Destination-trainer sampling weights proportional to owned reducers
+        # (uniform over reducers overall).
+        probs = np.array(self.reducers_per_trainer, dtype=np.float64)
+        probs /= probs.sum()
+        self._dest_uniform = len(set(self.reducers_per_trainer)) == 1
+        self._dest_cum = torch.from_numpy(np.cumsum(probs)).to(
+            self.device, torch.float32
+        )
+
+        if schema is None:
+            if not self.filenames:
+                raise ValueError("need filenames or an explicit schema")
+            schema = infer_schema(self.filenames[0])
+        self.base_schema = schema
+        self.out_schema = fuse_schema(schema, feature_matrix)
+
+        if seed is None:
+            seed = int(time.time_ns() % (2**31))
+        # Per-rank independent stream; ranks draw iid assignments.
+        gen_seed = seed * 1000003 + self.rank
+        if self.device.type == "cuda":
+            self._gen = torch.Generator(device=self.device)
+        else:
+            self._gen = torch.Generator()
+        self._gen.manual_seed(gen_seed)
+
+        self._cached_source: Optional[torch.Tensor] = None
+        self._stream = (
+            torch.cuda.Stream(device=self.device)
+            if self.device.type == "cuda"
+            else None
+        )
+        self._thread: Optional[threading.Thread] = None
+        self._error: Optional[BaseException] = None
+        self.duration: Optional[float] = None
+
+    # ----- source ------------------------------------------------------------
+
+    def _get_source(self, epoch: int) -> torch.Tensor:
+        if self._cached_source is not None:
+            src = self._cached_source
+            if src.device != self.device:
+                src = src.to(self.device, non_blocking=True)
+            return src
+        if self.stats:
+            # One map "task" per file (reference shuffle.py:111-114).
+            for _ in self.my_files:
+                self.stats.map_start(epoch)
+        t0 = time.perf_counter()
+        src = read_files_packed(
+            self.my_files, self.base_schema, self.device, self.reader_threads
+        )
+        read_dur = time.perf_counter() - t0
+        if self.stats:
+            per_file = read_dur / max(1, len(self.my_files))
+            for _ in self.my_files:
+                self.stats.map_done(epoch, per_file, per_file)
+        if self.source_cache in ("auto", "device"):
+            self._cached_source = src
+        elif self.source_cache == "host":
+            self._cached_source = src.cpu()
+        return src
+
+    # ----- epoch pipeline -----------------------------------------------------
+
+    def _sample_dest(self, n: int) -> torch.Tensor:
+        if self._dest_uniform:
+            return torch.randint(
+                self.num_trainers,
+                (n,),
+                device=self.device,
+                generator=self._gen,
+                dtype=torch.long,
+            )
+        u = torch.rand(n, device=self.device, generator=self._gen)
+        return torch.searchsorted(self._dest_cum, u, right=True).clamp_(
+            max=self.num_trainers - 1
+        )
+
+    def _make_partition(
+        self, rows: torch.Tensor, idx: torch.Tensor
+    ) -> RowBlock:
+        """One reducer partition: fused gather-permute (+ cast/pack) of the
+        selected rows into a RowBlock."""
+        schema = self.out_schema
+        hom = homogeneous_dtype(schema)
+        use_views = self.output == "views" or (
+            self.output == "auto" and hom is not None
+        )
+        if use_views and hom is not None:
+            esz = dtype_bytes(hom)
+            gathered = gather_rows(rows, idx)
+            typed = gathered.view(hom)  # [n, row_stride/esz]
+            cols: Dict[str, torch.Tensor] = {}
+            for spec in schema.columns:
+                o = schema.offsets[spec.name] // esz
+                if spec.numel == 1:
+                    cols[spec.name] = typed[:, o]
+                else:
+                    cols[spec.name] = typed[:, o : o + spec.numel]
+            return RowBlock(cols)
+        cols = unpack_permute(rows, schema, perm=idx.to(torch.long))
+        return RowBlock(cols)
+
+    def _shuffle_epoch(self, epoch: int) -> None:
+        if self.stats:
+            self.stats.epoch_start(epoch)
+        t0 = time.perf_counter()
+        src = self._get_source(epoch)
+        n = src.shape[0]
+
+        # --- map side: destination assignment + grouping + exchange --------
+        if self.distributed:
+            dest = self._sample_dest(n)
+            grouped, send_counts = partition_rows(
+                src, dest, self.num_trainers
+            )
+            recv, _ = fabric.exchange_rows(
+                grouped, send_counts, self.group
+            )
+            rows_per_trainer = {self.rank: recv}
+        elif self.num_trainers == 1:
+            rows_per_trainer = {0: src}
+        else:
+            dest = self._sample_dest(n)
+            grouped, counts = partition_rows(src, dest, self.num_trainers)
+            offs = torch.zeros(
+                self.num_trainers + 1, dtype=torch.long
+            )
+            torch.cumsum(counts.cpu(), 0, out=offs[1:])
+            rows_per_trainer = {
+                t: grouped[offs[t] : offs[t + 1]]
+                for t in range(self.num_trainers)
+            }
+
+        # --- reduce side: per owned trainer, multinomial reducer split +
+        #     fused permute-gather ------------------------------------------
+        for t in self.owned_trainers:
+            rows = rows_per_trainer[t]
+            m = rows.shape[0]
+            r_t = self.reducers_per_trainer[t]
+            # multinomial sizes == counts of iid uniform reducer assignment
+            assign = torch.randint(
+                r_t,
+                (m,),
+                device=self.device,
+                generator=self._gen,
+            )
+            sizes = torch.bincount(assign, minlength=r_t).cpu()
+            perm = torch.randperm(
+                m, device=self.device, generator=self._gen
+            )
+            offs = [0]
+            for s in sizes.tolist():
+                offs.append(offs[-1] + s)
+            parts = []
+            for k in range(r_t):
+                if self.stats:
+                    self.stats.reduce_start(epoch)
+                rt0 = time.perf_counter()
+                idx = perm[offs[k] : offs[k + 1]]
+                parts.append(self._make_partition(rows, idx))
+                if self.stats:
+                    self.stats.reduce_done(
+                        epoch, time.perf_counter() - rt0
+                    )
+            if self._stream is not None:
+                # Partitions are produced on the side stream; make them safe
+                # for the consumer's stream before they enter the queue.
+                torch.cuda.current_stream(self.device).synchronize()
+            self.consumer.consume(t, epoch, parts)
+            self.consumer.producer_done(t, epoch)
+        if self.stats:
+            self.stats.epoch_done(epoch, time.perf_counter() - t0)
+
+    # ----- driver loop --------------------------------------------------------
+
+    def run(self) -> float:
+        """The shuffle driver loop (reference shuffle.py:51-86): per epoch,
+        gate on the consumer's epoch window, then shuffle."""
+        start = time.perf_counter()
+        ctx = (
+            torch.cuda.stream(self._stream)
+            if self._stream is not None
+            else _nullcontext()
+        )
+        with ctx:
+            for epoch in range(self.num_epochs):
+                self.consumer.wait_until_ready(epoch)
+                self._shuffle_epoch(epoch)
+        self.consumer.wait_until_all_epochs_done()
+        self.duration = time.perf_counter() - start
+        if self.stats:
+            self.stats.trial_done(self.duration)
+        return self.duration
+
+    def start(self) -> None:
+        """Run the driver loop on a background thread."""
+
+        def _target():
+            try:
+                self.run()
+            except BaseException as e:  # surfaced via join()
+                self._error = e
+
+        self._thread = threading.Thread(
+            target=_target, name=f"rsdl-shuffle-r{self.rank}", daemon=True
+        )
+        self._thread.start()
+
+    def join(self, timeout: Optional[float] = None) -> None:
+        if self._thread is not None:
+            self._thread.join(timeout)
+            if self._error is not None:
+                raise RuntimeError(
+                    "shuffle engine worker failed"
+                ) from self._error
+
+
+class _nullcontext:
+    def __enter__(self):
+        return None
+
+    def __exit__(self, *a):
+        return False

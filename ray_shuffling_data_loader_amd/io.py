@@ -1,0 +1,117 @@
+"""Parquet ingest: files -> packed row tensors.
+
+Replaces the reference's ``pd.read_parquet`` map-stage load (reference
+shuffle.py:151) with multi-threaded pyarrow reads (host C++ Arrow decode,
+GIL-released) that land in torch tensors and are packed into the row-major
+exchange layout — on GPU via the pack_columns HIP kernel when a device is
+given, else via numpy strided views.
+"""
+
+from concurrent.futures import ThreadPoolExecutor
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+import pyarrow.parquet as pq
+import torch
+
+from ray_shuffling_data_loader_amd.ops.shuffle_ops import pack_columns
+from ray_shuffling_data_loader_amd.utils.schema import (
+    ColumnSpec,
+    NUMPY_TO_TORCH_DTYPE,
+    Schema,
+)
+
+
+def infer_schema(filename: str, drop_columns: Sequence[str] = ()) -> Schema:
+    """Schema from a Parquet file's arrow schema (fixed-width columns)."""
+    pf = pq.ParquetFile(filename)
+    specs = []
+    for fld in pf.schema_arrow:
+        if fld.name in drop_columns:
+            continue
+        np_dt = np.dtype(fld.type.to_pandas_dtype())
+        specs.append(ColumnSpec(fld.name, NUMPY_TO_TORCH_DTYPE[np_dt], 1))
+    return Schema(specs)
+
+
+def fuse_schema(base: Schema, feature_matrix: Optional[Tuple[str, List[str]]]
+                ) -> Schema:
+    """Fuse a run of same-dtype scalar columns that are CONSECUTIVE in the
+    packed layout into one vector column (e.g. 100 float32 feature columns
+    -> one [N,100] 'features' matrix). Offsets must already be contiguous —
+    this is a reinterpretation of the packed bytes, not a repack."""
+    if feature_matrix is None:
+        return base
+    name, members = feature_matrix
+    dts = {base.col(m).dtype for m in members}
+    if len(dts) != 1:
+        raise ValueError("feature-matrix columns must share one dtype")
+    dt = dts.pop()
+    offs = sorted(base.offsets[m] for m in members)
+    esz = offs[1] - offs[0] if len(offs) > 1 else 0
+    for a, b in zip(offs, offs[1:]):
+        if b - a != esz:
+            raise ValueError(
+                "feature-matrix columns are not contiguous in packed layout"
+            )
+    fused_specs = [ColumnSpec(name, dt, len(members))]
+    for c in base.columns:
+        if c.name not in members:
+            fused_specs.append(c)
+    fused = Schema(fused_specs)
+    # The fused layout must reinterpret the SAME bytes.
+    if fused.offsets[name] != offs[0] or fused.row_stride != base.row_stride:
+        raise ValueError(
+            "fused schema layout does not match base packed layout "
+            f"({fused.offsets[name]} != {offs[0]} or "
+            f"{fused.row_stride} != {base.row_stride})"
+        )
+    return fused
+
+
+def read_file_columns(
+    filename: str,
+    schema: Schema,
+) -> Dict[str, np.ndarray]:
+    """Read one Parquet file into contiguous numpy columns (only the schema's
+    columns)."""
+    table = pq.read_table(filename, columns=schema.names)
+    out = {}
+    for spec in schema.columns:
+        arr = table.column(spec.name).to_numpy(zero_copy_only=False)
+        # Arrow hands back read-only buffers; torch.from_numpy needs
+        # writable C-contiguous memory (copies only when required).
+        out[spec.name] = np.require(arr, requirements=["C", "W"])
+    return out
+
+
+def read_files_packed(
+    filenames: Sequence[str],
+    schema: Schema,
+    device: torch.device,
+    reader_threads: int = 8,
+) -> torch.Tensor:
+    """Read + pack many Parquet files into one [N, row_stride] uint8 tensor
+    on ``device``. Reads are threaded (Arrow releases the GIL); packing runs
+    on the GPU (pack_columns kernel) when device is cuda."""
+    if not filenames:
+        return torch.empty(0, schema.row_stride, dtype=torch.uint8,
+                           device=device)
+    use_gpu = device.type == "cuda"
+
+    def load(fn):
+        return read_file_columns(fn, schema)
+
+    packed_parts = []
+    with ThreadPoolExecutor(max_workers=max(1, reader_threads)) as pool:
+        for cols_np in pool.map(load, filenames):
+            cols = {}
+            for name, arr in cols_np.items():
+                t = torch.from_numpy(arr)
+                if use_gpu:
+                    t = t.to(device, non_blocking=True)
+                cols[name] = t
+            packed_parts.append(pack_columns(cols, schema))
+    if len(packed_parts) == 1:
+        return packed_parts[0]
+    return torch.cat(packed_parts, dim=0)

@@ -1,0 +1,218 @@
+"""Shuffle hot ops: gather-permute, pack, unpack — HIP on GPU, torch/numpy on
+CPU.
+
+GPU tensors dispatch to the gfx950 HIP kernels in ``_rsdl_hip``
+(csrc/shuffle_kernels.hip). If the extension is missing on a GPU box the ops
+raise immediately — there is deliberately NO eager/PyTorch fallback on GPU,
+so a silent-slow path can't masquerade as the native one. CPU tensors use
+torch/numpy implementations, which double as the fp32 oracle for the GPU
+numerics tests.
+
+Reference ops replaced here: ``pd.concat`` + ``df.sample(frac=1)``
+(reference shuffle.py:192-194) -> :func:`gather_rows`;
+``convert_to_tensor`` column cast/pack (reference torch_dataset.py:204-236)
+-> :func:`unpack_permute`.
+"""
+
+from typing import Dict, List, Optional
+
+import numpy as np
+import torch
+
+from ray_shuffling_data_loader_amd.utils.schema import (
+    Schema,
+    TORCH_TO_NUMPY_DTYPE,
+    dtype_bytes,
+)
+
+_hip = None
+_hip_err = None
+
+
+def _load_hip():
+    global _hip, _hip_err
+    if _hip is None and _hip_err is None:
+        try:
+            from ray_shuffling_data_loader_amd import _rsdl_hip
+
+            _hip = _rsdl_hip
+        except ImportError as e:  # remember why, re-raise on GPU use
+            _hip_err = e
+    if _hip is None:
+        raise RuntimeError(
+            "ray_shuffling_data_loader_amd._rsdl_hip extension is not built "
+            "but a GPU tensor was passed to a shuffle op. Build it with "
+            "`python setup.py build_ext --inplace` (PYTORCH_ROCM_ARCH="
+            f"gfx950). Original import error: {_hip_err}"
+        )
+    return _hip
+
+
+_DT_CODE = None
+
+
+def _dtype_code(dtype: torch.dtype) -> int:
+    global _DT_CODE
+    if _DT_CODE is None:
+        hip = _load_hip()
+        _DT_CODE = {
+            torch.float32: hip.DT_F32,
+            torch.float64: hip.DT_F64,
+            torch.int32: hip.DT_I32,
+            torch.int64: hip.DT_I64,
+            torch.float16: hip.DT_F16,
+            torch.bfloat16: hip.DT_BF16,
+            torch.uint8: hip.DT_U8,
+        }
+    return _DT_CODE[dtype]
+
+
+# ---------------------------------------------------------------------------
+# gather_rows: out[i,:] = src[perm[i],:]  on a packed byte matrix.
+# ---------------------------------------------------------------------------
+
+
+def gather_rows(
+    src: torch.Tensor,
+    perm: torch.Tensor,
+    out: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """Row gather of a 2-D contiguous tensor. The reducer-side full row
+    permutation (reference shuffle.py:194) fused with the implicit concat."""
+    if src.is_cuda:
+        hip = _load_hip()
+        if out is not None:
+            hip.gather_rows_out(src, perm, out)
+            return out[: perm.numel()]
+        return hip.gather_rows(src, perm)
+    res = torch.index_select(src, 0, perm.to(torch.long))
+    if out is not None:
+        out[: perm.numel()] = res
+        return out[: perm.numel()]
+    return res
+
+
+# ---------------------------------------------------------------------------
+# pack / unpack between column tensors and a packed row-major byte matrix.
+# ---------------------------------------------------------------------------
+
+
+def _np_strided_view(
+    packed_np: np.ndarray, schema: Schema, name: str, numel: int, np_dt
+) -> np.ndarray:
+    """Writable strided numpy view of one column inside a packed row buffer."""
+    n = packed_np.shape[0]
+    stride = packed_np.shape[1]
+    itemsize = np_dt.itemsize
+    return np.ndarray(
+        shape=(n, numel),
+        dtype=np_dt,
+        buffer=packed_np,
+        offset=schema.offsets[name],
+        strides=(stride, itemsize),
+    )
+
+
+def pack_columns(
+    columns: Dict[str, torch.Tensor],
+    schema: Schema,
+    perm: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """Interleave column tensors into packed rows (cast to the schema dtype
+    per column). ``perm`` scatters row i of the input to row perm[i] of the
+    output."""
+    first = next(iter(columns.values()))
+    n = first.shape[0]
+    if first.is_cuda:
+        hip = _load_hip()
+        cols, offs, codes = [], [], []
+        for spec in schema.columns:
+            t = columns[spec.name]
+            cols.append(t.contiguous())
+            offs.append(schema.offsets[spec.name])
+            codes.append(_dtype_code(spec.dtype))
+        return hip.pack_columns(
+            cols, offs, codes, schema.row_stride, perm
+        )
+    packed = torch.zeros(n, schema.row_stride, dtype=torch.uint8)
+    packed_np = packed.numpy()
+    perm_np = perm.cpu().numpy() if perm is not None else None
+    for spec in schema.columns:
+        t = columns[spec.name].detach()
+        np_dt = TORCH_TO_NUMPY_DTYPE[spec.dtype]
+        view = _np_strided_view(packed_np, schema, spec.name, spec.numel, np_dt)
+        src = t.cpu().numpy().reshape(n, spec.numel).astype(np_dt, copy=False)
+        if perm_np is not None:
+            view[perm_np] = src
+        else:
+            view[:] = src
+    return packed
+
+
+def unpack_permute(
+    packed: torch.Tensor,
+    schema: Schema,
+    perm: Optional[torch.Tensor] = None,
+    out_dtypes: Optional[Dict[str, torch.dtype]] = None,
+) -> Dict[str, torch.Tensor]:
+    """Fused gather-permute + per-column cast + pack into contiguous torch
+    column tensors: out[name][i] = cast(packed[perm[i], off_name]).
+
+    This single op replaces the reference's reduce-side
+    concat+sample+convert_to_tensor chain (shuffle.py:192-194 +
+    torch_dataset.py:204-236)."""
+    out_dtypes = out_dtypes or {}
+    n = perm.numel() if perm is not None else packed.shape[0]
+    if packed.is_cuda:
+        hip = _load_hip()
+        outs, offs, codes, names = [], [], [], []
+        result: Dict[str, torch.Tensor] = {}
+        for spec in schema.columns:
+            dst_dt = out_dtypes.get(spec.name, spec.dtype)
+            shape = (n,) if spec.numel == 1 else (n, spec.numel)
+            o = torch.empty(shape, dtype=dst_dt, device=packed.device)
+            outs.append(o)
+            offs.append(schema.offsets[spec.name])
+            codes.append(_dtype_code(spec.dtype))
+            names.append(spec.name)
+            result[spec.name] = o
+        hip.unpack_permute(packed, perm, outs, offs, codes)
+        return result
+    packed_np = packed.numpy()
+    perm_np = perm.cpu().numpy() if perm is not None else None
+    result = {}
+    for spec in schema.columns:
+        np_dt = TORCH_TO_NUMPY_DTYPE[spec.dtype]
+        view = _np_strided_view(packed_np, schema, spec.name, spec.numel, np_dt)
+        arr = view[perm_np] if perm_np is not None else np.ascontiguousarray(
+            view
+        )
+        t = torch.from_numpy(arr)
+        dst_dt = out_dtypes.get(spec.name, spec.dtype)
+        if dst_dt != spec.dtype:
+            t = t.to(dst_dt)
+        if spec.numel == 1:
+            t = t.reshape(n)
+        result[spec.name] = t
+    return result
+
+
+# ---------------------------------------------------------------------------
+# Row partition by destination (map-side scatter; reference shuffle.py:156-161
+# boolean-mask loop). v1: stable sort + native gather; the sort keys are tiny
+# (uint8-range dest ids) so torch's radix sort is cheap next to the row move.
+# ---------------------------------------------------------------------------
+
+
+def partition_rows(
+    packed: torch.Tensor,
+    dest: torch.Tensor,
+    num_dests: int,
+):
+    """Group packed rows by destination id. Returns (regrouped rows, counts
+    per destination). Row order within a destination is the stable input
+    order (like the reference's boolean-mask partition)."""
+    counts = torch.bincount(dest, minlength=num_dests)
+    order = torch.argsort(dest, stable=True)
+    grouped = gather_rows(packed, order)
+    return grouped, counts

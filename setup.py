@@ -40,7 +40,7 @@ try:
     hip_sources = ["csrc/shuffle_ops.cpp", "csrc/shuffle_kernels.hip"]
     if all(os.path.exists(s) for s in hip_sources):
         ext_modules.append(
-            torch_cpp_ext.CppExtension(
+            torch_cpp_ext.CUDAExtension(
                 name="ray_shuffling_data_loader_amd._rsdl_hip",
                 sources=hip_sources,
                 extra_compile_args={

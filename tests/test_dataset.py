@@ -1,0 +1,242 @@
+"""End-to-end ShufflingDataset / TorchShufflingDataset tests on tiny
+synthetic Parquet (CPU). Mirrors the reference's example-as-smoke coverage
+(reference run_ci_examples.sh + dataset.py:208-252) plus correctness
+invariants the reference never asserted: every row exactly once per epoch,
+batch sizing with leftover carry, drop_last, the set_epoch guard, and
+shuffledness across epochs."""
+
+import numpy as np
+import pytest
+import torch
+
+from ray_shuffling_data_loader_amd.data_generation import (
+    DATA_SPEC,
+    float_data_spec,
+    generate_data,
+)
+from ray_shuffling_data_loader_amd.dataset import ShufflingDataset
+from ray_shuffling_data_loader_amd.torch_dataset import TorchShufflingDataset
+from ray_shuffling_data_loader_amd.utils.schema import NUMPY_TO_TORCH_DTYPE
+
+
+@pytest.fixture(scope="module")
+def small_data(tmp_path_factory):
+    data_dir = tmp_path_factory.mktemp("parquet")
+    num_rows = 20000
+    filenames, _ = generate_data(num_rows, 4, 2, 0.0, str(data_dir))
+    return list(filenames), num_rows
+
+
+@pytest.fixture(scope="module")
+def float_data(tmp_path_factory):
+    data_dir = tmp_path_factory.mktemp("parquet_float")
+    num_rows = 10000
+    spec = float_data_spec(8)
+    filenames, _ = generate_data(
+        num_rows, 2, 1, 0.0, str(data_dir), spec=spec, include_key=False
+    )
+    return list(filenames), num_rows
+
+
+def collect_epoch(ds, epoch):
+    ds.set_epoch(epoch)
+    return list(iter(ds))
+
+
+def test_every_row_exactly_once(small_data):
+    filenames, num_rows = small_data
+    batch_size = 1024
+    num_epochs = 2
+    ds = ShufflingDataset(
+        filenames,
+        num_epochs,
+        num_trainers=1,
+        batch_size=batch_size,
+        rank=0,
+        num_reducers=4,
+        seed=1234,
+    )
+    for epoch in range(num_epochs):
+        batches = collect_epoch(ds, epoch)
+        keys = torch.cat([b["key"] for b in batches])
+        assert len(keys) == num_rows
+        assert torch.equal(
+            torch.sort(keys).values, torch.arange(num_rows)
+        ), "each row must appear exactly once per epoch"
+        # All but the last batch must be exactly batch_size.
+        sizes = [len(b) for b in batches]
+        assert all(s == batch_size for s in sizes[:-1])
+        assert sizes[-1] == num_rows - batch_size * (len(sizes) - 1)
+
+
+def test_epochs_are_differently_shuffled(small_data):
+    filenames, num_rows = small_data
+    ds = ShufflingDataset(
+        filenames,
+        2,
+        num_trainers=1,
+        batch_size=5000,
+        rank=0,
+        num_reducers=4,
+        seed=99,
+    )
+    keys0 = torch.cat([b["key"] for b in collect_epoch(ds, 0)])
+    keys1 = torch.cat([b["key"] for b in collect_epoch(ds, 1)])
+    assert not torch.equal(keys0, keys1)
+    assert not torch.equal(keys0, torch.arange(num_rows))
+
+
+def test_drop_last(small_data):
+    filenames, num_rows = small_data
+    batch_size = 1536  # 20000 % 1536 != 0
+    ds = ShufflingDataset(
+        filenames,
+        1,
+        num_trainers=1,
+        batch_size=batch_size,
+        rank=0,
+        drop_last=True,
+        num_reducers=4,
+    )
+    batches = collect_epoch(ds, 0)
+    assert all(len(b) == batch_size for b in batches)
+    assert len(batches) == num_rows // batch_size
+
+
+def test_set_epoch_guard(small_data):
+    filenames, _ = small_data
+    ds = ShufflingDataset(
+        filenames, 2, num_trainers=1, batch_size=4096, rank=0, num_reducers=2
+    )
+    with pytest.raises(ValueError, match="set_epoch"):
+        next(iter(ds))
+    collect_epoch(ds, 0)
+    # Re-iterating the same epoch must raise.
+    with pytest.raises(ValueError, match="set_epoch"):
+        next(iter(ds))
+    collect_epoch(ds, 1)
+
+
+def test_local_multi_trainer_partition(small_data):
+    # Local central mode with 2 trainers: rank 0 produces for both; rows are
+    # disjoint and complete across trainers.
+    filenames, num_rows = small_data
+    ds0 = ShufflingDataset(
+        filenames,
+        1,
+        num_trainers=2,
+        batch_size=1000,
+        rank=0,
+        num_reducers=4,
+        queue_name=f"test_mt_{torch.initial_seed() % 100000}",
+        seed=7,
+    )
+    # Second consumer in the same process: use the dataset's own queue
+    # directly under rank 1 (cross-process connection is covered in
+    # test_batch_queue).
+    # Drain rank 1 concurrently: rank 0's final engine join waits for ALL
+    # trainers' queues (like the reference's ray.get(shuffle_result)).
+    import threading
+
+    q = ds0._batch_queue
+    got = []
+
+    def drain_rank1():
+        while True:
+            items = q.get_batch(1, 0)
+            done = bool(items) and items[-1] is None
+            if done:
+                items.pop()
+            got.extend(items)
+            n_ack = len(items) + (1 if done else 0)
+            if n_ack:
+                q.task_done(1, 0, n_ack)
+            if done:
+                return
+
+    t = threading.Thread(target=drain_rank1, daemon=True)
+    t.start()
+    ds0.set_epoch(0)
+    keys0 = torch.cat([b["key"] for b in ds0])
+    t.join(timeout=30)
+    assert not t.is_alive()
+    keys1 = torch.cat([b["key"] for b in got])
+    allk = torch.cat([keys0, keys1])
+    assert len(allk) == num_rows
+    assert torch.equal(torch.sort(allk).values, torch.arange(num_rows))
+
+
+def test_torch_dataset_types_and_shapes(small_data):
+    filenames, num_rows = small_data
+    feature_columns = list(DATA_SPEC.keys())
+    feature_types = [
+        NUMPY_TO_TORCH_DTYPE[np.dtype(dt)] for _, _, dt in DATA_SPEC.values()
+    ]
+    label_column = feature_columns.pop()
+    label_type = feature_types.pop()
+    batch_size = 4096
+    ds = TorchShufflingDataset(
+        filenames,
+        1,
+        num_trainers=1,
+        batch_size=batch_size,
+        rank=0,
+        num_reducers=4,
+        feature_columns=feature_columns,
+        feature_types=feature_types,
+        label_column=label_column,
+        label_type=label_type,
+    )
+    ds.set_epoch(0)
+    total = 0
+    for data, target in ds:
+        assert len(data) == len(feature_columns)
+        for t, dt in zip(data, feature_types):
+            assert t.dtype == dt
+            assert t.shape == (len(target), 1)
+        assert target.dtype == label_type
+        assert target.shape[1] == 1
+        total += len(target)
+    assert total == num_rows
+
+
+def test_torch_dataset_feature_matrix(float_data):
+    # MI355X fast path: fused [N, C] feature matrix, float32.
+    filenames, num_rows = float_data
+    feature_columns = [f"f{i}" for i in range(8)]
+    ds = TorchShufflingDataset(
+        filenames,
+        1,
+        num_trainers=1,
+        batch_size=1000,
+        rank=0,
+        num_reducers=2,
+        feature_columns=feature_columns,
+        label_column="labels",
+        feature_matrix=True,
+    )
+    ds.set_epoch(0)
+    total = 0
+    for data, target in ds:
+        assert len(data) == 1
+        assert data[0].shape == (len(target), 8)
+        assert data[0].dtype == torch.float32
+        total += len(target)
+    assert total == num_rows
+
+
+def test_source_cache_reuse(small_data):
+    # Cached source: epoch 2 must still see all rows exactly once.
+    filenames, num_rows = small_data
+    ds = ShufflingDataset(
+        filenames,
+        3,
+        num_trainers=1,
+        batch_size=2048,
+        rank=0,
+        num_reducers=4,
+        source_cache="host",
+    )
+    for epoch in range(3):
+        keys = torch.cat([b["key"] for b in collect_epoch(ds, epoch)])
+        assert torch.equal(torch.sort(keys).values, torch.arange(num_rows))

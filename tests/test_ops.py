@@ -1,0 +1,136 @@
+"""Unit tests for schema/pack/unpack/gather/partition ops (CPU reference
+implementations; the GPU kernels are tested against these in
+test_gpu_kernels.py)."""
+
+import numpy as np
+import pytest
+import torch
+
+from ray_shuffling_data_loader_amd.ops.shuffle_ops import (
+    gather_rows,
+    pack_columns,
+    partition_rows,
+    unpack_permute,
+)
+from ray_shuffling_data_loader_amd.utils.schema import (
+    ColumnSpec,
+    Schema,
+    homogeneous_dtype,
+)
+
+
+def make_schema_het():
+    return Schema(
+        [
+            ColumnSpec("a", torch.int64, 1),
+            ColumnSpec("b", torch.float32, 1),
+            ColumnSpec("c", torch.float64, 1),
+            ColumnSpec("d", torch.float32, 4),
+        ]
+    )
+
+
+def test_schema_layout_alignment():
+    s = make_schema_het()
+    # Descending dtype size: int64/float64 first, then float32s.
+    for spec in s.columns:
+        off = s.offsets[spec.name]
+        assert off % min(8, spec.row_bytes // spec.numel) == 0
+    assert s.row_stride % 16 == 0
+    assert s.payload_bytes == 8 + 8 + 4 + 16
+    assert s.row_stride == 48
+
+
+def test_schema_homogeneous():
+    s = Schema([ColumnSpec("x", torch.float32, 100),
+                ColumnSpec("y", torch.float32, 1)])
+    assert homogeneous_dtype(s) == torch.float32
+    assert s.offsets["x"] == 0
+    assert s.offsets["y"] == 400
+    assert s.row_stride == 416
+    assert homogeneous_dtype(make_schema_het()) is None
+
+
+def rand_columns(n, schema, seed=0):
+    g = torch.Generator().manual_seed(seed)
+    cols = {}
+    for spec in schema.columns:
+        shape = (n,) if spec.numel == 1 else (n, spec.numel)
+        if spec.dtype.is_floating_point:
+            cols[spec.name] = torch.randn(shape, generator=g).to(spec.dtype)
+        else:
+            cols[spec.name] = torch.randint(
+                0, 1000, shape, generator=g
+            ).to(spec.dtype)
+    return cols
+
+
+def test_pack_unpack_roundtrip():
+    schema = make_schema_het()
+    n = 1000
+    cols = rand_columns(n, schema)
+    packed = pack_columns(cols, schema)
+    assert packed.shape == (n, schema.row_stride)
+    out = unpack_permute(packed, schema)
+    for name in cols:
+        assert torch.equal(
+            out[name].reshape(cols[name].shape), cols[name]
+        ), name
+
+
+def test_unpack_with_permutation():
+    schema = make_schema_het()
+    n = 500
+    cols = rand_columns(n, schema)
+    packed = pack_columns(cols, schema)
+    perm = torch.randperm(n)
+    out = unpack_permute(packed, schema, perm=perm)
+    for name in cols:
+        expected = cols[name][perm]
+        assert torch.equal(out[name].reshape(expected.shape), expected), name
+
+
+def test_unpack_with_cast():
+    schema = Schema([ColumnSpec("x", torch.float64, 1)])
+    n = 100
+    cols = {"x": torch.randn(n, dtype=torch.float64)}
+    packed = pack_columns(cols, schema)
+    out = unpack_permute(
+        packed, schema, out_dtypes={"x": torch.float32}
+    )
+    assert out["x"].dtype == torch.float32
+    assert torch.allclose(out["x"], cols["x"].float())
+
+
+def test_gather_rows_cpu():
+    n, stride = 200, 32
+    src = torch.randint(0, 256, (n, stride), dtype=torch.uint8)
+    perm = torch.randperm(n)
+    out = gather_rows(src, perm)
+    assert torch.equal(out, src[perm])
+
+
+def test_partition_rows_cpu():
+    n, stride = 300, 16
+    src = torch.randint(0, 256, (n, stride), dtype=torch.uint8)
+    dest = torch.randint(0, 4, (n,))
+    grouped, counts = partition_rows(src, dest, 4)
+    assert counts.sum() == n
+    # Rows grouped by destination, stable order within each group.
+    off = 0
+    for d in range(4):
+        expected = src[dest == d]
+        got = grouped[off : off + counts[d]]
+        assert torch.equal(got, expected), d
+        off += int(counts[d])
+
+
+def test_pack_with_scatter_perm():
+    # pack_columns with perm scatters input row i to packed row perm[i].
+    schema = Schema([ColumnSpec("x", torch.float32, 1)])
+    n = 64
+    cols = {"x": torch.arange(n, dtype=torch.float32)}
+    perm = torch.randperm(n)
+    packed = pack_columns(cols, schema, perm=perm)
+    out = unpack_permute(packed, schema)
+    assert torch.equal(out["x"][perm], cols["x"])
