@@ -1,0 +1,210 @@
+"""GPU (MI355X) tests: HIP kernels vs the CPU torch/numpy oracle, and the
+end-to-end GPU loader path. All marked @pytest.mark.gpu."""
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from ray_shuffling_data_loader_amd.ops.shuffle_ops import (
+    gather_rows,
+    pack_columns,
+    partition_rows,
+    unpack_permute,
+)
+from ray_shuffling_data_loader_amd.utils.schema import ColumnSpec, Schema
+
+
+@pytest.fixture(scope="module")
+def dev():
+    assert torch.cuda.is_available()
+    return torch.device("cuda", 0)
+
+
+def test_hip_extension_loaded(dev):
+    # The native path must be the one that runs on GPU: importing must
+    # succeed, no silent fallback.
+    from ray_shuffling_data_loader_amd import _rsdl_hip  # noqa: F401
+
+
+def test_gather_rows_matches_cpu(dev):
+    n, stride = 4096, 416
+    src = torch.randint(0, 256, (n, stride), dtype=torch.uint8)
+    perm = torch.randperm(n)
+    expected = gather_rows(src, perm)  # CPU oracle (index_select)
+    got = gather_rows(src.to(dev), perm.to(dev)).cpu()
+    assert torch.equal(got, expected)
+
+
+def test_gather_rows_int32_perm(dev):
+    n, stride = 1000, 48
+    src = torch.randint(0, 256, (n, stride), dtype=torch.uint8)
+    perm = torch.randperm(n).to(torch.int32)
+    got = gather_rows(src.to(dev), perm.to(dev)).cpu()
+    assert torch.equal(got, src[perm.long()])
+
+
+def het_schema():
+    return Schema(
+        [
+            ColumnSpec("i64", torch.int64, 1),
+            ColumnSpec("f64", torch.float64, 1),
+            ColumnSpec("f32", torch.float32, 1),
+            ColumnSpec("vec", torch.float32, 4),
+            ColumnSpec("i32", torch.int32, 1),
+        ]
+    )
+
+
+def rand_cols(n, schema, device=None):
+    g = torch.Generator().manual_seed(7)
+    cols = {}
+    for spec in schema.columns:
+        shape = (n,) if spec.numel == 1 else (n, spec.numel)
+        if spec.dtype.is_floating_point:
+            t = torch.randn(shape, generator=g).to(spec.dtype)
+        else:
+            t = torch.randint(0, 100000, shape, generator=g).to(spec.dtype)
+        cols[spec.name] = t.to(device) if device else t
+    return cols
+
+
+def test_pack_columns_matches_cpu(dev):
+    schema = het_schema()
+    n = 2048
+    cols = rand_cols(n, schema)
+    expected = pack_columns(cols, schema)  # CPU oracle
+    got = pack_columns(
+        {k: v.to(dev) for k, v in cols.items()}, schema
+    ).cpu()
+    # Compare payload bytes only (padding bytes are uninitialized on GPU).
+    for spec in schema.columns:
+        off = schema.offsets[spec.name]
+        nb = spec.row_bytes
+        assert torch.equal(got[:, off : off + nb], expected[:, off : off + nb]), spec.name
+
+
+def test_unpack_permute_matches_cpu(dev):
+    schema = het_schema()
+    n = 2048
+    cols = rand_cols(n, schema)
+    packed = pack_columns(cols, schema)
+    perm = torch.randperm(n)
+    expected = unpack_permute(packed, schema, perm=perm)
+    got = unpack_permute(packed.to(dev), schema, perm=perm.to(dev))
+    for name, t in expected.items():
+        assert torch.equal(got[name].cpu(), t), name
+
+
+def test_unpack_permute_cast(dev):
+    # fp64 -> fp32 and int64 -> fp32 casts inside the fused kernel,
+    # verified against the plain fp32 torch reference.
+    schema = Schema(
+        [
+            ColumnSpec("a", torch.float64, 1),
+            ColumnSpec("b", torch.int64, 1),
+        ]
+    )
+    n = 4096
+    cols = {
+        "a": torch.randn(n, dtype=torch.float64),
+        "b": torch.randint(0, 10**6, (n,)),
+    }
+    packed = pack_columns(cols, schema)
+    perm = torch.randperm(n)
+    got = unpack_permute(
+        packed.to(dev),
+        schema,
+        perm=perm.to(dev),
+        out_dtypes={"a": torch.float32, "b": torch.float32},
+    )
+    assert got["a"].dtype == torch.float32
+    ref_a = cols["a"][perm].to(torch.float32)
+    ref_b = cols["b"][perm].to(torch.float32)
+    assert torch.equal(got["a"].cpu(), ref_a)
+    assert torch.equal(got["b"].cpu(), ref_b)
+
+
+def test_unpack_permute_bf16(dev):
+    schema = Schema([ColumnSpec("x", torch.float32, 8)])
+    n = 1024
+    cols = {"x": torch.randn(n, 8)}
+    packed = pack_columns(cols, schema)
+    perm = torch.randperm(n)
+    got = unpack_permute(
+        packed.to(dev), schema, perm=perm.to(dev),
+        out_dtypes={"x": torch.bfloat16},
+    )
+    ref = cols["x"][perm].to(torch.bfloat16)
+    assert torch.equal(got["x"].cpu(), ref)
+
+
+def test_partition_rows_gpu(dev):
+    n, stride = 100_000, 32
+    src = torch.randint(0, 256, (n, stride), dtype=torch.uint8, device=dev)
+    dest = torch.randint(0, 8, (n,), device=dev)
+    grouped, counts = partition_rows(src, dest, 8)
+    assert int(counts.sum()) == n
+    # Oracle on CPU.
+    g_cpu, c_cpu = partition_rows(src.cpu(), dest.cpu(), 8)
+    assert torch.equal(counts.cpu(), c_cpu)
+    assert torch.equal(grouped.cpu(), g_cpu)
+
+
+def test_end_to_end_gpu_loader(dev, tmp_path):
+    from ray_shuffling_data_loader_amd.data_generation import (
+        float_data_spec,
+        generate_data,
+    )
+    from ray_shuffling_data_loader_amd.dataset import ShufflingDataset
+
+    num_rows = 100_000
+    spec = float_data_spec(16)
+    filenames, _ = generate_data(
+        num_rows, 2, 1, 0.0, str(tmp_path), spec=spec, include_key=False
+    )
+    ds = ShufflingDataset(
+        list(filenames),
+        2,
+        num_trainers=1,
+        batch_size=10_000,
+        rank=0,
+        num_reducers=4,
+        device=dev,
+        feature_matrix=("__features__", [f"f{i}" for i in range(16)]),
+    )
+    for epoch in range(2):
+        ds.set_epoch(epoch)
+        total = 0
+        for b in ds:
+            assert b.device.type == "cuda"
+            assert b["__features__"].shape[1] == 16
+            total += len(b)
+        assert total == num_rows
+
+
+def test_gather_rows_bandwidth(dev):
+    # Perf sanity: row gather of a 1 GB packed block should sustain a large
+    # fraction of HBM bandwidth (this is the reducer-side hot op).
+    n, stride = 2_500_000, 416  # ~1 GB
+    src = torch.randint(0, 256, (n, stride), dtype=torch.uint8, device=dev)
+    perm = torch.randperm(n, device=dev)
+    out = torch.empty_like(src)
+    # warmup
+    gather_rows(src, perm, out=out)
+    torch.cuda.synchronize()
+    t0 = torch.cuda.Event(enable_timing=True)
+    t1 = torch.cuda.Event(enable_timing=True)
+    iters = 10
+    t0.record()
+    for _ in range(iters):
+        gather_rows(src, perm, out=out)
+    t1.record()
+    torch.cuda.synchronize()
+    ms = t0.elapsed_time(t1) / iters
+    gbps = 2 * n * stride / (ms * 1e-3) / 1e9  # read+write
+    print(f"gather_rows: {ms:.2f} ms, {gbps:.0f} GB/s")
+    # MI355X HBM3E ~6300 GB/s achievable; random-row gather should still
+    # clear 1 TB/s by a wide margin.
+    assert gbps > 1000, f"gather_rows too slow: {gbps:.0f} GB/s"
