@@ -50,6 +50,14 @@ def parse_args():
         "--device", type=str, default=None, help="override (cpu for debug)"
     )
     p.add_argument(
+        "--loader-output",
+        type=str,
+        default="auto",
+        choices=["auto", "views", "columns"],
+        help="views: zero-copy strided feature matrix; columns: fused "
+        "unpack kernel -> contiguous feature matrix (better GEMM input)",
+    )
+    p.add_argument(
         "--source-cache",
         type=str,
         default="auto",
@@ -141,6 +149,7 @@ def main():
         feature_matrix=True,
         device=device,
         source_cache=args.source_cache,
+        output=args.loader_output,
     )
 
     model = TabularMLP(args.num_cols).to(device)

@@ -100,6 +100,12 @@ class ShuffleEngine:
                     f"rank={rank} disagrees with torch.distributed rank "
                     f"{dist_rank}"
                 )
+            if self.group is None:
+                # Dedicated communicator: the engine's collectives run on a
+                # background thread and must not interleave with trainer
+                # collectives (DDP allreduce) on the default group.
+                # (__init__ runs on the main thread on every rank.)
+                self.group = fabric.get_shuffle_group()
         self.rank = rank
 
         if device is None:

@@ -49,8 +49,29 @@ def init_from_env(
     return dist.get_world_size(), dist.get_rank()
 
 
+_shuffle_group = None
+
+
+def get_shuffle_group():
+    """A dedicated communicator for the shuffle engine's collectives.
+
+    The engine runs on a background thread while DDP allreduces run on the
+    trainer's main thread; if both used the default process group their
+    collectives could interleave in different orders on different ranks
+    (undefined behavior / deadlock). A separate group gives the shuffle its
+    own NCCL/gloo communicator, within which its collectives are strictly
+    epoch-ordered. Must first be called from the MAIN thread on all ranks
+    in the same order (dataset construction does this)."""
+    global _shuffle_group
+    if _shuffle_group is None:
+        _shuffle_group = dist.new_group(
+            ranks=list(range(dist.get_world_size()))
+        )
+    return _shuffle_group
+
+
 def exchange_counts(
-    send_counts: torch.Tensor, group=None
+    send_counts: torch.Tensor, group=None, device=None
 ) -> torch.Tensor:
     """Size-exchange phase: every rank contributes its [world] row counts per
     destination; returns this rank's [world] receive counts (rows arriving
@@ -58,10 +79,16 @@ def exchange_counts(
     (SURVEY.md §7 hard part (b))."""
     world = dist.get_world_size(group)
     rank = dist.get_rank(group)
-    send_counts = send_counts.to(torch.long).cpu()
+    backend = dist.get_backend(group)
+    send_counts = send_counts.to(torch.long)
+    if backend == "nccl":
+        # RCCL collectives need device tensors.
+        send_counts = send_counts.to(device or "cuda")
+    else:
+        send_counts = send_counts.cpu()
     gathered = [torch.zeros_like(send_counts) for _ in range(world)]
     dist.all_gather(gathered, send_counts, group=group)
-    return torch.stack(gathered)[:, rank].contiguous()
+    return torch.stack(gathered)[:, rank].cpu().contiguous()
 
 
 def exchange_rows(
@@ -77,7 +104,7 @@ def exchange_rows(
     """
     world = dist.get_world_size(group)
     rank = dist.get_rank(group)
-    recv_counts = exchange_counts(send_counts, group)
+    recv_counts = exchange_counts(send_counts, group, device=grouped.device)
     in_splits = [int(c) for c in send_counts]
     out_splits = [int(c) for c in recv_counts]
     recv = torch.empty(

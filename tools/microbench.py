@@ -1,0 +1,115 @@
+#!/usr/bin/env python3
+"""GPU micro-benchmarks for the shuffle hot ops + trainer GEMM input layout.
+
+Run on an MI355X box:  python tools/microbench.py
+Writes a summary to stdout (capture into gpurun_out/)."""
+
+import time
+
+import torch
+
+from ray_shuffling_data_loader_amd.ops.shuffle_ops import (
+    gather_rows,
+    pack_columns,
+    partition_rows,
+    unpack_permute,
+)
+from ray_shuffling_data_loader_amd.utils.schema import ColumnSpec, Schema
+
+
+def timeit_gpu(fn, iters=20, warmup=3):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = torch.cuda.Event(enable_timing=True)
+    t1 = torch.cuda.Event(enable_timing=True)
+    t0.record()
+    for _ in range(iters):
+        fn()
+    t1.record()
+    torch.cuda.synchronize()
+    return t0.elapsed_time(t1) / iters  # ms
+
+
+def main():
+    assert torch.cuda.is_available()
+    dev = torch.device("cuda", 0)
+    print(f"device: {torch.cuda.get_device_name(0)}")
+
+    n = 2_500_000
+    stride = 416  # 100 fp32 features + label + pad
+    src = torch.randint(0, 256, (n, stride), dtype=torch.uint8, device=dev)
+    perm64 = torch.randperm(n, device=dev)
+    perm32 = perm64.to(torch.int32)
+    out = torch.empty_like(src)
+
+    ms = timeit_gpu(lambda: gather_rows(src, perm64, out=out))
+    print(
+        f"gather_rows i64 perm : {ms:7.3f} ms  "
+        f"{2 * n * stride / ms / 1e6:7.0f} GB/s"
+    )
+    ms = timeit_gpu(lambda: gather_rows(src, perm32, out=out))
+    print(
+        f"gather_rows i32 perm : {ms:7.3f} ms  "
+        f"{2 * n * stride / ms / 1e6:7.0f} GB/s"
+    )
+    ms = timeit_gpu(lambda: out.copy_(src))
+    print(
+        f"plain copy (roofline): {ms:7.3f} ms  "
+        f"{2 * n * stride / ms / 1e6:7.0f} GB/s"
+    )
+
+    # Fused unpack+permute into contiguous feature matrix + labels.
+    schema = Schema(
+        [
+            ColumnSpec("features", torch.float32, 100),
+            ColumnSpec("labels", torch.float32, 1),
+        ]
+    )
+    packed = src  # reinterpret (bytes are random, fine for perf)
+    ms = timeit_gpu(
+        lambda: unpack_permute(packed, schema, perm=perm64), iters=10
+    )
+    bytes_moved = n * (schema.payload_bytes + stride)  # read row, write cols
+    print(
+        f"unpack_permute fused : {ms:7.3f} ms  "
+        f"{bytes_moved / ms / 1e6:7.0f} GB/s"
+    )
+
+    # partition_rows (argsort+gather) for 8 destinations
+    dest = torch.randint(0, 8, (n,), device=dev)
+    ms = timeit_gpu(lambda: partition_rows(src, dest, 8), iters=10)
+    print(f"partition_rows 8-way : {ms:7.3f} ms")
+
+    # GEMM input layout A/B: [250k,100] fp32 strided (lda=104) vs contiguous
+    b = 250_000
+    lin = torch.nn.Linear(100, 512).to(dev)
+    xs_raw = torch.randn(b, 104, device=dev)
+    x_strided = xs_raw[:, :100]
+    x_contig = x_strided.contiguous()
+    ms = timeit_gpu(lambda: lin(x_strided), iters=30)
+    print(f"Linear(100->512) strided A : {ms:7.3f} ms")
+    ms = timeit_gpu(lambda: lin(x_contig), iters=30)
+    print(f"Linear(100->512) contig  A : {ms:7.3f} ms")
+
+    # Full fwd+bwd step A/B
+    from ray_shuffling_data_loader_amd.models.mlp import TabularMLP
+
+    model = TabularMLP(100).to(dev)
+    opt = torch.optim.SGD(model.parameters(), lr=1e-3, momentum=0.9)
+    y = torch.randn(b, 1, device=dev)
+
+    def step(x):
+        opt.zero_grad(set_to_none=True)
+        loss = torch.nn.functional.mse_loss(model(x), y)
+        loss.backward()
+        opt.step()
+
+    ms = timeit_gpu(lambda: step(x_strided), iters=20)
+    print(f"MLP step strided input     : {ms:7.3f} ms")
+    ms = timeit_gpu(lambda: step(x_contig), iters=20)
+    print(f"MLP step contig input      : {ms:7.3f} ms")
+
+
+if __name__ == "__main__":
+    main()
