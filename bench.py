@@ -50,6 +50,14 @@ def parse_args():
         "--device", type=str, default=None, help="override (cpu for debug)"
     )
     p.add_argument(
+        "--dtype",
+        type=str,
+        default="bf16",
+        choices=["bf16", "fp32"],
+        help="train-step compute dtype; bf16 uses MFMA via autocast and the "
+        "loader's fused fp32->bf16 cast kernel for features",
+    )
+    p.add_argument(
         "--loader-output",
         type=str,
         default="auto",
@@ -135,6 +143,7 @@ def main():
     num_epochs = (total_steps + steps_per_epoch - 1) // steps_per_epoch + 1
 
     feature_columns = [f"f{i}" for i in range(args.num_cols)]
+    use_bf16 = args.dtype == "bf16"
     ds = TorchShufflingDataset(
         filenames,
         num_epochs,
@@ -145,6 +154,9 @@ def main():
         num_reducers=args.reducers_per_gpu * world,
         max_concurrent_epochs=args.max_concurrent_epochs,
         feature_columns=feature_columns,
+        feature_types=(
+            [torch.bfloat16] * args.num_cols if use_bf16 else None
+        ),
         label_column="labels",
         feature_matrix=True,
         device=device,
@@ -175,6 +187,15 @@ def main():
         if world > 1:
             torch.distributed.barrier()
 
+    import contextlib
+
+    def amp():
+        if use_bf16:
+            return torch.autocast(
+                device_type=device.type, dtype=torch.bfloat16
+            )
+        return contextlib.nullcontext()
+
     def one_step():
         t_wait0 = time.perf_counter()
         data, target = next(it)
@@ -184,8 +205,9 @@ def main():
             x = x.to(device, non_blocking=True)
             target = target.to(device, non_blocking=True)
         opt.zero_grad(set_to_none=True)
-        out = model(x)
-        loss = loss_fn(out, target)
+        with amp():
+            out = model(x)
+            loss = loss_fn(out.float(), target)
         loss.backward()
         opt.step()
         return wait
@@ -237,7 +259,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "fp32",
+            "dtype": args.dtype,
             "data": (
                 "synthetic parquet, generated locally; "
                 f"cached {'in HBM' if args.source_cache in ('auto', 'device') else args.source_cache} "

@@ -74,6 +74,7 @@ class ShuffleEngine:
         schema: Optional[Schema] = None,
         feature_matrix: Optional[Tuple[str, List[str]]] = None,
         output: str = "auto",  # "auto" | "views" | "columns"
+        out_dtypes: Optional[Dict[str, torch.dtype]] = None,
         stats_collector=None,
     ):
         self.filenames = list(filenames)
@@ -85,6 +86,7 @@ class ShuffleEngine:
         self.stats = stats_collector
         self.reader_threads = reader_threads
         self.output = output
+        self.out_dtypes = out_dtypes or {}
         self.source_cache = source_cache
 
         world, dist_rank, initialized = fabric.dist_info()
@@ -219,8 +221,11 @@ class ShuffleEngine:
         selected rows into a RowBlock."""
         schema = self.out_schema
         hom = homogeneous_dtype(schema)
-        use_views = self.output == "views" or (
-            self.output == "auto" and hom is not None
+        # Output dtype casts (e.g. fp32 -> bf16 features) happen inside the
+        # fused unpack kernel, which requires the columns path.
+        use_views = not self.out_dtypes and (
+            self.output == "views"
+            or (self.output == "auto" and hom is not None)
         )
         if use_views and hom is not None:
             esz = dtype_bytes(hom)
@@ -234,7 +239,9 @@ class ShuffleEngine:
                 else:
                     cols[spec.name] = typed[:, o : o + spec.numel]
             return RowBlock(cols)
-        cols = unpack_permute(rows, schema, perm=idx.to(torch.long))
+        cols = unpack_permute(
+            rows, schema, perm=idx.to(torch.long), out_dtypes=self.out_dtypes
+        )
         return RowBlock(cols)
 
     def _shuffle_epoch(self, epoch: int) -> None:

@@ -58,6 +58,14 @@ class TorchShufflingDataset(IterableDataset):
             engine_kwargs.setdefault(
                 "feature_matrix", ("__features__", list(feature_columns))
             )
+            # A requested non-fp32 feature dtype (e.g. bf16 for MFMA compute)
+            # is produced by the fused unpack+cast kernel, not per batch.
+            if feature_types and isinstance(feature_types, list):
+                ft = feature_types[0]
+                if isinstance(ft, torch.dtype) and ft != torch.float32:
+                    engine_kwargs.setdefault(
+                        "out_dtypes", {"__features__": ft}
+                    )
         self._ds = ShufflingDataset(
             filenames,
             num_epochs,

@@ -4,7 +4,10 @@
 Run on an MI355X box:  python tools/microbench.py
 Writes a summary to stdout (capture into gpurun_out/)."""
 
-import time
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
