@@ -67,6 +67,13 @@ def get_shuffle_group():
         _shuffle_group = dist.new_group(
             ranks=list(range(dist.get_world_size()))
         )
+        # Eagerly initialize the communicator HERE on the main thread, so
+        # its (collective) init can never race the default communicator's
+        # first DDP use from another thread.
+        probe = torch.zeros(1)
+        if dist.get_backend(_shuffle_group) == "nccl":
+            probe = probe.cuda()
+        dist.all_reduce(probe, group=_shuffle_group)
     return _shuffle_group
 
 
