@@ -135,6 +135,53 @@ __device__ __forceinline__ double cast_elem(__hip_bfloat16 v) {
   return (double)__bfloat162float(v);
 }
 
+// Vectorized unpack specializations for the hot flagship combos: the
+// feature-matrix column is (f32 -> f32) or (f32 -> bf16) with numel % 4 == 0
+// and a 16-B-aligned packed offset, so each lane moves one float4 (16 B
+// load) instead of four scalar dwords. ~2x on the fused unpack
+// (profiles/microbench: scalar 2.6 TB/s vs gather roofline 4.8 TB/s).
+__device__ __forceinline__ void unpack_vec4_f32_f32(
+    const uint8_t* __restrict__ packed, int64_t row_stride,
+    const int64_t* __restrict__ perm, float* __restrict__ dst,
+    int32_t packed_off, int32_t numel, int64_t n_rows, int64_t tid,
+    int64_t nthreads) {
+  const int32_t groups = numel >> 2;
+  const int64_t total = n_rows * groups;
+  for (int64_t i = tid; i < total; i += nthreads) {
+    const int64_t row = i / groups;
+    const int64_t g = i - row * groups;
+    const int64_t srow = perm ? perm[row] : row;
+    const float4 v = *reinterpret_cast<const float4*>(
+        packed + srow * row_stride + packed_off + (g << 4));
+    *reinterpret_cast<float4*>(dst + row * numel + (g << 2)) = v;
+  }
+}
+
+__device__ __forceinline__ void unpack_vec4_f32_bf16(
+    const uint8_t* __restrict__ packed, int64_t row_stride,
+    const int64_t* __restrict__ perm, __hip_bfloat16* __restrict__ dst,
+    int32_t packed_off, int32_t numel, int64_t n_rows, int64_t tid,
+    int64_t nthreads) {
+  const int32_t groups = numel >> 2;
+  const int64_t total = n_rows * groups;
+  for (int64_t i = tid; i < total; i += nthreads) {
+    const int64_t row = i / groups;
+    const int64_t g = i - row * groups;
+    const int64_t srow = perm ? perm[row] : row;
+    const float4 v = *reinterpret_cast<const float4*>(
+        packed + srow * row_stride + packed_off + (g << 4));
+    union {
+      __hip_bfloat16 h[4];
+      uint2 u;
+    } o;
+    o.h[0] = __float2bfloat16(v.x);
+    o.h[1] = __float2bfloat16(v.y);
+    o.h[2] = __float2bfloat16(v.z);
+    o.h[3] = __float2bfloat16(v.w);
+    *reinterpret_cast<uint2*>(dst + row * numel + (g << 2)) = o.u;
+  }
+}
+
 // Unpack direction: column[i*numel + e] = cast(packed[perm[i]*stride + off +
 // e*sizeof(S)]). Writes coalesced; reads random-row.
 template <typename S, typename D>
@@ -221,6 +268,22 @@ __global__ void unpack_permute_kernel(
   const ColDesc d = table.cols[c];
   const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int64_t nthreads = (int64_t)gridDim.x * blockDim.x;
+  // Hot-path vector specializations (flagship feature matrix).
+  if (d.src_dtype == DT_F32 && (d.numel & 3) == 0 &&
+      (d.packed_off & 15) == 0) {
+    if (d.dst_dtype == DT_F32) {
+      unpack_vec4_f32_f32(packed, row_stride, perm,
+                          reinterpret_cast<float*>(d.col_ptr), d.packed_off,
+                          d.numel, n_rows, tid, nthreads);
+      return;
+    }
+    if (d.dst_dtype == DT_BF16) {
+      unpack_vec4_f32_bf16(packed, row_stride, perm,
+                           reinterpret_cast<__hip_bfloat16*>(d.col_ptr),
+                           d.packed_off, d.numel, n_rows, tid, nthreads);
+      return;
+    }
+  }
   RSDL_DISPATCH_PAIR(
       d.src_dtype, d.dst_dtype,
       (unpack_col_loop<S, D>(packed, row_stride, perm,

@@ -78,6 +78,20 @@ def main():
         f"unpack_permute fused : {ms:7.3f} ms  "
         f"{bytes_moved / ms / 1e6:7.0f} GB/s"
     )
+    ms = timeit_gpu(
+        lambda: unpack_permute(
+            packed,
+            schema,
+            perm=perm64,
+            out_dtypes={"features": torch.bfloat16},
+        ),
+        iters=10,
+    )
+    bytes_bf16 = n * (stride + 100 * 2 + 4)
+    print(
+        f"unpack_permute ->bf16: {ms:7.3f} ms  "
+        f"{bytes_bf16 / ms / 1e6:7.0f} GB/s"
+    )
 
     # partition_rows (argsort+gather) for 8 destinations
     dest = torch.randint(0, 8, (n,), device=dev)
