@@ -22,9 +22,27 @@ Launch (driver contract):
 import argparse
 import json
 import os
+import shutil
 import statistics
 import tempfile
 import time
+
+# hipBLASLt algorithm selection: load the pre-tuned gfx950 GEMM table
+# (profiles/tunableop_gfx950.csv — picks split-K kernels for the huge-K
+# wgrad shapes; ~20% step time) and keep tuning enabled for any shape not
+# in it (tuning runs during warmup). Must be set before torch loads blaslt.
+_repo = os.path.dirname(os.path.abspath(__file__))
+_tuned_src = os.path.join(_repo, "profiles", "tunableop_gfx950.csv")
+if os.environ.get("RSDL_TUNABLEOP", "1") == "1":
+    _tuned_dst = os.path.join(
+        tempfile.gettempdir(), f"rsdl_tunableop_{os.getpid()}.csv"
+    )
+    if os.path.exists(_tuned_src):
+        shutil.copyfile(_tuned_src, _tuned_dst)
+    os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", _tuned_dst)
+    os.environ.setdefault("PYTORCH_TUNABLEOP_VERBOSE", "0")
 
 import torch
 

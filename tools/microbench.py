@@ -127,6 +127,31 @@ def main():
     ms = timeit_gpu(lambda: step(x_contig), iters=20)
     print(f"MLP step contig input      : {ms:7.3f} ms")
 
+    x_bf16 = x_contig.to(torch.bfloat16)
+
+    def step_amp(x):
+        opt.zero_grad(set_to_none=True)
+        with torch.autocast("cuda", torch.bfloat16):
+            out = model(x)
+            loss = torch.nn.functional.mse_loss(out.float(), y)
+        loss.backward()
+        opt.step()
+
+    ms = timeit_gpu(lambda: step_amp(x_bf16), iters=20)
+    print(f"MLP step bf16 autocast     : {ms:7.3f} ms")
+
+    model_bf = TabularMLP(100).to(dev).to(torch.bfloat16)
+    opt_bf = torch.optim.SGD(model_bf.parameters(), lr=1e-3, momentum=0.9)
+
+    def step_full_bf16(x):
+        opt_bf.zero_grad(set_to_none=True)
+        loss = torch.nn.functional.mse_loss(model_bf(x).float(), y)
+        loss.backward()
+        opt_bf.step()
+
+    ms = timeit_gpu(lambda: step_full_bf16(x_bf16), iters=20)
+    print(f"MLP step pure bf16 model   : {ms:7.3f} ms")
+
 
 if __name__ == "__main__":
     main()
