@@ -26,7 +26,10 @@ import threading
 from typing import List, Optional
 
 from ray_shuffling_data_loader_amd.batch_queue import BatchQueue
-from ray_shuffling_data_loader_amd.engine import ShuffleEngine
+from ray_shuffling_data_loader_amd.engine import (
+    ShuffleEngine,
+    ShuffleEngineFailure,
+)
 from ray_shuffling_data_loader_amd.shuffle import BatchConsumer
 from ray_shuffling_data_loader_amd.utils.rowblock import RowBlock
 from ray_shuffling_data_loader_amd.parallel import fabric
@@ -202,6 +205,10 @@ class ShufflingDataset:
             num_outstanding = len(pending)
 
             for block in pending:
+                if isinstance(block, ShuffleEngineFailure):
+                    raise RuntimeError(
+                        "shuffle engine worker failed"
+                    ) from block.error
                 if len(block) == 0:
                     continue
                 buffer_len = len(buffer) if buffer is not None else 0

@@ -57,6 +57,18 @@ from ray_shuffling_data_loader_amd.utils.schema import (
 )
 
 
+class ShuffleEngineFailure:
+    """Marker pushed into the batch queues when the engine worker dies, so
+    blocked consumers raise instead of hanging (the reference had no analog:
+    a dead Ray shuffle task surfaced only at the final ray.get,
+    dataset.py:186-188, while iterating trainers hung)."""
+
+    __slots__ = ("error",)
+
+    def __init__(self, error: BaseException):
+        self.error = error
+
+
 class ShuffleEngine:
     def __init__(
         self,
@@ -342,8 +354,16 @@ class ShuffleEngine:
         def _target():
             try:
                 self.run()
-            except BaseException as e:  # surfaced via join()
+            except BaseException as e:  # surfaced via join() AND the queue
                 self._error = e
+                failure = ShuffleEngineFailure(e)
+                try:
+                    for epoch in range(self.num_epochs):
+                        for t in self.owned_trainers:
+                            self.consumer.consume(t, epoch, [failure])
+                            self.consumer.producer_done(t, epoch)
+                except BaseException:
+                    pass  # queue may already be shut down
 
         self._thread = threading.Thread(
             target=_target, name=f"rsdl-shuffle-r{self.rank}", daemon=True
