@@ -340,7 +340,14 @@ class ShuffleEngine:
         )
         with ctx:
             for epoch in range(self.num_epochs):
+                t_gate = time.perf_counter()
                 self.consumer.wait_until_ready(epoch)
+                if self.stats:
+                    # Epoch-window throttle time (reference shuffle.py:74 +
+                    # stats.py:166-168 throttle_done).
+                    self.stats.epoch_throttle_done(
+                        epoch, time.perf_counter() - t_gate
+                    )
                 self._shuffle_epoch(epoch)
         self.consumer.wait_until_all_epochs_done()
         self.duration = time.perf_counter() - start

@@ -47,8 +47,10 @@ def fuse_schema(base: Schema, feature_matrix: Optional[Tuple[str, List[str]]]
     if len(dts) != 1:
         raise ValueError("feature-matrix columns must share one dtype")
     dt = dts.pop()
+    from ray_shuffling_data_loader_amd.utils.schema import dtype_bytes
+
     offs = sorted(base.offsets[m] for m in members)
-    esz = offs[1] - offs[0] if len(offs) > 1 else 0
+    esz = dtype_bytes(dt)
     for a, b in zip(offs, offs[1:]):
         if b - a != esz:
             raise ValueError(
