@@ -311,8 +311,144 @@ __global__ void pack_columns_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Radix partition by destination id (replaces torch argsort+bincount on the
+// map side; reference shuffle.py:156-161 boolean-mask loop). Three phases:
+//   1. per-block LDS histogram of dest ids -> block_counts[B][T]
+//   2. exclusive scan on host side (tiny [B,T] tensor, torch cumsum)
+//   3. rank+scatter: per-block LDS running counters assign each row its
+//      output slot; writes the GATHER permutation (perm[dst] = src row),
+//      which then drives the roofline-speed gather_rows kernel.
+// Within-destination order is block-local-nondeterministic, which is fine:
+// a full random permutation is applied downstream either way.
+// ---------------------------------------------------------------------------
+
+#define RSDL_PART_MAX_DESTS 1024
+
+__global__ void partition_hist_kernel(
+    const int32_t* __restrict__ dest,
+    int32_t* __restrict__ block_counts,  // [gridDim.x][num_dests]
+    int64_t n,
+    int32_t num_dests) {
+  extern __shared__ int32_t lds_counts[];
+  for (int32_t d = threadIdx.x; d < num_dests; d += blockDim.x)
+    lds_counts[d] = 0;
+  __syncthreads();
+  const int64_t chunk = (n + gridDim.x - 1) / gridDim.x;
+  const int64_t lo = (int64_t)blockIdx.x * chunk;
+  const int64_t hi = min(lo + chunk, n);
+  for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
+    atomicAdd(&lds_counts[dest[i]], 1);
+  __syncthreads();
+  for (int32_t d = threadIdx.x; d < num_dests; d += blockDim.x)
+    block_counts[(int64_t)blockIdx.x * num_dests + d] = lds_counts[d];
+}
+
+__global__ void partition_scatter_kernel(
+    const int32_t* __restrict__ dest,
+    const int32_t* __restrict__ block_base,  // [gridDim.x][num_dests]
+    int32_t* __restrict__ perm_out,          // [n] gather indices
+    int64_t n,
+    int32_t num_dests) {
+  extern __shared__ int32_t lds_next[];
+  for (int32_t d = threadIdx.x; d < num_dests; d += blockDim.x)
+    lds_next[d] = block_base[(int64_t)blockIdx.x * num_dests + d];
+  __syncthreads();
+  const int64_t chunk = (n + gridDim.x - 1) / gridDim.x;
+  const int64_t lo = (int64_t)blockIdx.x * chunk;
+  const int64_t hi = min(lo + chunk, n);
+  for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+    const int32_t pos = atomicAdd(&lds_next[dest[i]], 1);
+    perm_out[pos] = (int32_t)i;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Tiled pack for the map side: C scalar/vector columns -> packed rows.
+// The naive per-element kernel stores 4-8 B at row_stride intervals
+// (~0.5 TB/s); this one stages a TILE_R-row tile in LDS (per-column
+// coalesced loads, conflict-spread by the padded row stride) and then
+// streams the tile out as one contiguous row-major block (16-B lanes,
+// perfectly coalesced).
+// ---------------------------------------------------------------------------
+
+__global__ void pack_tiled_kernel(
+    uint8_t* __restrict__ packed,
+    int64_t row_stride,
+    ColTable table,
+    int32_t num_cols,
+    int64_t n_rows,
+    int32_t tile_rows) {
+  extern __shared__ uint8_t lds_tile[];  // [tile_rows][lds_stride]
+  // +8 B pad keeps 8-B column alignment inside LDS and makes the dword
+  // stride ≡ 2 (mod 4): per-column lane strides cover 16 of 32 banks
+  // (gcd=2 -> at worst 2-way, which is free for ds_write_b32).
+  const int64_t lds_stride = row_stride + 8;
+  const int64_t row0 = (int64_t)blockIdx.x * tile_rows;
+  const int32_t rows_here =
+      (int32_t)min((int64_t)tile_rows, n_rows - row0);
+  if (rows_here <= 0) return;
+
+  // Phase 1: per-column coalesced loads -> LDS rows (cast on the fly).
+  for (int32_t c = 0; c < num_cols; ++c) {
+    const ColDesc d = table.cols[c];
+    const int64_t elems = (int64_t)rows_here * d.numel;
+    for (int64_t t = threadIdx.x; t < elems; t += blockDim.x) {
+      const int32_t r = (int32_t)(t / d.numel);
+      const int32_t e = (int32_t)(t - (int64_t)r * d.numel);
+      uint8_t* dst_base = lds_tile + (int64_t)r * lds_stride + d.packed_off;
+      RSDL_DISPATCH_PAIR(
+          d.dst_dtype, d.src_dtype,
+          (reinterpret_cast<D*>(dst_base)[e] = cast_elem<S, D>(
+               reinterpret_cast<const S*>(d.col_ptr)[row0 * d.numel + t])));
+    }
+  }
+  __syncthreads();
+
+  // Phase 2: stream the tile to global as contiguous row-major dwords
+  // (consecutive lanes -> consecutive banks and fully coalesced stores;
+  // the padded LDS stride rules out 16-B-aligned reads).
+  const int64_t row_dw = row_stride >> 2;
+  const int64_t lds_stride_dw = lds_stride >> 2;
+  const int64_t total_dw = (int64_t)rows_here * row_dw;
+  uint32_t* gout = reinterpret_cast<uint32_t*>(packed + row0 * row_stride);
+  const uint32_t* lin = reinterpret_cast<const uint32_t*>(lds_tile);
+  for (int64_t t = threadIdx.x; t < total_dw; t += blockDim.x) {
+    const int64_t r = t / row_dw;
+    const int64_t o = t - r * row_dw;
+    gout[t] = lin[r * lds_stride_dw + o];
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Host-side launchers (called from shuffle_ops.cpp).
 // ---------------------------------------------------------------------------
+
+void launch_partition_hist(
+    const int32_t* dest, int32_t* block_counts, int64_t n,
+    int32_t num_dests, int32_t num_blocks, hipStream_t stream) {
+  hipLaunchKernelGGL(partition_hist_kernel, dim3(num_blocks), dim3(256),
+                     num_dests * sizeof(int32_t), stream, dest, block_counts,
+                     n, num_dests);
+}
+
+void launch_partition_scatter(
+    const int32_t* dest, const int32_t* block_base, int32_t* perm_out,
+    int64_t n, int32_t num_dests, int32_t num_blocks, hipStream_t stream) {
+  hipLaunchKernelGGL(partition_scatter_kernel, dim3(num_blocks), dim3(256),
+                     num_dests * sizeof(int32_t), stream, dest, block_base,
+                     perm_out, n, num_dests);
+}
+
+void launch_pack_tiled(
+    void* packed, int64_t row_stride, const ColTable& table,
+    int32_t num_cols, int64_t n_rows, int32_t tile_rows, int64_t lds_bytes,
+    hipStream_t stream) {
+  const int64_t blocks = (n_rows + tile_rows - 1) / tile_rows;
+  hipLaunchKernelGGL(pack_tiled_kernel, dim3((uint32_t)blocks), dim3(256),
+                     (uint32_t)lds_bytes, stream,
+                     reinterpret_cast<uint8_t*>(packed), row_stride, table,
+                     num_cols, n_rows, tile_rows);
+}
 
 void launch_gather_rows(
     const void* src, void* dst, const int64_t* perm, int64_t n_rows,

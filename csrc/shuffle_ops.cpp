@@ -50,6 +50,15 @@ void launch_unpack_permute(const void* packed, int64_t row_stride,
 void launch_pack_columns(void* packed, int64_t row_stride, const int64_t* perm,
                          const ColTable& table, int32_t num_cols,
                          int64_t n_rows, hipStream_t stream);
+void launch_partition_hist(const int32_t* dest, int32_t* block_counts,
+                           int64_t n, int32_t num_dests, int32_t num_blocks,
+                           hipStream_t stream);
+void launch_partition_scatter(const int32_t* dest, const int32_t* block_base,
+                              int32_t* perm_out, int64_t n, int32_t num_dests,
+                              int32_t num_blocks, hipStream_t stream);
+void launch_pack_tiled(void* packed, int64_t row_stride, const ColTable& table,
+                       int32_t num_cols, int64_t n_rows, int32_t tile_rows,
+                       int64_t lds_bytes, hipStream_t stream);
 
 namespace {
 
@@ -208,6 +217,82 @@ at::Tensor pack_columns(const std::vector<at::Tensor>& cols,
   return packed;
 }
 
+// Build the gather permutation that groups rows by destination id.
+// Returns (perm int32 [n], counts int64 [num_dests]).
+std::vector<at::Tensor> partition_build_perm(const at::Tensor& dest,
+                                             int64_t num_dests) {
+  TORCH_CHECK(dest.is_cuda() && dest.is_contiguous(),
+              "dest must be contiguous GPU tensor");
+  TORCH_CHECK(num_dests >= 1 && num_dests <= 1024,
+              "num_dests must be in [1, 1024]");
+  auto dest32 = dest.scalar_type() == at::kInt ? dest : dest.to(at::kInt);
+  int64_t n = dest32.numel();
+  int32_t num_blocks = 1024;
+  if (n < (int64_t)num_blocks * 256) {
+    num_blocks = (int32_t)std::max<int64_t>(1, (n + 255) / 256);
+  }
+  auto opts32 = dest32.options();
+  auto block_counts = at::empty({num_blocks, num_dests}, opts32);
+  auto stream = current_stream();
+  if (n > 0) {
+    launch_partition_hist(dest32.data_ptr<int32_t>(),
+                          block_counts.data_ptr<int32_t>(), n,
+                          (int32_t)num_dests, num_blocks, stream);
+  } else {
+    block_counts.zero_();
+  }
+  // Exclusive scan (dest-major, then block-major within a dest), done with
+  // torch ops on the same stream: tiny [B, T] tensor.
+  auto counts64 = block_counts.to(at::kLong);
+  auto per_dest = counts64.sum(0);                    // [T]
+  auto dest_base = per_dest.cumsum(0) - per_dest;     // exclusive [T]
+  auto within = counts64.cumsum(0) - counts64;        // exclusive over blocks
+  auto base = (within + dest_base.unsqueeze(0)).to(at::kInt).contiguous();
+  auto perm = at::empty({n}, opts32);
+  if (n > 0) {
+    launch_partition_scatter(dest32.data_ptr<int32_t>(),
+                             base.data_ptr<int32_t>(),
+                             perm.data_ptr<int32_t>(), n, (int32_t)num_dests,
+                             num_blocks, stream);
+  }
+  return {perm, per_dest};
+}
+
+// LDS-tiled pack (no scatter perm): cols -> packed rows.
+at::Tensor pack_columns_tiled(const std::vector<at::Tensor>& cols,
+                              const std::vector<int64_t>& packed_offsets,
+                              const std::vector<int64_t>& packed_dtype_codes,
+                              int64_t row_stride) {
+  TORCH_CHECK(!cols.empty() && cols.size() <= 128, "1..128 columns");
+  TORCH_CHECK(row_stride % 16 == 0, "row_stride must be multiple of 16");
+  int64_t n_rows = cols[0].size(0);
+  auto packed =
+      at::empty({n_rows, row_stride}, cols[0].options().dtype(at::kByte));
+  ColTable table{};
+  for (size_t c = 0; c < cols.size(); ++c) {
+    const auto& t = cols[c];
+    TORCH_CHECK(t.is_cuda() && t.is_contiguous(),
+                "input column must be contiguous GPU tensor");
+    TORCH_CHECK(t.size(0) == n_rows, "column rows mismatch");
+    table.cols[c] = ColDesc{
+        reinterpret_cast<int64_t>(t.data_ptr()),
+        (int32_t)packed_offsets[c],
+        (int32_t)packed_dtype_codes[c],
+        dtype_code(t.scalar_type()),
+        (int32_t)(t.numel() / n_rows),
+    };
+  }
+  const int64_t lds_stride = row_stride + 8;
+  int32_t tile_rows = (int32_t)std::min<int64_t>(128, 65536 / lds_stride);
+  TORCH_CHECK(tile_rows >= 1, "row_stride too large for tiled pack");
+  if (n_rows > 0) {
+    launch_pack_tiled(packed.data_ptr(), row_stride, table,
+                      (int32_t)cols.size(), n_rows, tile_rows,
+                      lds_stride * tile_rows, current_stream());
+  }
+  return packed;
+}
+
 }  // namespace
 }  // namespace rsdl
 
@@ -223,6 +308,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_columns", &rsdl::pack_columns, py::arg("cols"),
         py::arg("packed_offsets"), py::arg("packed_dtype_codes"),
         py::arg("row_stride"), py::arg("perm") = c10::nullopt);
+  m.def("pack_columns_tiled", &rsdl::pack_columns_tiled, py::arg("cols"),
+        py::arg("packed_offsets"), py::arg("packed_dtype_codes"),
+        py::arg("row_stride"));
+  m.def("partition_build_perm", &rsdl::partition_build_perm, py::arg("dest"),
+        py::arg("num_dests"));
   m.attr("DT_F32") = (int)rsdl::DT_F32;
   m.attr("DT_F64") = (int)rsdl::DT_F64;
   m.attr("DT_I32") = (int)rsdl::DT_I32;

@@ -131,6 +131,12 @@ def pack_columns(
             cols.append(t.contiguous())
             offs.append(schema.offsets[spec.name])
             codes.append(_dtype_code(spec.dtype))
+        if perm is None:
+            # LDS-tiled transpose pack: coalesced column reads + one
+            # contiguous row-major store stream.
+            return hip.pack_columns_tiled(
+                cols, offs, codes, schema.row_stride
+            )
         return hip.pack_columns(
             cols, offs, codes, schema.row_stride, perm
         )
@@ -210,8 +216,18 @@ def partition_rows(
     num_dests: int,
 ):
     """Group packed rows by destination id. Returns (regrouped rows, counts
-    per destination). Row order within a destination is the stable input
-    order (like the reference's boolean-mask partition)."""
+    per destination).
+
+    GPU: fused histogram -> scan -> rank/scatter builds the gather
+    permutation in two O(N) index passes (no radix sort); the row move is
+    the roofline gather. Order within a destination is block-local (a full
+    random permutation is applied downstream either way). CPU: stable
+    argsort (order matches the reference's boolean-mask partition)."""
+    if packed.is_cuda:
+        hip = _load_hip()
+        perm, counts = hip.partition_build_perm(dest, num_dests)
+        grouped = gather_rows(packed, perm)
+        return grouped, counts
     counts = torch.bincount(dest, minlength=num_dests)
     order = torch.argsort(dest, stable=True)
     grouped = gather_rows(packed, order)

@@ -93,10 +93,31 @@ def main():
         f"{bytes_bf16 / ms / 1e6:7.0f} GB/s"
     )
 
-    # partition_rows (argsort+gather) for 8 destinations
-    dest = torch.randint(0, 8, (n,), device=dev)
+    # partition_rows (fused hist+scan+scatter+gather) for 8 destinations
+    dest = torch.randint(0, 8, (n,), device=dev, dtype=torch.int32)
     ms = timeit_gpu(lambda: partition_rows(src, dest, 8), iters=10)
     print(f"partition_rows 8-way : {ms:7.3f} ms")
+
+    def part_sort():
+        order = torch.argsort(dest.long(), stable=True)
+        return gather_rows(src, order)
+
+    ms = timeit_gpu(part_sort, iters=10)
+    print(f"partition via argsort: {ms:7.3f} ms")
+
+    # Tiled pack: 101 fp32 scalar columns -> packed rows (map side).
+    from ray_shuffling_data_loader_amd.utils.schema import Schema as Sch
+
+    pcols = {f"c{i}": torch.randn(n, device=dev) for i in range(101)}
+    psch = Sch(
+        [ColumnSpec(f"c{i}", torch.float32, 1) for i in range(101)]
+    )
+    ms = timeit_gpu(lambda: pack_columns(pcols, psch), iters=10)
+    pbytes = n * (101 * 4 + psch.row_stride)
+    print(
+        f"pack_columns tiled   : {ms:7.3f} ms  "
+        f"{pbytes / ms / 1e6:7.0f} GB/s"
+    )
 
     # GEMM input layout A/B: [250k,100] fp32 strided (lda=104) vs contiguous
     b = 250_000
