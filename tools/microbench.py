@@ -119,6 +119,9 @@ def main():
         f"{pbytes / ms / 1e6:7.0f} GB/s"
     )
 
+    if os.environ.get("RSDL_MB_KERNELS_ONLY") == "1":
+        return
+
     # GEMM input layout A/B: [250k,100] fp32 strided (lda=104) vs contiguous
     b = 250_000
     lin = torch.nn.Linear(100, 512).to(dev)
