@@ -111,7 +111,14 @@ def read_files_packed(
             for name, arr in cols_np.items():
                 t = torch.from_numpy(arr)
                 if use_gpu:
-                    t = t.to(device, non_blocking=True)
+                    # Stage through a pinned-host arena so the H2D copy is
+                    # a true async DMA overlapping the next file's Arrow
+                    # read and the pack kernel (torch's caching host
+                    # allocator keeps the pinned block alive until the
+                    # copy's stream work completes).
+                    pin = torch.empty_like(t, pin_memory=True)
+                    pin.copy_(t)
+                    t = pin.to(device, non_blocking=True)
                 cols[name] = t
             packed_parts.append(pack_columns(cols, schema))
     if len(packed_parts) == 1:
