@@ -22,7 +22,7 @@ dataset.py:52-84):
 """
 
 import os
-import threading
+
 from typing import List, Optional
 
 from ray_shuffling_data_loader_amd.batch_queue import BatchQueue

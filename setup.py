@@ -11,7 +11,7 @@ Usage:  python setup.py build_ext --inplace
 """
 
 import os
-import sys
+
 
 from setuptools import setup
 

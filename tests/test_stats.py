@@ -3,7 +3,7 @@
 
 import csv
 import os
-import threading
+
 import time
 
 import pytest
