@@ -58,6 +58,10 @@ def parse_args():
     p.add_argument("--mock-train-step-time", type=float, default=None,
                    help="replace the real fwd/bwd with sleep(T) "
                    "(the reference example's methodology)")
+    p.add_argument("--spec", choices=["float", "tabular"], default="float",
+                   help="float: 100 fp32 cols + TabularMLP; tabular: the "
+                   "reference DATA_SPEC (19 int64 categoricals) + "
+                   "EmbeddingMLP")
     p.add_argument("--data-dir", type=str, default=None)
     p.add_argument("--cache-files", action="store_true")
     p.add_argument("--lr", type=float, default=1e-3)
@@ -74,10 +78,14 @@ def train_main(args, world, rank):
     else:
         device = torch.device("cpu")
 
+    from ray_shuffling_data_loader_amd.data_generation import DATA_SPEC
+    from ray_shuffling_data_loader_amd.models.mlp import EmbeddingMLP
+
+    tabular = args.spec == "tabular"
     data_dir = args.data_dir or os.path.join(
-        tempfile.gettempdir(), "rsdl_ddp_example_data"
+        tempfile.gettempdir(), f"rsdl_ddp_example_data_{args.spec}"
     )
-    spec = float_data_spec(args.num_cols)
+    spec = DATA_SPEC if tabular else float_data_spec(args.num_cols)
     if rank == 0 and (
         not args.cache_files
         or not os.path.isdir(data_dir)
@@ -91,7 +99,7 @@ def train_main(args, world, rank):
             0.0,
             data_dir,
             spec=spec,
-            include_key=False,
+            include_key=not tabular,
         )
     if world > 1:
         torch.distributed.barrier()
@@ -99,22 +107,41 @@ def train_main(args, world, rank):
         os.path.join(data_dir, f) for f in os.listdir(data_dir)
     )
 
-    feature_columns = [f"f{i}" for i in range(args.num_cols)]
-    ds = TorchShufflingDataset(
-        filenames,
-        args.num_epochs,
-        num_trainers=world,
-        batch_size=args.batch_size,
-        rank=rank,
-        num_reducers=args.num_reducers,
-        max_concurrent_epochs=args.max_concurrent_epochs,
-        feature_columns=feature_columns,
-        label_column="labels",
-        feature_matrix=True,
-        device=device,
-    )
-
-    model = TabularMLP(args.num_cols).to(device)
+    if tabular:
+        # Reference example shape: 19 int64 categorical columns + fp64
+        # label (ray_torch_shuffle.py:256-278 over DATA_SPEC).
+        feature_columns = [c for c in DATA_SPEC if c != "labels"]
+        cardinalities = [DATA_SPEC[c][1] for c in feature_columns]
+        ds = TorchShufflingDataset(
+            filenames,
+            args.num_epochs,
+            num_trainers=world,
+            batch_size=args.batch_size,
+            rank=rank,
+            num_reducers=args.num_reducers,
+            max_concurrent_epochs=args.max_concurrent_epochs,
+            feature_columns=feature_columns,
+            feature_types=[torch.int64] * len(feature_columns),
+            label_column="labels",
+            device=device,
+        )
+        model = EmbeddingMLP(cardinalities).to(device)
+    else:
+        feature_columns = [f"f{i}" for i in range(args.num_cols)]
+        ds = TorchShufflingDataset(
+            filenames,
+            args.num_epochs,
+            num_trainers=world,
+            batch_size=args.batch_size,
+            rank=rank,
+            num_reducers=args.num_reducers,
+            max_concurrent_epochs=args.max_concurrent_epochs,
+            feature_columns=feature_columns,
+            label_column="labels",
+            feature_matrix=True,
+            device=device,
+        )
+        model = TabularMLP(args.num_cols).to(device)
     if world > 1:
         model = torch.nn.parallel.DistributedDataParallel(model)
     opt = torch.optim.SGD(model.parameters(), lr=args.lr, momentum=0.9)
@@ -131,7 +158,7 @@ def train_main(args, world, rank):
             if args.mock_train_step_time is not None:
                 time.sleep(args.mock_train_step_time)
             else:
-                x = data[0]
+                x = data if tabular else data[0]
                 opt.zero_grad(set_to_none=True)
                 loss = loss_fn(model(x), target)
                 loss.backward()
