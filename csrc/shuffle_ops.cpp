@@ -58,7 +58,7 @@ void launch_partition_scatter(const int32_t* dest, const int32_t* block_base,
                               int32_t num_blocks, hipStream_t stream);
 void launch_pack_tiled(void* packed, int64_t row_stride, const ColTable& table,
                        int32_t num_cols, int64_t n_rows, int32_t tile_rows,
-                       int64_t lds_bytes, hipStream_t stream);
+                       int64_t lds_bytes, int32_t n8, hipStream_t stream);
 
 namespace {
 
@@ -285,10 +285,29 @@ at::Tensor pack_columns_tiled(const std::vector<at::Tensor>& cols,
   const int64_t lds_stride = row_stride + 8;
   int32_t tile_rows = (int32_t)std::min<int64_t>(128, 65536 / lds_stride);
   TORCH_CHECK(tile_rows >= 1, "row_stride too large for tiled pack");
+  // Fast-path eligibility: scalar columns, no casts, 8-B columns first
+  // (guaranteed by Schema's descending-size packing) then 4-B columns.
+  int32_t n8 = 0;
+  bool fast = true;
+  bool in8 = true;
+  for (size_t c = 0; c < cols.size(); ++c) {
+    const ColDesc& d = table.cols[c];
+    if (d.numel != 1 || d.src_dtype != d.dst_dtype) { fast = false; break; }
+    int esz = (d.src_dtype == DT_F64 || d.src_dtype == DT_I64) ? 8
+              : (d.src_dtype == DT_F32 || d.src_dtype == DT_I32) ? 4 : 0;
+    if (esz == 0) { fast = false; break; }
+    if (esz == 8) {
+      if (!in8) { fast = false; break; }
+      n8++;
+    } else {
+      in8 = false;
+    }
+  }
   if (n_rows > 0) {
     launch_pack_tiled(packed.data_ptr(), row_stride, table,
                       (int32_t)cols.size(), n_rows, tile_rows,
-                      lds_stride * tile_rows, current_stream());
+                      lds_stride * tile_rows, fast ? n8 : -1,
+                      current_stream());
   }
   return packed;
 }
