@@ -249,3 +249,47 @@ def test_pack_tiled_odd_tail(dev):
         packed = pack_columns(cols, schema)
         out = unpack_permute(packed, schema)
         assert torch.equal(out["x"].cpu(), cols["x"].cpu()), n
+
+
+def test_chunked_linear_grads_match(dev):
+    from ray_shuffling_data_loader_amd.models.mlp import ChunkedLinear
+
+    torch.manual_seed(0)
+    m, k, n = 4096, 100, 512
+    x = torch.randn(m, k, device=dev, requires_grad=True)
+    ref = torch.nn.Linear(k, n).to(dev)
+    chk = ChunkedLinear(k, n).to(dev)
+    chk.load_state_dict(ref.state_dict())
+
+    y_ref = ref(x)
+    g = torch.randn_like(y_ref)
+    y_ref.backward(g)
+    gx_ref, gw_ref, gb_ref = (
+        x.grad.clone(),
+        ref.weight.grad.clone(),
+        ref.bias.grad.clone(),
+    )
+    x.grad = None
+    y = chk(x)
+    assert torch.equal(y, y_ref)
+    y.backward(g)
+    assert torch.allclose(x.grad, gx_ref, atol=1e-5)
+    assert torch.allclose(chk.weight.grad, gw_ref, atol=1e-3, rtol=1e-4)
+    assert torch.allclose(chk.bias.grad, gb_ref, atol=1e-4)
+
+
+def test_chunked_linear_autocast_step(dev):
+    from ray_shuffling_data_loader_amd.models.mlp import TabularMLP
+
+    model = TabularMLP(100).to(dev)
+    opt = torch.optim.SGD(model.parameters(), lr=1e-3)
+    x = torch.randn(8192, 100, device=dev, dtype=torch.bfloat16)
+    y = torch.randn(8192, 1, device=dev)
+    with torch.autocast("cuda", torch.bfloat16):
+        loss = torch.nn.functional.mse_loss(model(x).float(), y)
+    loss.backward()
+    opt.step()
+    assert all(
+        p.grad is not None and torch.isfinite(p.grad).all()
+        for p in model.parameters()
+    )

@@ -60,5 +60,58 @@ def main():
     rep("fwd L1 NT b=62.5k", t(lambda: xs @ W1.T), f1 / 4, b1 / 4)
 
 
-if __name__ == "__main__":
+
+
+def wgrad_variants():
+    dev = "cuda"
+    M, N, K = 250_000, 512, 100
+    dt = torch.bfloat16
+    dy = torch.randn(M, N, device=dev, dtype=dt)
+    x = torch.randn(M, K, device=dev, dtype=dt)
+    fl = 2 * M * N * K
+    by = (M * N + M * K) * 2
+
+    def rep(name, ms):
+        print(f"{name:<34} {ms:8.3f} ms  {fl/ms/1e9:7.1f} TF/s  "
+              f"{by/ms/1e6:7.0f} GB/s")
+
+    rep("wgrad mm dy.T@x", t(lambda: dy.T @ x))
+    for c in (4, 16, 64, 256):
+        mc = M // c
+        dyc = dy.view(c, mc, N)
+        xc = x.view(c, mc, K)
+
+        def f(dyc=dyc, xc=xc):
+            return torch.bmm(dyc.transpose(1, 2), xc).sum(0)
+
+        rep(f"wgrad bmm c={c}", t(f))
+    # L2 shape
+    N2, K2 = 256, 512
+    dy2 = torch.randn(M, N2, device=dev, dtype=dt)
+    y1 = torch.randn(M, K2, device=dev, dtype=dt)
+    fl2 = 2 * M * N2 * K2
+    by2 = (M * N2 + M * K2) * 2
+
+    def rep2(name, ms):
+        print(f"{name:<34} {ms:8.3f} ms  {fl2/ms/1e9:7.1f} TF/s  "
+              f"{by2/ms/1e6:7.0f} GB/s")
+
+    rep2("wgrad L2 mm", t(lambda: dy2.T @ y1))
+    for c in (16, 64):
+        mc = M // c
+
+        def f2(c=c, mc=mc):
+            return torch.bmm(
+                dy2.view(c, mc, N2).transpose(1, 2), y1.view(c, mc, K2)
+            ).sum(0)
+
+        rep2(f"wgrad L2 bmm c={c}", t(f2))
+
+
+if os.environ.get("RSDL_WGRAD") == "1":
+    wgrad_variants()
+    raise SystemExit
+
+
+if __name__ == "__main__" and os.environ.get("RSDL_WGRAD") != "1":
     main()
