@@ -176,6 +176,35 @@ def main():
     ms = timeit_gpu(lambda: step_full_bf16(x_bf16), iters=20)
     print(f"MLP step pure bf16 model   : {ms:7.3f} ms")
 
+    # Step decomposition (bf16 autocast path)
+    with torch.autocast("cuda", torch.bfloat16):
+        out_f = model(x_bf16)
+        loss_f = torch.nn.functional.mse_loss(out_f.float(), y)
+
+    def fwd_only():
+        with torch.autocast("cuda", torch.bfloat16):
+            out = model(x_bf16)
+            return torch.nn.functional.mse_loss(out.float(), y)
+
+    ms = timeit_gpu(fwd_only, iters=20)
+    print(f"  fwd+loss only            : {ms:7.3f} ms")
+
+    def fwd_bwd():
+        opt.zero_grad(set_to_none=True)
+        loss = fwd_only()
+        loss.backward()
+
+    ms = timeit_gpu(fwd_bwd, iters=20)
+    print(f"  fwd+bwd (no opt)         : {ms:7.3f} ms")
+
+    gparams = [p for p in model.parameters()]
+
+    def opt_only():
+        opt.step()
+
+    ms = timeit_gpu(opt_only, iters=20)
+    print(f"  opt.step only            : {ms:7.3f} ms")
+
 
 if __name__ == "__main__":
     main()
