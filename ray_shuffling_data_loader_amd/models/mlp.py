@@ -43,7 +43,14 @@ class _ChunkedLinearFn(torch.autograd.Function):
         dy = dy.contiguous()
         dx = dy @ weight
         m = x.shape[0]
-        c = _wgrad_chunks(m)
+        # Chunking only pays for wide-output, tall-K wgrads; for skinny
+        # outputs (e.g. the final [*,1] head) hipBLASLt's plain mm is fine
+        # and the batched kernel is pathological (11.7 ms for N=1 vs 0.13).
+        c = (
+            _wgrad_chunks(m)
+            if min(dy.shape[1], x.shape[1]) >= 32 and m >= 1 << 16
+            else 1
+        )
         if c > 1:
             dw = (
                 torch.bmm(
