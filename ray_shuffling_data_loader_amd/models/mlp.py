@@ -81,6 +81,60 @@ class ChunkedLinear(nn.Linear):
         return super().forward(x)
 
 
+class _LinearReLUFn(torch.autograd.Function):
+    """Linear + bias + ReLU with the GEMM-epilogue fusion
+    (aten._addmm_activation -> hipBLASLt ReLU epilogue: no separate
+    bias-add/relu elementwise pass over the [M, N] activation) and the
+    split-K chunked weight gradient. The ReLU mask for backward is derived
+    from the saved OUTPUT (y > 0), so no extra mask tensor."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        y = torch.ops.aten._addmm_activation(
+            bias, x, weight.t(), use_gelu=False
+        )
+        ctx.save_for_backward(x, weight, y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, y = ctx.saved_tensors
+        dy = dy * (y > 0)  # relu'
+        dx = dy @ weight
+        m = x.shape[0]
+        c = (
+            _wgrad_chunks(m)
+            if min(dy.shape[1], x.shape[1]) >= 32 and m >= 1 << 16
+            else 1
+        )
+        if c > 1:
+            dw = (
+                torch.bmm(
+                    dy.view(c, m // c, dy.shape[1]).transpose(1, 2),
+                    x.view(c, m // c, x.shape[1]),
+                )
+                .sum(0)
+            )
+        else:
+            dw = dy.t() @ x
+        return dx, dw, dy.sum(0)
+
+
+class LinearReLU(nn.Linear):
+    """Fused Linear+ReLU layer (GEMM epilogue ReLU + chunked wgrad)."""
+
+    def forward(self, x):
+        if x.is_cuda and x.dim() == 2 and self.bias is not None:
+            w, b = self.weight, self.bias
+            if torch.is_autocast_enabled():
+                adt = torch.get_autocast_dtype("cuda")
+                x = x.to(adt)
+                w = w.to(adt)
+                b = b.to(adt)
+            return _LinearReLUFn.apply(x, w, b)
+        return torch.relu(super().forward(x))
+
+
 class TabularMLP(nn.Module):
     """MLP over a [B, num_features] float batch -> scalar regression."""
 
@@ -96,7 +150,10 @@ class TabularMLP(nn.Module):
         d = num_features
         for i in range(depth):
             h = hidden // (2**i)
-            layers += [linear_cls(d, h), nn.ReLU()]
+            if linear_cls is ChunkedLinear:
+                layers.append(LinearReLU(d, h))
+            else:
+                layers += [linear_cls(d, h), nn.ReLU()]
             d = h
         layers.append(linear_cls(d, 1))
         self.net = nn.Sequential(*layers)

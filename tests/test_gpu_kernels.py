@@ -293,3 +293,29 @@ def test_chunked_linear_autocast_step(dev):
         p.grad is not None and torch.isfinite(p.grad).all()
         for p in model.parameters()
     )
+
+
+def test_linear_relu_fused_grads_match(dev):
+    from ray_shuffling_data_loader_amd.models.mlp import LinearReLU
+
+    torch.manual_seed(3)
+    m, k, n = 4096, 64, 128
+    x = torch.randn(m, k, device=dev, requires_grad=True)
+    ref_lin = torch.nn.Linear(k, n).to(dev)
+    fused = LinearReLU(k, n).to(dev)
+    fused.load_state_dict(ref_lin.state_dict())
+
+    y_ref = torch.relu(ref_lin(x))
+    g = torch.randn_like(y_ref)
+    y_ref.backward(g)
+    gx_ref = x.grad.clone()
+    gw_ref = ref_lin.weight.grad.clone()
+    gb_ref = ref_lin.bias.grad.clone()
+
+    x.grad = None
+    y = fused(x)
+    assert torch.allclose(y, y_ref, atol=1e-5)
+    y.backward(g)
+    assert torch.allclose(x.grad, gx_ref, atol=1e-5)
+    assert torch.allclose(fused.weight.grad, gw_ref, atol=1e-3, rtol=1e-4)
+    assert torch.allclose(fused.bias.grad, gb_ref, atol=1e-4)
