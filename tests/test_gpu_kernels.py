@@ -319,3 +319,48 @@ def test_linear_relu_fused_grads_match(dev):
     assert torch.allclose(x.grad, gx_ref, atol=1e-5)
     assert torch.allclose(fused.weight.grad, gw_ref, atol=1e-3, rtol=1e-4)
     assert torch.allclose(fused.bias.grad, gb_ref, atol=1e-4)
+
+
+def test_end_to_end_gpu_tabular_dataspec(dev, tmp_path):
+    # Heterogeneous DATA_SPEC path on GPU (int64+fp64 columns -> fused
+    # unpack kernel, no feature matrix): the reference's canonical shape.
+    from ray_shuffling_data_loader_amd.data_generation import (
+        DATA_SPEC,
+        generate_data,
+    )
+    from ray_shuffling_data_loader_amd.torch_dataset import (
+        TorchShufflingDataset,
+    )
+    from ray_shuffling_data_loader_amd.utils.schema import (
+        NUMPY_TO_TORCH_DTYPE,
+    )
+
+    num_rows = 50_000
+    filenames, _ = generate_data(num_rows, 2, 1, 0.0, str(tmp_path))
+    feature_columns = list(DATA_SPEC.keys())
+    feature_types = [
+        NUMPY_TO_TORCH_DTYPE[np.dtype(dt)] for _, _, dt in DATA_SPEC.values()
+    ]
+    label_column = feature_columns.pop()
+    label_type = feature_types.pop()
+    ds = TorchShufflingDataset(
+        list(filenames),
+        1,
+        num_trainers=1,
+        batch_size=5000,
+        rank=0,
+        num_reducers=4,
+        feature_columns=feature_columns,
+        feature_types=feature_types,
+        label_column=label_column,
+        label_type=label_type,
+        device=dev,
+    )
+    ds.set_epoch(0)
+    total = 0
+    for data, target in ds:
+        assert all(t.is_cuda for t in data)
+        for t, dt in zip(data, feature_types):
+            assert t.dtype == dt
+        total += len(target)
+    assert total == num_rows
