@@ -297,6 +297,14 @@ def main():
                 "mean_batch_wait_ms": (
                     sum(waits) / len(waits) * 1000 if waits else None
                 ),
+                "p95_batch_wait_ms": (
+                    sorted(waits)[int(0.95 * len(waits))] * 1000
+                    if waits
+                    else None
+                ),
+                "max_batch_wait_ms": (
+                    max(waits) * 1000 if waits else None
+                ),
             },
         }
         print(json.dumps(result), flush=True)
