@@ -240,6 +240,13 @@ def main():
     waits = []
     for _ in range(args.steps):
         waits.append(one_step())
+    if os.environ.get("RSDL_DEBUG_WAITS") == "1" and rank == 0:
+        top = sorted(enumerate(waits), key=lambda kv: -kv[1])[:5]
+        print(
+            "[debug] top waits (step_idx, ms):",
+            [(i, round(w * 1e3, 2)) for i, w in top],
+            flush=True,
+        )
     sync()
     barrier()
     sync()
