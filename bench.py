@@ -55,8 +55,8 @@ from ray_shuffling_data_loader_amd.torch_dataset import TorchShufflingDataset
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=40)
-    p.add_argument("--warmup", type=int, default=12)
+    p.add_argument("--steps", type=int, default=80)
+    p.add_argument("--warmup", type=int, default=16)
     p.add_argument("--batch-size", type=int, default=250_000)
     p.add_argument("--rows-per-gpu", type=int, default=12_500_000)
     p.add_argument("--num-cols", type=int, default=100)
