@@ -59,6 +59,9 @@ void launch_partition_scatter(const int32_t* dest, const int32_t* block_base,
 void launch_pack_tiled(void* packed, int64_t row_stride, const ColTable& table,
                        int32_t num_cols, int64_t n_rows, int32_t tile_rows,
                        int64_t lds_bytes, int32_t n8, hipStream_t stream);
+void launch_wgrad_bf16(const void* dy, const void* x, float* dW, float* db,
+                       int64_t M, int32_t N, int32_t K, int32_t split_m,
+                       hipStream_t stream);
 
 namespace {
 
@@ -312,6 +315,35 @@ at::Tensor pack_columns_tiled(const std::vector<at::Tensor>& cols,
   return packed;
 }
 
+// MFMA split-M weight gradient: dW = dy^T @ x (+ db = dy.sum(0)).
+// Returns (dW fp32 [N,K], db fp32 [N] or empty).
+std::vector<at::Tensor> wgrad_bf16(const at::Tensor& dy, const at::Tensor& x,
+                                   bool with_bias) {
+  TORCH_CHECK(dy.is_cuda() && x.is_cuda(), "wgrad: inputs on GPU");
+  TORCH_CHECK(dy.scalar_type() == at::kBFloat16 &&
+                  x.scalar_type() == at::kBFloat16,
+              "wgrad: bf16 inputs only");
+  TORCH_CHECK(dy.dim() == 2 && x.dim() == 2 && dy.size(0) == x.size(0),
+              "wgrad: dy [M,N], x [M,K]");
+  TORCH_CHECK(dy.is_contiguous() && x.is_contiguous(),
+              "wgrad: contiguous inputs");
+  int64_t M = dy.size(0);
+  int32_t N = (int32_t)dy.size(1);
+  int32_t K = (int32_t)x.size(1);
+  auto dW = at::zeros({N, K}, dy.options().dtype(at::kFloat));
+  auto db = with_bias ? at::zeros({N}, dy.options().dtype(at::kFloat))
+                      : at::Tensor();
+  int32_t tiles = ((N + 63) / 64) * ((K + 63) / 64);
+  int32_t split = (int32_t)std::min<int64_t>(
+      std::max<int64_t>(1, 2048 / tiles), std::max<int64_t>(1, M / 64));
+  if (M > 0) {
+    launch_wgrad_bf16(dy.data_ptr(), x.data_ptr(), dW.data_ptr<float>(),
+                      with_bias ? db.data_ptr<float>() : nullptr, M, N, K,
+                      split, current_stream());
+  }
+  return {dW, db};
+}
+
 }  // namespace
 }  // namespace rsdl
 
@@ -332,6 +364,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("row_stride"));
   m.def("partition_build_perm", &rsdl::partition_build_perm, py::arg("dest"),
         py::arg("num_dests"));
+  m.def("wgrad_bf16", &rsdl::wgrad_bf16, py::arg("dy"), py::arg("x"),
+        py::arg("with_bias") = true);
   m.attr("DT_F32") = (int)rsdl::DT_F32;
   m.attr("DT_F64") = (int)rsdl::DT_F64;
   m.attr("DT_I32") = (int)rsdl::DT_I32;

@@ -9,8 +9,20 @@ loader yields, and the step executes a genuine forward+backward+optimizer
 update on the GPU.
 """
 
+import os
+
 import torch
 import torch.nn as nn
+
+_USE_WGRAD_KERNEL = os.environ.get("RSDL_WGRAD_KERNEL", "1") == "1"
+
+
+def _wgrad_bf16_kernel(dy, x, with_bias):
+    """Fused MFMA split-M wgrad (+bias grad) — csrc/wgrad_kernel.hip."""
+    from ray_shuffling_data_loader_amd.ops.shuffle_ops import wgrad
+
+    dw, db = wgrad(dy, x, with_bias)
+    return dw.to(dy.dtype), (db.to(dy.dtype) if db is not None else None)
 
 
 def _wgrad_chunks(m: int, target: int = 16) -> int:
@@ -103,6 +115,15 @@ class _LinearReLUFn(torch.autograd.Function):
         dy = torch.ops.aten.threshold_backward(dy, y, 0)
         dx = dy @ weight
         m = x.shape[0]
+        if (
+            _USE_WGRAD_KERNEL
+            and dy.is_cuda
+            and dy.dtype == torch.bfloat16
+            and x.dtype == torch.bfloat16
+            and m >= 1 << 16
+        ):
+            dw, db = _wgrad_bf16_kernel(dy, x, with_bias=True)
+            return dx, dw, db
         c = (
             _wgrad_chunks(m)
             if min(dy.shape[1], x.shape[1]) >= 32 and m >= 1 << 16

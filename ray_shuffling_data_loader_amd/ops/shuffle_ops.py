@@ -232,3 +232,21 @@ def partition_rows(
     order = torch.argsort(dest, stable=True)
     grouped = gather_rows(packed, order)
     return grouped, counts
+
+
+# ---------------------------------------------------------------------------
+# MFMA split-M weight gradient (trainer-side hot op; csrc/wgrad_kernel.hip).
+# ---------------------------------------------------------------------------
+
+
+def wgrad(dy: torch.Tensor, x: torch.Tensor, with_bias: bool = True):
+    """dW = dy^T @ x (fp32) and db = dy.sum(0) in one fused MFMA kernel
+    (bf16 inputs, split-M atomic reduction). CPU / non-bf16 fallback uses
+    plain matmul."""
+    if dy.is_cuda and dy.dtype == torch.bfloat16 and x.dtype == torch.bfloat16:
+        hip = _load_hip()
+        dw, db = hip.wgrad_bf16(dy.contiguous(), x.contiguous(), with_bias)
+        return dw, (db if with_bias else None)
+    dw = dy.t().float() @ x.float()
+    db = dy.float().sum(0) if with_bias else None
+    return dw, db

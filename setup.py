@@ -37,7 +37,11 @@ cmdclass = {}
 try:
     from torch.utils import cpp_extension as torch_cpp_ext
 
-    hip_sources = ["csrc/shuffle_ops.cpp", "csrc/shuffle_kernels.hip"]
+    hip_sources = [
+        "csrc/shuffle_ops.cpp",
+        "csrc/shuffle_kernels.hip",
+        "csrc/wgrad_kernel.hip",
+    ]
     if all(os.path.exists(s) for s in hip_sources):
         ext_modules.append(
             torch_cpp_ext.CUDAExtension(

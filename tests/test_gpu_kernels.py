@@ -364,3 +364,26 @@ def test_end_to_end_gpu_tabular_dataspec(dev, tmp_path):
             assert t.dtype == dt
         total += len(target)
     assert total == num_rows
+
+
+def test_wgrad_kernel_matches_reference(dev):
+    from ray_shuffling_data_loader_amd.ops.shuffle_ops import wgrad
+
+    torch.manual_seed(11)
+    for m, n, k in [
+        (250_000, 512, 100),
+        (250_000, 256, 512),
+        (65_536, 128, 256),
+        (100_003, 64, 100),  # odd M, K not multiple of 64
+        (70_000, 1, 128),    # degenerate head shape
+    ]:
+        dy = torch.randn(m, n, device=dev, dtype=torch.bfloat16)
+        x = torch.randn(m, k, device=dev, dtype=torch.bfloat16)
+        dw, db = wgrad(dy, x, with_bias=True)
+        ref_dw = dy.t().float() @ x.float()
+        ref_db = dy.float().sum(0)
+        # fp32 accumulation in both; ordering differs -> tiny tolerance
+        assert torch.allclose(dw, ref_dw, atol=2.0, rtol=1e-2), (
+            m, n, k, (dw - ref_dw).abs().max().item(),
+        )
+        assert torch.allclose(db, ref_db, atol=2.0, rtol=1e-2), (m, n, k)
