@@ -123,19 +123,31 @@ wgrad_bf16_kernel(const short* __restrict__ dy,  // [M, N] bf16 bits
 
   f32x16 acc = {};
   float bias_acc = 0.0f;
+  // Bias grad comes straight from the A fragments already in registers
+  // (each lane's a[] covers dy[mfrag..mfrag+7][n = wn + lane&31]; summed
+  // over stages + the lane-half split this is exactly sum_m dy[m,n]) —
+  // no LDS traffic, no bank conflicts.
   const bool do_bias = (db != nullptr) && (k0 == 0) && (wk == 0);
 
-  StageRegs regs;
+  // Two-deep software pipeline: loads for stage s are issued at stage s-2,
+  // so each stage's global latency has ~2 stages of MFMA+LDS work to hide
+  // under (1-deep measured 73% SQ_WAIT_ANY).
+  StageRegs regs[2];
   int buf = 0;
   if (m_lo < m_hi) {
-    load_stage(dy, x, M, N, K, n0, k0, m_lo, m_hi, tid, regs);
+    load_stage(dy, x, M, N, K, n0, k0, m_lo, m_hi, tid, regs[0]);
   }
+  if (m_lo + MT < m_hi) {
+    load_stage(dy, x, M, N, K, n0, k0, m_lo + MT, m_hi, tid, regs[1]);
+  }
+  int32_t parity = 0;
   for (int64_t m0 = m_lo; m0 < m_hi; m0 += MT) {
-    write_stage(lds[buf][0], lds[buf][1], tid, regs);
+    write_stage(lds[buf][0], lds[buf][1], tid, regs[parity]);
     __syncthreads();
-    // Issue next stage's global loads while computing on this one.
-    if (m0 + MT < m_hi) {
-      load_stage(dy, x, M, N, K, n0, k0, m0 + MT, m_hi, tid, regs);
+    // Refill the set we just drained, two stages ahead.
+    if (m0 + 2 * MT < m_hi) {
+      load_stage(dy, x, M, N, K, n0, k0, m0 + 2 * MT, m_hi, tid,
+                 regs[parity]);
     }
     const short* dyT = lds[buf][0];
     const short* xT = lds[buf][1];
@@ -147,21 +159,19 @@ wgrad_bf16_kernel(const short* __restrict__ dy,  // [M, N] bf16 bits
           &dyT[(wn + (lane & 31)) * LDS_STRIDE + mfrag]);
       *reinterpret_cast<uint4*>(&b) = *reinterpret_cast<const uint4*>(
           &xT[(wk + (lane & 31)) * LDS_STRIDE + mfrag]);
+      if (do_bias) {
+        #pragma unroll
+        for (int j = 0; j < 8; j++) {
+          __hip_bfloat16 h;
+          short sv = a[j];
+          *reinterpret_cast<short*>(&h) = sv;
+          bias_acc += __bfloat162float(h);
+        }
+      }
       acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
     }
-    if (do_bias) {
-      const int32_t n = wn + (lane & 31);
-      const int32_t mb = (lane >> 5) * (MT / 2);
-      float s = 0.0f;
-      #pragma unroll
-      for (int32_t m = 0; m < MT / 2; m++) {
-        __hip_bfloat16 h;
-        *reinterpret_cast<short*>(&h) = dyT[n * LDS_STRIDE + mb + m];
-        s += __bfloat162float(h);
-      }
-      bias_acc += s;
-    }
     buf ^= 1;
+    parity ^= 1;
     __syncthreads();
   }
 
