@@ -129,35 +129,29 @@ wgrad_bf16_kernel(const short* __restrict__ dy,  // [M, N] bf16 bits
   // no LDS traffic, no bank conflicts.
   const bool do_bias = (db != nullptr) && (k0 == 0) && (wk == 0);
 
-  // Two-deep software pipeline: loads for stage s are issued at stage s-2,
-  // so each stage's global latency has ~2 stages of MFMA+LDS work to hide
-  // under (1-deep measured 73% SQ_WAIT_ANY).
-  StageRegs regs[2];
+  // Two-deep software pipeline with STATIC register sets (a regs[parity]
+  // runtime index would push the arrays to scratch): the loop body is
+  // unrolled over two stages, r0/r1 alternating; loads for stage s issue
+  // at stage s-2 (1-deep measured 73% SQ_WAIT_ANY).
+  StageRegs r0, r1;
   int buf = 0;
   if (m_lo < m_hi) {
-    load_stage(dy, x, M, N, K, n0, k0, m_lo, m_hi, tid, regs[0]);
+    load_stage(dy, x, M, N, K, n0, k0, m_lo, m_hi, tid, r0);
   }
   if (m_lo + MT < m_hi) {
-    load_stage(dy, x, M, N, K, n0, k0, m_lo + MT, m_hi, tid, regs[1]);
+    load_stage(dy, x, M, N, K, n0, k0, m_lo + MT, m_hi, tid, r1);
   }
-  int32_t parity = 0;
-  for (int64_t m0 = m_lo; m0 < m_hi; m0 += MT) {
-    write_stage(lds[buf][0], lds[buf][1], tid, regs[parity]);
-    __syncthreads();
-    // Refill the set we just drained, two stages ahead.
-    if (m0 + 2 * MT < m_hi) {
-      load_stage(dy, x, M, N, K, n0, k0, m0 + 2 * MT, m_hi, tid,
-                 regs[parity]);
-    }
-    const short* dyT = lds[buf][0];
-    const short* xT = lds[buf][1];
+
+  auto compute_stage = [&](int32_t b) {
+    const short* dyT = lds[b][0];
+    const short* xT = lds[b][1];
     #pragma unroll
     for (int32_t ms = 0; ms < MT; ms += 16) {
       const int32_t mfrag = ms + ((lane >> 5) * 8);
-      bf16x8 a, b;
+      bf16x8 a, bfr;
       *reinterpret_cast<uint4*>(&a) = *reinterpret_cast<const uint4*>(
           &dyT[(wn + (lane & 31)) * LDS_STRIDE + mfrag]);
-      *reinterpret_cast<uint4*>(&b) = *reinterpret_cast<const uint4*>(
+      *reinterpret_cast<uint4*>(&bfr) = *reinterpret_cast<const uint4*>(
           &xT[(wk + (lane & 31)) * LDS_STRIDE + mfrag]);
       if (do_bias) {
         #pragma unroll
@@ -168,11 +162,31 @@ wgrad_bf16_kernel(const short* __restrict__ dy,  // [M, N] bf16 bits
           bias_acc += __bfloat162float(h);
         }
       }
-      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
+      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, bfr, acc, 0, 0, 0);
     }
-    buf ^= 1;
-    parity ^= 1;
+  };
+
+  for (int64_t m0 = m_lo; m0 < m_hi; m0 += 2 * MT) {
+    // stage A (r0)
+    write_stage(lds[buf][0], lds[buf][1], tid, r0);
     __syncthreads();
+    if (m0 + 2 * MT < m_hi) {
+      load_stage(dy, x, M, N, K, n0, k0, m0 + 2 * MT, m_hi, tid, r0);
+    }
+    compute_stage(buf);
+    buf ^= 1;
+    __syncthreads();
+    // stage B (r1)
+    if (m0 + MT < m_hi) {
+      write_stage(lds[buf][0], lds[buf][1], tid, r1);
+      __syncthreads();
+      if (m0 + 3 * MT < m_hi) {
+        load_stage(dy, x, M, N, K, n0, k0, m0 + 3 * MT, m_hi, tid, r1);
+      }
+      compute_stage(buf);
+      buf ^= 1;
+      __syncthreads();
+    }
   }
 
   #pragma unroll
