@@ -46,20 +46,51 @@ struct StageRegs {
   uint32_t x_v[4][4];
 };
 
-__device__ __forceinline__ void load_stage(
-    const short* __restrict__ dy, const short* __restrict__ x, int64_t M,
-    int32_t N, int32_t K, int32_t n0, int32_t k0, int64_t m0, int64_t m_hi,
-    int32_t tid, StageRegs& r) {
-  // thread t covers rows {m0 + (t>>4)*2 + p*32 : p=0..? } — MT=64 rows via
-  // 4 row-pairs: rows (t>>4)*2 + {0,1} + 32*p for p in 0..1? 256 threads:
-  // t>>4 in 0..15 -> base rows 0..31 (pairs); pairs p=0..1 cover rows
-  // 0..63. cols: c0 = (t&15)*4 .. +3.
+// Thread t covers 4 rows x 4 cols per operand: rows (t>>4)*2 + {0,1} +
+// 32*p (p = 0,1), cols c0 = (t&15)*4 .. +3. The fast path loads each row's
+// 4 consecutive bf16 as one 8-B vector load, UNGUARDED — per-element
+// bounds ternaries make hipcc branch around each load and wait vmcnt(0)
+// per element (serialized L2 round trips; cdna_hip_programming.md §5
+// trap 4(c)), which cost this kernel ~5x.
+typedef __attribute__((__vector_size__(4 * sizeof(short)))) short bf16x4;
+
+__device__ __forceinline__ void load_stage_fast(
+    const short* __restrict__ dy, const short* __restrict__ x, int32_t N,
+    int32_t K, int32_t n0, int32_t k0, int64_t m0, int32_t tid,
+    StageRegs& r) {
   const int32_t c0 = (tid & 15) * 4;
-  const int32_t rp = (tid >> 4) * 2;  // row pair base within 32
+  const int32_t rp = (tid >> 4) * 2;
   #pragma unroll
   for (int p = 0; p < 2; p++) {
     #pragma unroll
-    for (int q = 0; q < 2; q++) {  // the two rows of the pair
+    for (int q = 0; q < 2; q++) {
+      const int64_t m = m0 + rp + 32 * p + q;
+      const bf16x4 dv = *reinterpret_cast<const bf16x4*>(&dy[m * N + n0 + c0]);
+      const bf16x4 xv = *reinterpret_cast<const bf16x4*>(&x[m * K + k0 + c0]);
+      #pragma unroll
+      for (int j = 0; j < 4; j++) {
+        if (q == 0) {
+          r.dy_v[p * 2][j] = (uint16_t)dv[j];
+          r.x_v[p * 2][j] = (uint16_t)xv[j];
+        } else {
+          r.dy_v[p * 2][j] |= ((uint32_t)(uint16_t)dv[j]) << 16;
+          r.x_v[p * 2][j] |= ((uint32_t)(uint16_t)xv[j]) << 16;
+        }
+      }
+    }
+  }
+}
+
+__device__ void load_stage_guarded(
+    const short* __restrict__ dy, const short* __restrict__ x, int64_t M,
+    int32_t N, int32_t K, int32_t n0, int32_t k0, int64_t m0, int64_t m_hi,
+    int32_t tid, StageRegs& r) {
+  const int32_t c0 = (tid & 15) * 4;
+  const int32_t rp = (tid >> 4) * 2;
+  #pragma unroll
+  for (int p = 0; p < 2; p++) {
+    #pragma unroll
+    for (int q = 0; q < 2; q++) {
       const int64_t m = m0 + rp + 32 * p + q;
       const bool ok = m < m_hi;
       #pragma unroll
@@ -71,14 +102,26 @@ __device__ __forceinline__ void load_stage(
         const uint16_t xv =
             (ok && k < K) ? (uint16_t)x[m * K + k] : (uint16_t)0;
         if (q == 0) {
-          r.dy_v[p * 2 + 0][j] = dv;  // lo half
-          r.x_v[p * 2 + 0][j] = xv;
+          r.dy_v[p * 2][j] = dv;
+          r.x_v[p * 2][j] = xv;
         } else {
-          r.dy_v[p * 2 + 0][j] |= ((uint32_t)dv) << 16;
-          r.x_v[p * 2 + 0][j] |= ((uint32_t)xv) << 16;
+          r.dy_v[p * 2][j] |= ((uint32_t)dv) << 16;
+          r.x_v[p * 2][j] |= ((uint32_t)xv) << 16;
         }
       }
     }
+  }
+}
+
+__device__ __forceinline__ void load_stage(
+    const short* __restrict__ dy, const short* __restrict__ x, int64_t M,
+    int32_t N, int32_t K, int32_t n0, int32_t k0, int64_t m0, int64_t m_hi,
+    int32_t tid, StageRegs& r, bool full_nk) {
+  // Wave-uniform branch: interior stages take the unguarded vector path.
+  if (full_nk && m0 + MT <= m_hi) {
+    load_stage_fast(dy, x, N, K, n0, k0, m0, tid, r);
+  } else {
+    load_stage_guarded(dy, x, M, N, K, n0, k0, m0, m_hi, tid, r);
   }
 }
 
@@ -121,6 +164,7 @@ wgrad_bf16_kernel(const short* __restrict__ dy,  // [M, N] bf16 bits
   const int32_t wn = (wave & 1) * 32;
   const int32_t wk = (wave >> 1) * 32;
 
+  const bool full_nk = (n0 + WG_TILE <= N) && (k0 + WG_TILE <= K);
   f32x16 acc = {};
   float bias_acc = 0.0f;
   // Bias grad comes straight from the A fragments already in registers
@@ -136,10 +180,11 @@ wgrad_bf16_kernel(const short* __restrict__ dy,  // [M, N] bf16 bits
   StageRegs r0, r1;
   int buf = 0;
   if (m_lo < m_hi) {
-    load_stage(dy, x, M, N, K, n0, k0, m_lo, m_hi, tid, r0);
+    load_stage(dy, x, M, N, K, n0, k0, m_lo, m_hi, tid, r0, full_nk);
   }
   if (m_lo + MT < m_hi) {
-    load_stage(dy, x, M, N, K, n0, k0, m_lo + MT, m_hi, tid, r1);
+    load_stage(dy, x, M, N, K, n0, k0, m_lo + MT, m_hi, tid, r1,
+               full_nk);
   }
 
   auto compute_stage = [&](int32_t b) {
@@ -171,7 +216,8 @@ wgrad_bf16_kernel(const short* __restrict__ dy,  // [M, N] bf16 bits
     write_stage(lds[buf][0], lds[buf][1], tid, r0);
     __syncthreads();
     if (m0 + 2 * MT < m_hi) {
-      load_stage(dy, x, M, N, K, n0, k0, m0 + 2 * MT, m_hi, tid, r0);
+      load_stage(dy, x, M, N, K, n0, k0, m0 + 2 * MT, m_hi, tid, r0,
+                 full_nk);
     }
     compute_stage(buf);
     buf ^= 1;
@@ -181,7 +227,8 @@ wgrad_bf16_kernel(const short* __restrict__ dy,  // [M, N] bf16 bits
       write_stage(lds[buf][0], lds[buf][1], tid, r1);
       __syncthreads();
       if (m0 + 3 * MT < m_hi) {
-        load_stage(dy, x, M, N, K, n0, k0, m0 + 3 * MT, m_hi, tid, r1);
+        load_stage(dy, x, M, N, K, n0, k0, m0 + 3 * MT, m_hi, tid, r1,
+                   full_nk);
       }
       compute_stage(buf);
       buf ^= 1;
