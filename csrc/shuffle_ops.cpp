@@ -62,6 +62,9 @@ void launch_pack_tiled(void* packed, int64_t row_stride, const ColTable& table,
 void launch_wgrad_bf16(const void* dy, const void* x, float* dW, float* db,
                        int64_t M, int32_t N, int32_t K, int32_t split_m,
                        hipStream_t stream);
+void launch_wgrad_wide(const void* dy, const void* x, float* dW, float* db,
+                       int64_t M, int32_t N, int32_t K, int32_t split_m,
+                       hipStream_t stream);
 
 namespace {
 
@@ -330,6 +333,29 @@ std::vector<at::Tensor> wgrad_bf16(const at::Tensor& dy, const at::Tensor& x,
   int64_t M = dy.size(0);
   int32_t N = (int32_t)dy.size(1);
   int32_t K = (int32_t)x.size(1);
+  bool wide = N >= 64 && K >= 64 && M >= 4096;
+  if (wide) {
+    // Wide-tile kernel wants 128-multiples; pad (cheap relative to the
+    // re-read amplification it removes) and narrow the result.
+    auto dyp = (N % 128) ? at::constant_pad_nd(dy, {0, 128 - N % 128})
+                         : dy.contiguous();
+    auto xp = (K % 128) ? at::constant_pad_nd(x, {0, 128 - K % 128})
+                        : x.contiguous();
+    int32_t Np = (int32_t)dyp.size(1), Kp = (int32_t)xp.size(1);
+    auto dWp = at::zeros({Np, Kp}, dy.options().dtype(at::kFloat));
+    auto dbp = with_bias ? at::zeros({Np}, dy.options().dtype(at::kFloat))
+                         : at::Tensor();
+    int32_t tiles = (Np / 128) * (Kp / 128);
+    int32_t split = (int32_t)std::min<int64_t>(
+        std::max<int64_t>(1, 2048 / std::max(1, tiles)),
+        std::max<int64_t>(1, M / 64));
+    launch_wgrad_wide(dyp.data_ptr(), xp.data_ptr(), dWp.data_ptr<float>(),
+                      with_bias ? dbp.data_ptr<float>() : nullptr, M, Np, Kp,
+                      split, current_stream());
+    auto dW = dWp.narrow(0, 0, N).narrow(1, 0, K).contiguous();
+    auto db = with_bias ? dbp.narrow(0, 0, N).contiguous() : at::Tensor();
+    return {dW, db};
+  }
   auto dW = at::zeros({N, K}, dy.options().dtype(at::kFloat));
   auto db = with_bias ? at::zeros({N}, dy.options().dtype(at::kFloat))
                       : at::Tensor();
