@@ -319,6 +319,22 @@ def main():
     if world > 1:
         torch.distributed.destroy_process_group()
 
+    # Driver hygiene: the SCALE tier runs N=1,2,4,8 back-to-back on one box;
+    # each N generates its own shard dir (up to ~40 GB at N=8). When using
+    # the DEFAULT data dir, each rank removes its own generated files so
+    # runs don't accumulate on /tmp. An explicit --data-dir is preserved
+    # (cached reuse).
+    if args.data_dir is None:
+        import shutil
+
+        for i in my_indices:
+            try:
+                os.remove(filenames[i])
+            except OSError:
+                pass
+        if rank == 0:
+            shutil.rmtree(shard_dir, ignore_errors=True)
+
 
 if __name__ == "__main__":
     main()
