@@ -21,7 +21,7 @@ Single process (no torchrun) also works: local central mode.
 
 import argparse
 import os
-import statistics
+
 import sys
 import tempfile
 import time

@@ -14,7 +14,7 @@ Reference ops replaced here: ``pd.concat`` + ``df.sample(frac=1)``
 -> :func:`unpack_permute`.
 """
 
-from typing import Dict, List, Optional
+from typing import Dict, Optional
 
 import numpy as np
 import torch
@@ -22,7 +22,6 @@ import torch
 from ray_shuffling_data_loader_amd.utils.schema import (
     Schema,
     TORCH_TO_NUMPY_DTYPE,
-    dtype_bytes,
 )
 
 _hip = None

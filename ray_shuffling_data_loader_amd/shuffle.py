@@ -8,7 +8,7 @@ actual pipeline lives in :class:`ray_shuffling_data_loader_amd.engine.
 ShuffleEngine` (map -> RCCL all-to-all -> fused permute on MI355X).
 """
 
-from typing import List, Optional, Sequence
+from typing import List, Sequence
 
 from ray_shuffling_data_loader_amd.engine import ShuffleEngine
 

@@ -14,7 +14,7 @@ per-column dtype handling of ``convert_to_tensor``
 """
 
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import Dict, List, Optional
 
 import numpy as np
 import torch

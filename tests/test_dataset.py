@@ -245,8 +245,6 @@ def test_source_cache_reuse(small_data):
 def test_engine_failure_propagates(tmp_path):
     # A dying engine (bad file) must raise in the consumer, not hang.
     bad = [str(tmp_path / "nonexistent.parquet")]
-    import pyarrow.parquet  # noqa: F401 (schema inference fails on open)
-
     with pytest.raises(Exception):
         ds = ShufflingDataset(
             bad, 1, num_trainers=1, batch_size=100, rank=0, num_reducers=1

@@ -6,7 +6,6 @@ all-to-all rows exchange (isend/irecv on gloo), per-rank reducer split, and
 the exactly-once global row invariant.
 """
 
-import multiprocessing
 import os
 import tempfile
 

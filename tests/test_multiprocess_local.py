@@ -3,10 +3,8 @@ named queue + shuffle driver; a SEPARATE trainer process connects by name
 and consumes its partition (reference dataset.py:52-84 + batch_queue
 connect retry). CPU-only."""
 
-import multiprocessing
 import time
 
-import pytest
 import torch
 
 

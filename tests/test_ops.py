@@ -2,8 +2,6 @@
 implementations; the GPU kernels are tested against these in
 test_gpu_kernels.py)."""
 
-import numpy as np
-import pytest
 import torch
 
 from ray_shuffling_data_loader_amd.ops.shuffle_ops import (

@@ -8,7 +8,6 @@ import pytest
 import torch
 
 from ray_shuffling_data_loader_amd.data_generation import generate_data
-from ray_shuffling_data_loader_amd.dataset import ShufflingDataset
 from ray_shuffling_data_loader_amd.engine import ShuffleEngine
 from ray_shuffling_data_loader_amd.shuffle import BatchConsumer
 

@@ -7,7 +7,6 @@ import os
 import time
 
 import pytest
-import torch
 
 from ray_shuffling_data_loader_amd.utils.stats import (
     CONSUMER_FIELDS,
