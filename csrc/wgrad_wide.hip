@@ -31,20 +31,27 @@ typedef __attribute__((__vector_size__(16 * sizeof(float)))) float w_f32x16;
 // Per-operand stage registers: operand with C columns contributes
 // C*WMT/256 elements per thread = C/32 packed row-pair words per thread
 // at 4 cols per group -> C/128 groups of 4 words.
+#define WG_GROUPS(C) (((C / 4) * (WMT / 2) + 511) / 512)
+
 template <int C>
 struct WStage {
-  uint32_t v[C / 128][4];  // [group][col j] packed (m, m+1) halfwords
+  uint32_t v[WG_GROUPS(C)][4];  // [group][col j] packed (m,m+1) halfwords
 };
+
+#define W_THREADS 512
 
 template <int C>
 __device__ __forceinline__ void w_load(const short* __restrict__ src,
                                        int64_t ld, int32_t c_base,
                                        int64_t m0, int32_t tid,
                                        WStage<C>& r) {
-  // unit u = g*256 + tid covers row-pair pr = u / (C/4), cols (u % (C/4))*4.
+  // unit u = g*W_THREADS + tid covers row-pair pr = u / (C/4),
+  // cols (u % (C/4))*4. For C*WMT/8 < W_THREADS the tail threads idle
+  // (wave-uniform for the shapes used).
   #pragma unroll
-  for (int g = 0; g < C / 128; g++) {
-    const int32_t u = g * 256 + tid;
+  for (int g = 0; g < WG_GROUPS(C); g++) {
+    const int32_t u = g * W_THREADS + tid;
+    if (u >= (C / 4) * (WMT / 2)) break;
     const int32_t c4 = u % (C / 4);
     const int32_t pr = u / (C / 4);
     const int64_t m = m0 + pr * 2;
@@ -65,8 +72,9 @@ __device__ __forceinline__ void w_load_guarded(
     const short* __restrict__ src, int64_t ld, int32_t c_base, int64_t m0,
     int64_t m_hi, int32_t tid, WStage<C>& r) {
   #pragma unroll
-  for (int g = 0; g < C / 128; g++) {
-    const int32_t u = g * 256 + tid;
+  for (int g = 0; g < WG_GROUPS(C); g++) {
+    const int32_t u = g * W_THREADS + tid;
+    if (u >= (C / 4) * (WMT / 2)) break;
     const int32_t c4 = u % (C / 4);
     const int32_t pr = u / (C / 4);
     const int64_t m = m0 + pr * 2;
@@ -87,8 +95,9 @@ template <int C>
 __device__ __forceinline__ void w_write(short* lds_t, int32_t tid,
                                         const WStage<C>& r) {
   #pragma unroll
-  for (int g = 0; g < C / 128; g++) {
-    const int32_t u = g * 256 + tid;
+  for (int g = 0; g < WG_GROUPS(C); g++) {
+    const int32_t u = g * W_THREADS + tid;
+    if (u >= (C / 4) * (WMT / 2)) break;
     const int32_t c4 = u % (C / 4);
     const int32_t pr = u / (C / 4);
     #pragma unroll
@@ -101,7 +110,7 @@ __device__ __forceinline__ void w_write(short* lds_t, int32_t tid,
 
 // NW x KW dW tile per WG; per wave: WN x WK subtile.
 template <int NW, int KW, int WN, int WK>
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(W_THREADS)
 wgrad_wide_kernel(const short* __restrict__ dy, const short* __restrict__ x,
                   float* __restrict__ dW, float* __restrict__ db, int64_t M,
                   int32_t N, int32_t K, int32_t split_m) {
@@ -119,10 +128,9 @@ wgrad_wide_kernel(const short* __restrict__ dy, const short* __restrict__ x,
   const int32_t tid = threadIdx.x;
   const int32_t lane = tid & 63;
   const int32_t wave = tid >> 6;
-  // wave subtile offsets within the WG tile
-  const int32_t wn = (NW / WN > 1) ? (wave % (NW / WN)) * WN : 0;
-  const int32_t wk = (NW / WN > 1) ? (wave / (NW / WN)) * WK
-                                   : (wave % (KW / WK)) * WK;
+  // wave subtile offsets within the WG tile ((NW/WN) x (KW/WK) == 8 waves)
+  const int32_t wn = (wave % (NW / WN)) * WN;
+  const int32_t wk = (wave / (NW / WN)) * WK;
 
   w_f32x16 acc[(WN / 32) * (WK / 32)] = {};
   // Per-A-fragment bias partials: afr[ai] covers n = wn + ai*32 + (lane&31)
@@ -260,22 +268,22 @@ void launch_wgrad_wide(const void* dy, const void* x, float* dW, float* db,
   // Tile choice: cover N fully when N >= 256, else favor K coverage.
   if (N % 256 == 0 && K % 128 == 0) {
     dim3 grid((N / 256) * (K / 128), split_m);
-    hipLaunchKernelGGL((wgrad_wide_kernel<256, 128, 64, 128>), grid,
-                       dim3(256), 0, stream,
+    hipLaunchKernelGGL((wgrad_wide_kernel<256, 128, 64, 64>), grid,
+                       dim3(W_THREADS), 0, stream,
                        reinterpret_cast<const short*>(dy),
                        reinterpret_cast<const short*>(x), dW, db, M, N, K,
                        split_m);
   } else if (N % 128 == 0 && K % 256 == 0) {
     dim3 grid((N / 128) * (K / 256), split_m);
-    hipLaunchKernelGGL((wgrad_wide_kernel<128, 256, 128, 64>), grid,
-                       dim3(256), 0, stream,
+    hipLaunchKernelGGL((wgrad_wide_kernel<128, 256, 64, 64>), grid,
+                       dim3(W_THREADS), 0, stream,
                        reinterpret_cast<const short*>(dy),
                        reinterpret_cast<const short*>(x), dW, db, M, N, K,
                        split_m);
   } else {
     dim3 grid((N / 128) * (K / 128), split_m);
-    hipLaunchKernelGGL((wgrad_wide_kernel<128, 128, 64, 64>), grid,
-                       dim3(256), 0, stream,
+    hipLaunchKernelGGL((wgrad_wide_kernel<128, 128, 32, 64>), grid,
+                       dim3(W_THREADS), 0, stream,
                        reinterpret_cast<const short*>(dy),
                        reinterpret_cast<const short*>(x), dW, db, M, N, K,
                        split_m);
