@@ -65,6 +65,10 @@ def parse_args():
     p.add_argument("--data-dir", type=str, default=None)
     p.add_argument("--cache-files", action="store_true")
     p.add_argument("--lr", type=float, default=1e-3)
+    p.add_argument("--fp16-allreduce", action="store_true",
+                   help="compress DDP gradients to fp16 for the RCCL "
+                   "allreduce (reference ray_torch_shuffle.py:184-185 "
+                   "hvd fp16 compression parity)")
     p.add_argument("--device", type=str, default=None)
     return p.parse_args()
 
@@ -144,8 +148,18 @@ def train_main(args, world, rank):
         model = TabularMLP(args.num_cols).to(device)
     if world > 1:
         model = torch.nn.parallel.DistributedDataParallel(model)
+        if args.fp16_allreduce:
+            from torch.distributed.algorithms.ddp_comm_hooks import (
+                default_hooks,
+            )
+
+            model.register_comm_hook(
+                None, default_hooks.fp16_compress_hook
+            )
     opt = torch.optim.SGD(model.parameters(), lr=args.lr, momentum=0.9)
     loss_fn = torch.nn.MSELoss()
+
+    import contextlib
 
     for epoch in range(args.num_epochs):
         ds.set_epoch(epoch)
@@ -153,18 +167,29 @@ def train_main(args, world, rank):
         n_batches = 0
         t_epoch = time.perf_counter()
         t_wait = time.perf_counter()
-        for data, target in ds:
-            wait_times.append(time.perf_counter() - t_wait)
-            if args.mock_train_step_time is not None:
-                time.sleep(args.mock_train_step_time)
-            else:
-                x = data if tabular else data[0]
-                opt.zero_grad(set_to_none=True)
-                loss = loss_fn(model(x), target)
-                loss.backward()
-                opt.step()
-            n_batches += 1
-            t_wait = time.perf_counter()
+        # Reducer partitions are binomial, so per-rank batch counts differ;
+        # DDP's join() shadows the missing allreduces on ranks that finish
+        # the epoch early (without it the gradient allreduce deadlocks —
+        # the reference example only avoids this because its train step is
+        # a sleep with the real fwd/bwd commented out).
+        join_ctx = (
+            model.join()
+            if world > 1 and args.mock_train_step_time is None
+            else contextlib.nullcontext()
+        )
+        with join_ctx:
+            for data, target in ds:
+                wait_times.append(time.perf_counter() - t_wait)
+                if args.mock_train_step_time is not None:
+                    time.sleep(args.mock_train_step_time)
+                else:
+                    x = data if tabular else data[0]
+                    opt.zero_grad(set_to_none=True)
+                    loss = loss_fn(model(x), target)
+                    loss.backward()
+                    opt.step()
+                n_batches += 1
+                t_wait = time.perf_counter()
         if device.type == "cuda":
             torch.cuda.synchronize(device)
         dur = time.perf_counter() - t_epoch
