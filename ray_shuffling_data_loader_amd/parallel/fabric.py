@@ -112,8 +112,8 @@ def exchange_rows(
     world = dist.get_world_size(group)
     rank = dist.get_rank(group)
     recv_counts = exchange_counts(send_counts, group, device=grouped.device)
-    in_splits = [int(c) for c in send_counts]
-    out_splits = [int(c) for c in recv_counts]
+    in_splits = send_counts.cpu().tolist()  # one sync, not world syncs
+    out_splits = recv_counts.tolist()
     recv = torch.empty(
         (sum(out_splits), grouped.shape[1]),
         dtype=grouped.dtype,
