@@ -177,6 +177,13 @@ class BatchQueue:
                 )
                 self._server_thread.start()
 
+    @property
+    def actor(self):
+        """Reference-API alias (batch_queue.py:63-65 exposes `.actor`): the
+        backing queue core (local mode) or client socket (connected mode);
+        None after shutdown."""
+        return self.core if self.core is not None else self._sock
+
     # ----- transport ---------------------------------------------------------
 
     def _call(self, method: str, *args, **kwargs):

@@ -370,3 +370,10 @@ def test_pull_from_streaming_batch_queue():
     assert not t.is_alive()
     assert len(consumed) == len(data)
     assert set(consumed) == set(data)
+
+
+def test_actor_alias():
+    q = make_queue()
+    assert q.actor is not None
+    q.shutdown()
+    assert q.actor is None
