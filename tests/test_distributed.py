@@ -99,3 +99,40 @@ def test_distributed_exactly_once(tmp_path, num_epochs, mp_spawn_context):
         assert len(results[1][epoch]) > num_rows // 4
     # Different epochs produce different per-rank orderings.
     assert results[0][0] != results[0][1]
+
+
+def test_distributed_fewer_files_than_ranks(tmp_path, mp_spawn_context):
+    # One Parquet file, two ranks: rank 1 maps zero files but still
+    # participates in the exchange and receives ~half the rows.
+    from ray_shuffling_data_loader_amd.data_generation import generate_data
+
+    num_rows = 6000
+    filenames, _ = generate_data(num_rows, 1, 1, 0.0, str(tmp_path))
+    filenames = list(filenames)
+
+    port_file = tempfile.NamedTemporaryFile(delete=False)
+    init_method = f"file://{port_file.name}"
+    os.unlink(port_file.name)
+
+    ctx = mp_spawn_context
+    result_q = ctx.Queue()
+    procs = [
+        ctx.Process(
+            target=_run_rank,
+            args=(r, 2, init_method, filenames, num_rows, result_q, 1),
+        )
+        for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, out = result_q.get(timeout=180)
+        assert not isinstance(out, str), out
+        results[rank] = out
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    all_keys = sorted(results[0][0] + results[1][0])
+    assert all_keys == list(range(num_rows))
+    assert len(results[1][0]) > num_rows // 4  # rank 1 still gets its share
