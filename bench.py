@@ -270,6 +270,14 @@ def main():
     except Exception:
         pass
 
+    if os.environ.get("RSDL_LOG_MEM") == "1" and is_cuda and rank == 0:
+        print(
+            f"[mem] allocated={torch.cuda.memory_allocated()/2**30:.2f}GiB "
+            f"max_allocated={torch.cuda.max_memory_allocated()/2**30:.2f}GiB "
+            f"reserved={torch.cuda.memory_reserved()/2**30:.2f}GiB",
+            flush=True,
+        )
+
     rows_per_sec = n_gpus * args.batch_size * args.steps / elapsed
     p50_wait_ms = statistics.median(waits) * 1000 if waits else None
     if rank == 0:
