@@ -115,7 +115,9 @@ class ShufflingDataset:
         self._engine = None
 
         world, _, initialized = fabric.dist_info()
-        distributed = initialized and world > 1
+        distributed = initialized and (
+            world > 1 or os.environ.get("RSDL_FORCE_COLLECTIVE") == "1"
+        )
 
         if distributed:
             # Symmetric mode: each rank owns a local single-column queue

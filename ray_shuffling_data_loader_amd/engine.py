@@ -102,7 +102,13 @@ class ShuffleEngine:
         self.source_cache = source_cache
 
         world, dist_rank, initialized = fabric.dist_info()
-        self.distributed = initialized and world > 1
+        # RSDL_FORCE_COLLECTIVE=1 runs the full collective path even at
+        # world 1 (self-exchange) — lets a single GPU exercise the exact
+        # multi-GPU code path end to end.
+        import os as _os
+
+        force = _os.environ.get("RSDL_FORCE_COLLECTIVE") == "1"
+        self.distributed = initialized and (world > 1 or force)
         if self.distributed:
             if num_trainers != world:
                 raise ValueError(
