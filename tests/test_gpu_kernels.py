@@ -387,3 +387,48 @@ def test_wgrad_kernel_matches_reference(dev):
             m, n, k, (dw - ref_dw).abs().max().item(),
         )
         assert torch.allclose(db, ref_db, atol=2.0, rtol=1e-2), (m, n, k)
+
+
+def test_collective_path_single_gpu(dev, tmp_path, monkeypatch):
+    # Full multi-GPU code path on one GPU: RCCL comm init, GPU radix
+    # partition, all_to_all_single (self-exchange), multinomial reducer
+    # split, fused permute — exactly what runs per epoch at N=8.
+    import torch.distributed as dist
+
+    from ray_shuffling_data_loader_amd.data_generation import (
+        float_data_spec,
+        generate_data,
+    )
+    from ray_shuffling_data_loader_amd.dataset import ShufflingDataset
+    from ray_shuffling_data_loader_amd.parallel import fabric
+
+    num_rows = 80_000
+    filenames, _ = generate_data(
+        num_rows, 2, 1, 0.0, str(tmp_path), spec=float_data_spec(8),
+        include_key=False,
+    )
+    monkeypatch.setenv("RSDL_FORCE_COLLECTIVE", "1")
+    dist.init_process_group(
+        "nccl",
+        init_method="tcp://127.0.0.1:29417",
+        rank=0,
+        world_size=1,
+    )
+    try:
+        ds = ShufflingDataset(
+            list(filenames),
+            2,
+            num_trainers=1,
+            batch_size=10_000,
+            rank=0,
+            num_reducers=4,
+            device=dev,
+            feature_matrix=("__features__", [f"f{i}" for i in range(8)]),
+        )
+        for epoch in range(2):
+            ds.set_epoch(epoch)
+            total = sum(len(b) for b in ds)
+            assert total == num_rows
+    finally:
+        dist.destroy_process_group()
+        fabric._shuffle_group = None
