@@ -7,6 +7,7 @@ exchange layout — on GPU via the pack_columns HIP kernel when a device is
 given, else via numpy strided views.
 """
 
+import warnings
 from concurrent.futures import ThreadPoolExecutor
 from typing import Dict, List, Optional, Sequence, Tuple
 
@@ -74,16 +75,29 @@ def fuse_schema(base: Schema, feature_matrix: Optional[Tuple[str, List[str]]]
 def read_file_columns(
     filename: str,
     schema: Schema,
-) -> Dict[str, np.ndarray]:
-    """Read one Parquet file into contiguous numpy columns (only the schema's
-    columns)."""
+    pin: bool = False,
+) -> Dict[str, torch.Tensor]:
+    """Read one Parquet file into contiguous column tensors (only the
+    schema's columns). With ``pin=True`` each column is copied ONCE from the
+    (read-only) Arrow buffer straight into a pinned-host tensor, ready for
+    async H2D — no intermediate writable copy."""
     table = pq.read_table(filename, columns=schema.names)
     out = {}
     for spec in schema.columns:
         arr = table.column(spec.name).to_numpy(zero_copy_only=False)
-        # Arrow hands back read-only buffers; torch.from_numpy needs
-        # writable C-contiguous memory (copies only when required).
-        out[spec.name] = np.require(arr, requirements=["C", "W"])
+        if not arr.flags["C_CONTIGUOUS"]:
+            arr = np.ascontiguousarray(arr)
+        with warnings.catch_warnings():
+            # Arrow buffers are read-only; we never write through this
+            # tensor (it is copied into pinned memory or consumed by the
+            # pack kernel input path), so the writability warning is noise.
+            warnings.simplefilter("ignore")
+            t = torch.from_numpy(arr)
+        if pin:
+            p = torch.empty_like(t, pin_memory=True)
+            p.copy_(t)
+            t = p
+        out[spec.name] = t
     return out
 
 
@@ -101,25 +115,24 @@ def read_files_packed(
                            device=device)
     use_gpu = device.type == "cuda"
 
+    # Reader threads do Arrow decode + the single copy into pinned-host
+    # arenas in parallel; the caller thread only issues async H2D DMAs and
+    # pack kernels, which overlap the remaining reads (torch's caching host
+    # allocator keeps each pinned block alive until its copy's stream work
+    # completes).
     def load(fn):
-        return read_file_columns(fn, schema)
+        return read_file_columns(fn, schema, pin=use_gpu)
 
     packed_parts = []
     with ThreadPoolExecutor(max_workers=max(1, reader_threads)) as pool:
-        for cols_np in pool.map(load, filenames):
-            cols = {}
-            for name, arr in cols_np.items():
-                t = torch.from_numpy(arr)
-                if use_gpu:
-                    # Stage through a pinned-host arena so the H2D copy is
-                    # a true async DMA overlapping the next file's Arrow
-                    # read and the pack kernel (torch's caching host
-                    # allocator keeps the pinned block alive until the
-                    # copy's stream work completes).
-                    pin = torch.empty_like(t, pin_memory=True)
-                    pin.copy_(t)
-                    t = pin.to(device, non_blocking=True)
-                cols[name] = t
+        for cols_host in pool.map(load, filenames):
+            if use_gpu:
+                cols = {
+                    name: t.to(device, non_blocking=True)
+                    for name, t in cols_host.items()
+                }
+            else:
+                cols = cols_host
             packed_parts.append(pack_columns(cols, schema))
     if len(packed_parts) == 1:
         return packed_parts[0]
