@@ -240,6 +240,10 @@ def main():
     waits = []
     for _ in range(args.steps):
         waits.append(one_step())
+    sync()
+    barrier()
+    sync()
+    elapsed = time.perf_counter() - t0
     if os.environ.get("RSDL_DEBUG_WAITS") == "1" and rank == 0:
         top = sorted(enumerate(waits), key=lambda kv: -kv[1])[:5]
         print(
@@ -247,10 +251,6 @@ def main():
             [(i, round(w * 1e3, 2)) for i, w in top],
             flush=True,
         )
-    sync()
-    barrier()
-    sync()
-    elapsed = time.perf_counter() - t0
 
     # MAX over ranks (slowest rank defines job time).
     if world > 1:
