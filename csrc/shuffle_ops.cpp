@@ -264,16 +264,29 @@ std::vector<at::Tensor> partition_build_perm(const at::Tensor& dest,
   return {perm, per_dest};
 }
 
-// LDS-tiled pack (no scatter perm): cols -> packed rows.
+// LDS-tiled pack (no scatter perm): cols -> packed rows. ``out`` may be a
+// contiguous [n_rows, row_stride] uint8 view (e.g. a row-slice of a larger
+// preallocated source tensor) to pack in place.
 at::Tensor pack_columns_tiled(const std::vector<at::Tensor>& cols,
                               const std::vector<int64_t>& packed_offsets,
                               const std::vector<int64_t>& packed_dtype_codes,
-                              int64_t row_stride) {
+                              int64_t row_stride,
+                              const c10::optional<at::Tensor>& out) {
   TORCH_CHECK(!cols.empty() && cols.size() <= 128, "1..128 columns");
   TORCH_CHECK(row_stride % 16 == 0, "row_stride must be multiple of 16");
   int64_t n_rows = cols[0].size(0);
-  auto packed =
-      at::empty({n_rows, row_stride}, cols[0].options().dtype(at::kByte));
+  at::Tensor packed;
+  if (out.has_value()) {
+    packed = *out;
+    TORCH_CHECK(packed.is_cuda() && packed.is_contiguous() &&
+                    packed.scalar_type() == at::kByte &&
+                    packed.size(0) == n_rows &&
+                    packed.size(1) == row_stride,
+                "out must be contiguous uint8 [n_rows, row_stride]");
+  } else {
+    packed =
+        at::empty({n_rows, row_stride}, cols[0].options().dtype(at::kByte));
+  }
   ColTable table{};
   for (size_t c = 0; c < cols.size(); ++c) {
     const auto& t = cols[c];
@@ -387,7 +400,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("row_stride"), py::arg("perm") = c10::nullopt);
   m.def("pack_columns_tiled", &rsdl::pack_columns_tiled, py::arg("cols"),
         py::arg("packed_offsets"), py::arg("packed_dtype_codes"),
-        py::arg("row_stride"));
+        py::arg("row_stride"), py::arg("out") = c10::nullopt);
   m.def("partition_build_perm", &rsdl::partition_build_perm, py::arg("dest"),
         py::arg("num_dests"));
   m.def("wgrad_bf16", &rsdl::wgrad_bf16, py::arg("dy"), py::arg("x"),

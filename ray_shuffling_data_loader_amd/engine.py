@@ -275,9 +275,13 @@ class ShuffleEngine:
             grouped, send_counts = partition_rows(
                 src, dest, self.num_trainers
             )
+            del dest
             recv, _ = fabric.exchange_rows(
                 grouped, send_counts, self.group
             )
+            # Free the send staging (one full shard) before the reduce-side
+            # gathers allocate the partitions.
+            del grouped
             rows_per_trainer = {self.rank: recv}
         elif self.num_trainers == 1:
             rows_per_trainer = {0: src}

@@ -116,6 +116,7 @@ def pack_columns(
     columns: Dict[str, torch.Tensor],
     schema: Schema,
     perm: Optional[torch.Tensor] = None,
+    out: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """Interleave column tensors into packed rows (cast to the schema dtype
     per column). ``perm`` scatters row i of the input to row perm[i] of the
@@ -132,14 +133,18 @@ def pack_columns(
             codes.append(_dtype_code(spec.dtype))
         if perm is None:
             # LDS-tiled transpose pack: coalesced column reads + one
-            # contiguous row-major store stream.
+            # contiguous row-major store stream (optionally in place).
             return hip.pack_columns_tiled(
-                cols, offs, codes, schema.row_stride
+                cols, offs, codes, schema.row_stride, out
             )
         return hip.pack_columns(
             cols, offs, codes, schema.row_stride, perm
         )
-    packed = torch.zeros(n, schema.row_stride, dtype=torch.uint8)
+    packed = (
+        out
+        if out is not None
+        else torch.zeros(n, schema.row_stride, dtype=torch.uint8)
+    )
     packed_np = packed.numpy()
     perm_np = perm.cpu().numpy() if perm is not None else None
     for spec in schema.columns:
