@@ -31,9 +31,18 @@ GPU work runs on a dedicated side HIP stream so the trainer's compute stream
 never stalls behind shuffle work.
 """
 
+import os
+import sys
 import threading
 import time
 from typing import Dict, List, Optional, Sequence, Tuple
+
+_VERBOSE = os.environ.get("RSDL_VERBOSE") == "1"
+
+
+def _vlog(msg: str) -> None:
+    if _VERBOSE:
+        print(f"[rsdl-engine] {msg}", file=sys.stderr, flush=True)
 
 import numpy as np
 import torch
@@ -276,12 +285,20 @@ class ShuffleEngine:
                 src, dest, self.num_trainers
             )
             del dest
+            _vlog(
+                f"rank {self.rank} epoch {epoch}: exchanging "
+                f"{int(grouped.shape[0])} rows"
+            )
             recv, _ = fabric.exchange_rows(
                 grouped, send_counts, self.group
             )
             # Free the send staging (one full shard) before the reduce-side
             # gathers allocate the partitions.
             del grouped
+            _vlog(
+                f"rank {self.rank} epoch {epoch}: received "
+                f"{int(recv.shape[0])} rows"
+            )
             rows_per_trainer = {self.rank: recv}
         elif self.num_trainers == 1:
             rows_per_trainer = {0: src}
@@ -334,6 +351,11 @@ class ShuffleEngine:
                 torch.cuda.current_stream(self.device).synchronize()
             self.consumer.consume(t, epoch, parts)
             self.consumer.producer_done(t, epoch)
+            _vlog(
+                f"rank {self.rank} epoch {epoch}: queued "
+                f"{len(parts)} partitions for trainer {t} "
+                f"({time.perf_counter() - t0:.3f}s since epoch start)"
+            )
         if self.stats:
             self.stats.epoch_done(epoch, time.perf_counter() - t0)
 
