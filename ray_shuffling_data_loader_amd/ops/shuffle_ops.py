@@ -175,7 +175,7 @@ def unpack_permute(
     n = perm.numel() if perm is not None else packed.shape[0]
     if packed.is_cuda:
         hip = _load_hip()
-        outs, offs, codes, names = [], [], [], []
+        outs, offs, codes = [], [], []
         result: Dict[str, torch.Tensor] = {}
         for spec in schema.columns:
             dst_dt = out_dtypes.get(spec.name, spec.dtype)
@@ -184,7 +184,6 @@ def unpack_permute(
             outs.append(o)
             offs.append(schema.offsets[spec.name])
             codes.append(_dtype_code(spec.dtype))
-            names.append(spec.name)
             result[spec.name] = o
         hip.unpack_permute(packed, perm, outs, offs, codes)
         return result

@@ -208,6 +208,7 @@ def _col_tensor(block, col) -> torch.Tensor:
             )
     return torch.as_tensor(np.ascontiguousarray(column))
 
+
 def convert_to_tensor(
     block,
     feature_columns: List[Any],
