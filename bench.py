@@ -236,10 +236,29 @@ def main():
     sync()
     barrier()
     sync()
+    prof = None
+    if os.environ.get("RSDL_TRACE_FIRST") == "1" and rank == 0:
+        prof = torch.profiler.profile(
+            activities=[
+                torch.profiler.ProfilerActivity.CPU,
+                torch.profiler.ProfilerActivity.CUDA,
+            ],
+            with_stack=False,
+        )
+        prof.__enter__()
     t0 = time.perf_counter()
     waits = []
-    for _ in range(args.steps):
+    for i in range(args.steps):
         waits.append(one_step())
+        if prof is not None and i == 2:
+            prof.__exit__(None, None, None)
+            print(
+                prof.key_averages().table(
+                    sort_by="self_cpu_time_total", row_limit=18
+                ),
+                flush=True,
+            )
+            prof = None
     sync()
     barrier()
     sync()
