@@ -82,7 +82,7 @@ __device__ __forceinline__ void load_stage_fast(
 }
 
 __device__ void load_stage_guarded(
-    const short* __restrict__ dy, const short* __restrict__ x, int64_t M,
+    const short* __restrict__ dy, const short* __restrict__ x,
     int32_t N, int32_t K, int32_t n0, int32_t k0, int64_t m0, int64_t m_hi,
     int32_t tid, StageRegs& r) {
   const int32_t c0 = (tid & 15) * 4;
@@ -121,7 +121,7 @@ __device__ __forceinline__ void load_stage(
   if (full_nk && m0 + MT <= m_hi) {
     load_stage_fast(dy, x, N, K, n0, k0, m0, tid, r);
   } else {
-    load_stage_guarded(dy, x, M, N, K, n0, k0, m0, m_hi, tid, r);
+    load_stage_guarded(dy, x, N, K, n0, k0, m0, m_hi, tid, r);
   }
 }
 
