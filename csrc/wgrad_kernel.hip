@@ -114,7 +114,7 @@ __device__ void load_stage_guarded(
 }
 
 __device__ __forceinline__ void load_stage(
-    const short* __restrict__ dy, const short* __restrict__ x, int64_t M,
+    const short* __restrict__ dy, const short* __restrict__ x,
     int32_t N, int32_t K, int32_t n0, int32_t k0, int64_t m0, int64_t m_hi,
     int32_t tid, StageRegs& r, bool full_nk) {
   // Wave-uniform branch: interior stages take the unguarded vector path.
@@ -180,11 +180,10 @@ wgrad_bf16_kernel(const short* __restrict__ dy,  // [M, N] bf16 bits
   StageRegs r0, r1;
   int buf = 0;
   if (m_lo < m_hi) {
-    load_stage(dy, x, M, N, K, n0, k0, m_lo, m_hi, tid, r0, full_nk);
+    load_stage(dy, x, N, K, n0, k0, m_lo, m_hi, tid, r0, full_nk);
   }
   if (m_lo + MT < m_hi) {
-    load_stage(dy, x, M, N, K, n0, k0, m_lo + MT, m_hi, tid, r1,
-               full_nk);
+    load_stage(dy, x, N, K, n0, k0, m_lo + MT, m_hi, tid, r1, full_nk);
   }
 
   auto compute_stage = [&](int32_t b) {
@@ -216,8 +215,7 @@ wgrad_bf16_kernel(const short* __restrict__ dy,  // [M, N] bf16 bits
     write_stage(lds[buf][0], lds[buf][1], tid, r0);
     __syncthreads();
     if (m0 + 2 * MT < m_hi) {
-      load_stage(dy, x, M, N, K, n0, k0, m0 + 2 * MT, m_hi, tid, r0,
-                 full_nk);
+      load_stage(dy, x, N, K, n0, k0, m0 + 2 * MT, m_hi, tid, r0, full_nk);
     }
     compute_stage(buf);
     buf ^= 1;
@@ -227,7 +225,7 @@ wgrad_bf16_kernel(const short* __restrict__ dy,  // [M, N] bf16 bits
       write_stage(lds[buf][0], lds[buf][1], tid, r1);
       __syncthreads();
       if (m0 + 3 * MT < m_hi) {
-        load_stage(dy, x, M, N, K, n0, k0, m0 + 3 * MT, m_hi, tid, r1,
+        load_stage(dy, x, N, K, n0, k0, m0 + 3 * MT, m_hi, tid, r1,
                    full_nk);
       }
       compute_stage(buf);
