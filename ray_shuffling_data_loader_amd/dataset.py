@@ -95,6 +95,7 @@ class ShufflingDataset:
         num_reducers: Optional[int] = None,
         max_concurrent_epochs: int = 2,
         queue_name: str = BATCHQUEUE_ACTOR_NAME,
+        start_epoch: int = 0,
         **engine_kwargs,
     ):
         if num_reducers is None:
@@ -137,6 +138,7 @@ class ShufflingDataset:
                 num_reducers=num_reducers,
                 num_trainers=num_trainers,
                 rank=rank,
+                start_epoch=start_epoch,
                 **engine_kwargs,
             )
             self._engine.start()
@@ -161,6 +163,7 @@ class ShufflingDataset:
                 num_reducers=num_reducers,
                 num_trainers=num_trainers,
                 rank=0,
+                start_epoch=start_epoch,
                 **engine_kwargs,
             )
             self._engine.start()

@@ -87,6 +87,7 @@ class ShuffleEngine:
         num_reducers: int,
         num_trainers: int,
         rank: int = 0,
+        start_epoch: int = 0,
         device: Optional[torch.device] = None,
         seed: Optional[int] = None,
         source_cache: str = "auto",  # "auto" | "device" | "host" | "none"
@@ -101,6 +102,7 @@ class ShuffleEngine:
         self.filenames = list(filenames)
         self.consumer = consumer
         self.num_epochs = num_epochs
+        self.start_epoch = start_epoch
         self.num_reducers = num_reducers
         self.num_trainers = num_trainers
         self.group = group
@@ -180,13 +182,11 @@ class ShuffleEngine:
 
         if seed is None:
             seed = int(time.time_ns() % (2**31))
-        # Per-rank independent stream; ranks draw iid assignments.
-        gen_seed = seed * 1000003 + self.rank
+        self.seed = seed
         if self.device.type == "cuda":
             self._gen = torch.Generator(device=self.device)
         else:
             self._gen = torch.Generator()
-        self._gen.manual_seed(gen_seed)
 
         self._cached_source: Optional[torch.Tensor] = None
         self._stream = (
@@ -271,7 +271,16 @@ class ShuffleEngine:
         )
         return RowBlock(cols)
 
+    def _epoch_seed(self, epoch: int) -> int:
+        # Deterministic per-(seed, rank, epoch) stream: epoch k's shuffle is
+        # independent of which epochs ran before it, so a resumed run
+        # (start_epoch=k, same seed) reproduces exactly the shuffles the
+        # original run would have produced. (The reference has no resume
+        # support at all — SURVEY §5 checkpoint/resume: none.)
+        return (self.seed * 1000003 + self.rank) ^ (epoch * 0x9E3779B1)
+
     def _shuffle_epoch(self, epoch: int) -> None:
+        self._gen.manual_seed(self._epoch_seed(epoch) % (2**63))
         if self.stats:
             self.stats.epoch_start(epoch)
         t0 = time.perf_counter()
@@ -371,7 +380,7 @@ class ShuffleEngine:
             else _nullcontext()
         )
         with ctx:
-            for epoch in range(self.num_epochs):
+            for epoch in range(self.start_epoch, self.num_epochs):
                 t_gate = time.perf_counter()
                 self.consumer.wait_until_ready(epoch)
                 if self.stats:
@@ -397,7 +406,7 @@ class ShuffleEngine:
                 self._error = e
                 failure = ShuffleEngineFailure(e)
                 try:
-                    for epoch in range(self.num_epochs):
+                    for epoch in range(self.start_epoch, self.num_epochs):
                         for t in self.owned_trainers:
                             self.consumer.consume(t, epoch, [failure])
                             self.consumer.producer_done(t, epoch)
