@@ -375,6 +375,9 @@ std::vector<at::Tensor> wgrad_bf16(const at::Tensor& dy, const at::Tensor& x,
   int32_t tiles = ((N + 63) / 64) * ((K + 63) / 64);
   int32_t split = (int32_t)std::min<int64_t>(
       std::max<int64_t>(1, 2048 / tiles), std::max<int64_t>(1, M / 64));
+  if (const char* ov = getenv("RSDL_WGRAD_SPLIT")) {
+    split = std::max(1, atoi(ov));
+  }
   if (M > 0) {
     launch_wgrad_bf16(dy.data_ptr(), x.data_ptr(), dW.data_ptr<float>(),
                       with_bias ? db.data_ptr<float>() : nullptr, M, N, K,
