@@ -377,3 +377,42 @@ def test_actor_alias():
     assert q.actor is not None
     q.shutdown()
     assert q.actor is None
+
+
+def test_torture_many_producers_consumers():
+    # 4 producers x 4 consumers hammering one sub-queue with maxsize
+    # backpressure; exact item conservation.
+    q = make_queue(maxsize=8)
+    n_per = 500
+    produced = 4 * n_per
+    consumed = []
+    lock = threading.Lock()
+
+    def producer(pid):
+        for i in range(n_per):
+            q.put(rank=0, epoch=0, item=(pid, i))
+
+    def consumer():
+        while True:
+            item = q.get(rank=0, epoch=0, timeout=5)
+            if item is None:
+                return
+            with lock:
+                consumed.append(item)
+
+    producers = [
+        threading.Thread(target=producer, args=(p,)) for p in range(4)
+    ]
+    consumers = [threading.Thread(target=consumer) for _ in range(4)]
+    for t in consumers + producers:
+        t.start()
+    for t in producers:
+        t.join(timeout=30)
+        assert not t.is_alive()
+    for _ in consumers:
+        q.put(rank=0, epoch=0, item=None)  # poison pills
+    for t in consumers:
+        t.join(timeout=30)
+        assert not t.is_alive()
+    assert len(consumed) == produced
+    assert len(set(consumed)) == produced
