@@ -284,3 +284,37 @@ def test_same_seed_reproducible(small_data):
         ds.set_epoch(0)
         runs.append(torch.cat([b["key"] for b in ds]))
     assert torch.equal(runs[0], runs[1])
+
+
+def test_output_modes_equivalent(float_data):
+    # views vs columns output paths must yield identical rows per epoch.
+    filenames, num_rows = float_data
+    keys_by_mode = {}
+    for mode in ("views", "columns"):
+        ds = ShufflingDataset(
+            filenames, 1, num_trainers=1, batch_size=1000, rank=0,
+            num_reducers=2, seed=42, output=mode,
+        )
+        ds.set_epoch(0)
+        blocks = list(ds)
+        total = sum(len(b) for b in blocks)
+        assert total == num_rows
+        # deterministic seed -> identical shuffled f0 stream across modes
+        keys_by_mode[mode] = torch.cat([b["f0"] for b in blocks])
+    assert torch.equal(keys_by_mode["views"], keys_by_mode["columns"])
+
+
+def test_out_dtypes_cpu(float_data):
+    filenames, num_rows = float_data
+    ds = ShufflingDataset(
+        filenames, 1, num_trainers=1, batch_size=1000, rank=0,
+        num_reducers=2,
+        feature_matrix=("__features__", [f"f{i}" for i in range(8)]),
+        out_dtypes={"__features__": torch.bfloat16},
+    )
+    ds.set_epoch(0)
+    total = 0
+    for b in ds:
+        assert b["__features__"].dtype == torch.bfloat16
+        total += len(b)
+    assert total == num_rows
