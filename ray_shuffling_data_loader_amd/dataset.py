@@ -184,12 +184,6 @@ class ShufflingDataset:
         iterator each epoch (reference dataset.py:96-106)."""
         self._epoch = epoch
 
-    def __len__(self):
-        raise NotImplementedError(
-            "ShufflingDataset is a streaming dataset with binomial partition "
-            "sizes; its length is not known up front."
-        )
-
     def __iter__(self):
         """Yields RowBlock batches of ``batch_size`` rows from the shuffle
         queue (reference dataset.py:108-188)."""
