@@ -346,7 +346,8 @@ std::vector<at::Tensor> wgrad_bf16(const at::Tensor& dy, const at::Tensor& x,
   int64_t M = dy.size(0);
   int32_t N = (int32_t)dy.size(1);
   int32_t K = (int32_t)x.size(1);
-  bool wide = N >= 64 && K >= 64 && M >= 4096;
+  bool wide = getenv("RSDL_WGRAD_WIDE") != nullptr && N >= 64 && K >= 64 &&
+              M >= 4096;
   if (wide) {
     // Wide-tile kernel wants 128-multiples; pad (cheap relative to the
     // re-read amplification it removes) and narrow the result.
@@ -369,20 +370,33 @@ std::vector<at::Tensor> wgrad_bf16(const at::Tensor& dy, const at::Tensor& x,
     auto db = with_bias ? dbp.narrow(0, 0, N).contiguous() : at::Tensor();
     return {dW, db};
   }
-  auto dW = at::zeros({N, K}, dy.options().dtype(at::kFloat));
-  auto db = with_bias ? at::zeros({N}, dy.options().dtype(at::kFloat))
-                      : at::Tensor();
-  int32_t tiles = ((N + 63) / 64) * ((K + 63) / 64);
+  // 64-tile kernel: pad N/K to 64-multiples so the hot loop runs only the
+  // unguarded vector-load stages (per-element guards serialize loads).
+  auto dyp = (N % 64) ? at::constant_pad_nd(dy, {0, 64 - N % 64})
+                      : dy.contiguous();
+  auto xp = (K % 64) ? at::constant_pad_nd(x, {0, 64 - K % 64})
+                     : x.contiguous();
+  int32_t Np = (int32_t)dyp.size(1), Kp = (int32_t)xp.size(1);
+  auto dWp = at::zeros({Np, Kp}, dy.options().dtype(at::kFloat));
+  auto dbp = with_bias ? at::zeros({Np}, dy.options().dtype(at::kFloat))
+                       : at::Tensor();
+  int32_t tiles = (Np / 64) * (Kp / 64);
   int32_t split = (int32_t)std::min<int64_t>(
       std::max<int64_t>(1, 2048 / tiles), std::max<int64_t>(1, M / 64));
   if (const char* ov = getenv("RSDL_WGRAD_SPLIT")) {
     split = std::max(1, atoi(ov));
   }
   if (M > 0) {
-    launch_wgrad_bf16(dy.data_ptr(), x.data_ptr(), dW.data_ptr<float>(),
-                      with_bias ? db.data_ptr<float>() : nullptr, M, N, K,
+    launch_wgrad_bf16(dyp.data_ptr(), xp.data_ptr(), dWp.data_ptr<float>(),
+                      with_bias ? dbp.data_ptr<float>() : nullptr, M, Np, Kp,
                       split, current_stream());
   }
+  auto dW = (Np == N && Kp == K)
+                ? dWp
+                : dWp.narrow(0, 0, N).narrow(1, 0, K).contiguous();
+  auto db = with_bias
+                ? ((Np == N) ? dbp : dbp.narrow(0, 0, N).contiguous())
+                : at::Tensor();
   return {dW, db};
 }
 

@@ -60,23 +60,27 @@ __device__ __forceinline__ void load_stage_fast(
     StageRegs& r) {
   const int32_t c0 = (tid & 15) * 4;
   const int32_t rp = (tid >> 4) * 2;
+  // Issue all 8 independent 8-B loads up front (separate destination and
+  // address registers), THEN pack: lets the compiler keep every load in
+  // flight with one counted wait instead of serializing on reused regs.
+  bf16x4 d[2][2], xv[2][2];
   #pragma unroll
   for (int p = 0; p < 2; p++) {
     #pragma unroll
     for (int q = 0; q < 2; q++) {
       const int64_t m = m0 + rp + 32 * p + q;
-      const bf16x4 dv = *reinterpret_cast<const bf16x4*>(&dy[m * N + n0 + c0]);
-      const bf16x4 xv = *reinterpret_cast<const bf16x4*>(&x[m * K + k0 + c0]);
-      #pragma unroll
-      for (int j = 0; j < 4; j++) {
-        if (q == 0) {
-          r.dy_v[p * 2][j] = (uint16_t)dv[j];
-          r.x_v[p * 2][j] = (uint16_t)xv[j];
-        } else {
-          r.dy_v[p * 2][j] |= ((uint32_t)(uint16_t)dv[j]) << 16;
-          r.x_v[p * 2][j] |= ((uint32_t)(uint16_t)xv[j]) << 16;
-        }
-      }
+      d[p][q] = *reinterpret_cast<const bf16x4*>(&dy[m * N + n0 + c0]);
+      xv[p][q] = *reinterpret_cast<const bf16x4*>(&x[m * K + k0 + c0]);
+    }
+  }
+  #pragma unroll
+  for (int p = 0; p < 2; p++) {
+    #pragma unroll
+    for (int j = 0; j < 4; j++) {
+      r.dy_v[p * 2][j] = (uint32_t)(uint16_t)d[p][0][j] |
+                         (((uint32_t)(uint16_t)d[p][1][j]) << 16);
+      r.x_v[p * 2][j] = (uint32_t)(uint16_t)xv[p][0][j] |
+                        (((uint32_t)(uint16_t)xv[p][1][j]) << 16);
     }
   }
 }
@@ -110,18 +114,6 @@ __device__ void load_stage_guarded(
         }
       }
     }
-  }
-}
-
-__device__ __forceinline__ void load_stage(
-    const short* __restrict__ dy, const short* __restrict__ x,
-    int32_t N, int32_t K, int32_t n0, int32_t k0, int64_t m0, int64_t m_hi,
-    int32_t tid, StageRegs& r, bool full_nk) {
-  // Wave-uniform branch: interior stages take the unguarded vector path.
-  if (full_nk && m0 + MT <= m_hi) {
-    load_stage_fast(dy, x, N, K, n0, k0, m0, tid, r);
-  } else {
-    load_stage_guarded(dy, x, N, K, n0, k0, m0, m_hi, tid, r);
   }
 }
 
@@ -177,13 +169,21 @@ wgrad_bf16_kernel(const short* __restrict__ dy,  // [M, N] bf16 bits
   // runtime index would push the arrays to scratch): the loop body is
   // unrolled over two stages, r0/r1 alternating; loads for stage s issue
   // at stage s-2 (1-deep measured 73% SQ_WAIT_ANY).
+  // Stage plan: n_full full stages run the UNGUARDED vector loads only
+  // (per-element guard branches in the hot loop serialize every load with
+  // a vmcnt(0) — seen in the disassembly); at most one guarded tail stage
+  // is peeled to the end.
+  const int64_t span = m_hi - m_lo;
+  const int64_t n_full = full_nk ? span / MT : 0;
+  const int64_t m_full_end = m_lo + n_full * MT;
+
   StageRegs r0, r1;
   int buf = 0;
-  if (m_lo < m_hi) {
-    load_stage(dy, x, N, K, n0, k0, m_lo, m_hi, tid, r0, full_nk);
+  if (m_lo < m_full_end) {
+    load_stage_fast(dy, x, N, K, n0, k0, m_lo, tid, r0);
   }
-  if (m_lo + MT < m_hi) {
-    load_stage(dy, x, N, K, n0, k0, m_lo + MT, m_hi, tid, r1, full_nk);
+  if (m_lo + MT < m_full_end) {
+    load_stage_fast(dy, x, N, K, n0, k0, m_lo + MT, tid, r1);
   }
 
   auto compute_stage = [&](int32_t b) {
@@ -210,28 +210,37 @@ wgrad_bf16_kernel(const short* __restrict__ dy,  // [M, N] bf16 bits
     }
   };
 
-  for (int64_t m0 = m_lo; m0 < m_hi; m0 += 2 * MT) {
+  for (int64_t m0 = m_lo; m0 < m_full_end; m0 += 2 * MT) {
     // stage A (r0)
     write_stage(lds[buf][0], lds[buf][1], tid, r0);
     __syncthreads();
-    if (m0 + 2 * MT < m_hi) {
-      load_stage(dy, x, N, K, n0, k0, m0 + 2 * MT, m_hi, tid, r0, full_nk);
+    if (m0 + 2 * MT < m_full_end) {
+      load_stage_fast(dy, x, N, K, n0, k0, m0 + 2 * MT, tid, r0);
     }
     compute_stage(buf);
     buf ^= 1;
     __syncthreads();
     // stage B (r1)
-    if (m0 + MT < m_hi) {
+    if (m0 + MT < m_full_end) {
       write_stage(lds[buf][0], lds[buf][1], tid, r1);
       __syncthreads();
-      if (m0 + 3 * MT < m_hi) {
-        load_stage(dy, x, N, K, n0, k0, m0 + 3 * MT, m_hi, tid, r1,
-                   full_nk);
+      if (m0 + 3 * MT < m_full_end) {
+        load_stage_fast(dy, x, N, K, n0, k0, m0 + 3 * MT, tid, r1);
       }
       compute_stage(buf);
       buf ^= 1;
       __syncthreads();
     }
+  }
+  // Peeled guarded tail (runs at most ceil((span - n_full*MT)/MT) stages;
+  // only edge WGs or the last m-chunk reach it).
+  for (int64_t m0 = m_full_end; m0 < m_hi; m0 += MT) {
+    load_stage_guarded(dy, x, N, K, n0, k0, m0, m_hi, tid, r0);
+    write_stage(lds[buf][0], lds[buf][1], tid, r0);
+    __syncthreads();
+    compute_stage(buf);
+    buf ^= 1;
+    __syncthreads();
   }
 
   #pragma unroll
