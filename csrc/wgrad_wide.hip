@@ -48,6 +48,7 @@ __device__ __forceinline__ void w_load(const short* __restrict__ src,
   // unit u = g*W_THREADS + tid covers row-pair pr = u / (C/4),
   // cols (u % (C/4))*4. For C*WMT/8 < W_THREADS the tail threads idle
   // (wave-uniform for the shapes used).
+  w_bf16x4 lo[WG_GROUPS(C)], hi[WG_GROUPS(C)];
   #pragma unroll
   for (int g = 0; g < WG_GROUPS(C); g++) {
     const int32_t u = g * W_THREADS + tid;
@@ -55,14 +56,18 @@ __device__ __forceinline__ void w_load(const short* __restrict__ src,
     const int32_t c4 = u % (C / 4);
     const int32_t pr = u / (C / 4);
     const int64_t m = m0 + pr * 2;
-    const w_bf16x4 lo =
-        *reinterpret_cast<const w_bf16x4*>(&src[m * ld + c_base + c4 * 4]);
-    const w_bf16x4 hi = *reinterpret_cast<const w_bf16x4*>(
+    lo[g] = *reinterpret_cast<const w_bf16x4*>(&src[m * ld + c_base + c4 * 4]);
+    hi[g] = *reinterpret_cast<const w_bf16x4*>(
         &src[(m + 1) * ld + c_base + c4 * 4]);
+  }
+  #pragma unroll
+  for (int g = 0; g < WG_GROUPS(C); g++) {
+    const int32_t u = g * W_THREADS + tid;
+    if (u >= (C / 4) * (WMT / 2)) break;
     #pragma unroll
     for (int j = 0; j < 4; j++) {
-      r.v[g][j] = (uint32_t)(uint16_t)lo[j] |
-                  (((uint32_t)(uint16_t)hi[j]) << 16);
+      r.v[g][j] = (uint32_t)(uint16_t)lo[g][j] |
+                  (((uint32_t)(uint16_t)hi[g][j]) << 16);
     }
   }
 }
@@ -139,27 +144,20 @@ wgrad_wide_kernel(const short* __restrict__ dy, const short* __restrict__ x,
   float bias_acc[WN / 32] = {};
   const bool do_bias = (db != nullptr) && (k0 == 0) && (wk == 0);
 
+  const int64_t span = m_hi - m_lo;
+  const int64_t n_full = span / WMT;
+  const int64_t m_full_end = m_lo + n_full * WMT;
+
   WStage<NW> dy0, dy1;
   WStage<KW> x0, x1;
   int buf = 0;
-  const int64_t m1 = m_lo + WMT;
-  if (m_lo < m_hi) {
-    if (m_lo + WMT <= m_hi) {
-      w_load<NW>(dy, N, n0, m_lo, tid, dy0);
-      w_load<KW>(x, K, k0, m_lo, tid, x0);
-    } else {
-      w_load_guarded<NW>(dy, N, n0, m_lo, m_hi, tid, dy0);
-      w_load_guarded<KW>(x, K, k0, m_lo, m_hi, tid, x0);
-    }
+  if (m_lo < m_full_end) {
+    w_load<NW>(dy, N, n0, m_lo, tid, dy0);
+    w_load<KW>(x, K, k0, m_lo, tid, x0);
   }
-  if (m1 < m_hi) {
-    if (m1 + WMT <= m_hi) {
-      w_load<NW>(dy, N, n0, m1, tid, dy1);
-      w_load<KW>(x, K, k0, m1, tid, x1);
-    } else {
-      w_load_guarded<NW>(dy, N, n0, m1, m_hi, tid, dy1);
-      w_load_guarded<KW>(x, K, k0, m1, m_hi, tid, x1);
-    }
+  if (m_lo + WMT < m_full_end) {
+    w_load<NW>(dy, N, n0, m_lo + WMT, tid, dy1);
+    w_load<KW>(x, K, k0, m_lo + WMT, tid, x1);
   }
 
   auto compute = [&](int32_t b) {
@@ -200,41 +198,40 @@ wgrad_wide_kernel(const short* __restrict__ dy, const short* __restrict__ x,
     }
   };
 
-  for (int64_t m0 = m_lo; m0 < m_hi; m0 += 2 * WMT) {
+  for (int64_t m0 = m_lo; m0 < m_full_end; m0 += 2 * WMT) {
     w_write<NW>(dyT[buf], tid, dy0);
     w_write<KW>(xT[buf], tid, x0);
     __syncthreads();
-    const int64_t mn = m0 + 2 * WMT;
-    if (mn < m_hi) {
-      if (mn + WMT <= m_hi) {
-        w_load<NW>(dy, N, n0, mn, tid, dy0);
-        w_load<KW>(x, K, k0, mn, tid, x0);
-      } else {
-        w_load_guarded<NW>(dy, N, n0, mn, m_hi, tid, dy0);
-        w_load_guarded<KW>(x, K, k0, mn, m_hi, tid, x0);
-      }
+    if (m0 + 2 * WMT < m_full_end) {
+      w_load<NW>(dy, N, n0, m0 + 2 * WMT, tid, dy0);
+      w_load<KW>(x, K, k0, m0 + 2 * WMT, tid, x0);
     }
     compute(buf);
     buf ^= 1;
     __syncthreads();
-    if (m0 + WMT < m_hi) {
+    if (m0 + WMT < m_full_end) {
       w_write<NW>(dyT[buf], tid, dy1);
       w_write<KW>(xT[buf], tid, x1);
       __syncthreads();
-      const int64_t mn2 = m0 + 3 * WMT;
-      if (mn2 < m_hi) {
-        if (mn2 + WMT <= m_hi) {
-          w_load<NW>(dy, N, n0, mn2, tid, dy1);
-          w_load<KW>(x, K, k0, mn2, tid, x1);
-        } else {
-          w_load_guarded<NW>(dy, N, n0, mn2, m_hi, tid, dy1);
-          w_load_guarded<KW>(x, K, k0, mn2, m_hi, tid, x1);
-        }
+      if (m0 + 3 * WMT < m_full_end) {
+        w_load<NW>(dy, N, n0, m0 + 3 * WMT, tid, dy1);
+        w_load<KW>(x, K, k0, m0 + 3 * WMT, tid, x1);
       }
       compute(buf);
       buf ^= 1;
       __syncthreads();
     }
+  }
+  // Peeled guarded tail (last partial stage of the last chunk only).
+  for (int64_t m0 = m_full_end; m0 < m_hi; m0 += WMT) {
+    w_load_guarded<NW>(dy, N, n0, m0, m_hi, tid, dy0);
+    w_load_guarded<KW>(x, K, k0, m0, m_hi, tid, x0);
+    w_write<NW>(dyT[buf], tid, dy0);
+    w_write<KW>(xT[buf], tid, x0);
+    __syncthreads();
+    compute(buf);
+    buf ^= 1;
+    __syncthreads();
   }
 
   #pragma unroll
