@@ -25,7 +25,7 @@ typedef __attribute__((__vector_size__(8 * sizeof(short)))) short w_bf16x8;
 typedef __attribute__((__vector_size__(4 * sizeof(short)))) short w_bf16x4;
 typedef __attribute__((__vector_size__(16 * sizeof(float)))) float w_f32x16;
 
-#define WMT 16
+#define WMT 32
 #define WLDS_STRIDE (WMT + 8)  // 24 halfwords = 48 B per [col] row
 
 // Per-operand stage registers: operand with C columns contributes
