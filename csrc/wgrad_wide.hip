@@ -163,37 +163,43 @@ wgrad_wide_kernel(const short* __restrict__ dy, const short* __restrict__ x,
   auto compute = [&](int32_t b) {
     const short* dT = dyT[b];
     const short* xTb = xT[b];
-    const int32_t mfrag = (lane >> 5) * 8;
-    w_bf16x8 afr[WN / 32];
-    w_bf16x8 bfr[WK / 32];
     #pragma unroll
-    for (int ai = 0; ai < WN / 32; ai++) {
-      *reinterpret_cast<uint4*>(&afr[ai]) = *reinterpret_cast<const uint4*>(
-          &dT[(wn + ai * 32 + (lane & 31)) * WLDS_STRIDE + mfrag]);
-    }
-    #pragma unroll
-    for (int bi = 0; bi < WK / 32; bi++) {
-      *reinterpret_cast<uint4*>(&bfr[bi]) = *reinterpret_cast<const uint4*>(
-          &xTb[(wk + bi * 32 + (lane & 31)) * WLDS_STRIDE + mfrag]);
-    }
-    if (do_bias) {
+    for (int32_t ms = 0; ms < WMT; ms += 16) {
+      const int32_t mfrag = ms + (lane >> 5) * 8;
+      w_bf16x8 afr[WN / 32];
+      w_bf16x8 bfr[WK / 32];
+      #pragma unroll
+      for (int ai = 0; ai < WN / 32; ai++) {
+        *reinterpret_cast<uint4*>(&afr[ai]) =
+            *reinterpret_cast<const uint4*>(
+                &dT[(wn + ai * 32 + (lane & 31)) * WLDS_STRIDE + mfrag]);
+      }
+      #pragma unroll
+      for (int bi = 0; bi < WK / 32; bi++) {
+        *reinterpret_cast<uint4*>(&bfr[bi]) =
+            *reinterpret_cast<const uint4*>(
+                &xTb[(wk + bi * 32 + (lane & 31)) * WLDS_STRIDE + mfrag]);
+      }
+      if (do_bias) {
+        #pragma unroll
+        for (int ai = 0; ai < WN / 32; ai++) {
+          #pragma unroll
+          for (int j = 0; j < 8; j++) {
+            __hip_bfloat16 h;
+            short sv = afr[ai][j];
+            *reinterpret_cast<short*>(&h) = sv;
+            bias_acc[ai] += __bfloat162float(h);
+          }
+        }
+      }
       #pragma unroll
       for (int ai = 0; ai < WN / 32; ai++) {
         #pragma unroll
-        for (int j = 0; j < 8; j++) {
-          __hip_bfloat16 h;
-          short sv = afr[ai][j];
-          *reinterpret_cast<short*>(&h) = sv;
-          bias_acc[ai] += __bfloat162float(h);
+        for (int bi = 0; bi < WK / 32; bi++) {
+          acc[ai * (WK / 32) + bi] =
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                  afr[ai], bfr[bi], acc[ai * (WK / 32) + bi], 0, 0, 0);
         }
-      }
-    }
-    #pragma unroll
-    for (int ai = 0; ai < WN / 32; ai++) {
-      #pragma unroll
-      for (int bi = 0; bi < WK / 32; bi++) {
-        acc[ai * (WK / 32) + bi] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            afr[ai], bfr[bi], acc[ai * (WK / 32) + bi], 0, 0, 0);
       }
     }
   };
