@@ -3,7 +3,7 @@
 loader feeding a real DDP training step.
 
 BASELINE.json metric: "shuffled rows/sec (whole node) + p50 batch-wait,
-1e8 x 100 float cols, 8 trainers". Weak scaling: 1.25e7 rows x 100 float32
+1e8 x 100 float cols, 8 trainers". Weak scaling: 1.25e7 rows x 100 float64
 cols per GPU (N=8 => the named 1e8-row config), batch_size 250k, synthetic
 Parquet (no network; generated locally on first run), random-init TabularMLP
 with genuine fwd+bwd+optimizer in every timed step (the reference example
@@ -313,7 +313,7 @@ def main():
             "vs_baseline": None,
             "dtype": args.dtype,
             "data": (
-                "synthetic parquet, generated locally; "
+                "synthetic parquet (float64 source cols), generated locally; "
                 f"cached {'in HBM' if args.source_cache in ('auto', 'device') else args.source_cache} "
                 "after first read; full per-epoch reshuffle "
                 "(assignment+all-to-all+permute) every epoch"

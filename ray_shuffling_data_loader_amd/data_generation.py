@@ -42,11 +42,16 @@ DATA_SPEC = {
 }
 
 
-def float_data_spec(num_features: int = 100) -> Dict[str, Tuple]:
-    """The flagship benchmark shape: ``num_features`` float32 feature columns
-    + one float32 label column (BASELINE.json: '1e8 rows x 100 float cols')."""
-    spec = {f"f{i}": (0.0, 1.0, np.float32) for i in range(num_features)}
-    spec["labels"] = (0.0, 1.0, np.float32)
+def float_data_spec(
+    num_features: int = 100, dtype=np.float64
+) -> Dict[str, Tuple]:
+    """The flagship benchmark shape: ``num_features`` float feature columns
+    + one float label column (BASELINE.json: '1e8 rows x 100 float cols').
+    Default float64 source columns to match the reference's DATA_SPEC
+    precision (reference data_generation.py DATA_SPEC: float64 features);
+    the fused unpack kernel casts fp64 -> fp32/bf16 on the GPU."""
+    spec = {f"f{i}": (0.0, 1.0, dtype) for i in range(num_features)}
+    spec["labels"] = (0.0, 1.0, dtype)
     return spec
 
 
