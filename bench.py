@@ -24,6 +24,7 @@ import json
 import os
 import shutil
 import statistics
+import sys
 import tempfile
 import time
 
@@ -113,7 +114,7 @@ def main():
     total_files = args.files_per_gpu * world
     rows_per_file = args.rows_per_gpu // args.files_per_gpu
     shard_dir = os.path.join(
-        data_dir, f"w{world}_{args.rows_per_gpu}x{args.num_cols}"
+        data_dir, f"w{world}_{args.rows_per_gpu}x{args.num_cols}f64"
     )
     os.makedirs(shard_dir, exist_ok=True)
     filenames = [
@@ -146,11 +147,13 @@ def main():
                     missing,
                 )
             )
+        # stderr: stdout carries exactly ONE JSON line (driver contract).
         print(
             f"[bench] rank {rank} generated {len(missing)} files "
             f"({len(missing) * rows_per_file} rows) in "
             f"{time.perf_counter() - t0:.1f}s",
             flush=True,
+            file=sys.stderr,
         )
     if world > 1:
         torch.distributed.barrier()
