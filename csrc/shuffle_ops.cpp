@@ -65,6 +65,9 @@ void launch_wgrad_bf16(const void* dy, const void* x, float* dW, float* db,
 void launch_wgrad_wide(const void* dy, const void* x, float* dW, float* db,
                        int64_t M, int32_t N, int32_t K, int32_t split_m,
                        hipStream_t stream);
+void launch_relu_bwd_bias(const void* dy, const void* y, void* dx,
+                          float* db_part, int64_t nvec, int32_t N,
+                          int32_t grid, hipStream_t stream);
 
 namespace {
 
@@ -400,6 +403,40 @@ std::vector<at::Tensor> wgrad_bf16(const at::Tensor& dy, const at::Tensor& x,
   return {dW, db};
 }
 
+
+// Fused ReLU-backward + bias gradient: dx = dy * (y > 0), db = dx.sum(0).
+// One pass over HBM instead of threshold_backward + a separate column
+// reduce (csrc/relu_bwd.hip). Requires bf16, contiguous, N a power of two
+// >= 8 (the LinearReLU widths); callers fall back to torch otherwise.
+std::vector<at::Tensor> relu_bwd_bias(const at::Tensor& dy,
+                                      const at::Tensor& y) {
+  TORCH_CHECK(dy.is_cuda() && y.is_cuda(), "relu_bwd_bias: inputs on GPU");
+  TORCH_CHECK(dy.scalar_type() == at::kBFloat16 &&
+                  y.scalar_type() == at::kBFloat16,
+              "relu_bwd_bias: bf16 inputs only");
+  TORCH_CHECK(dy.dim() == 2 && dy.sizes() == y.sizes(),
+              "relu_bwd_bias: dy and y must be equal 2-D shapes");
+  TORCH_CHECK(dy.is_contiguous() && y.is_contiguous(),
+              "relu_bwd_bias: contiguous inputs");
+  const int64_t M = dy.size(0);
+  const int64_t N = dy.size(1);
+  TORCH_CHECK(N >= 8 && N <= 2048 && (N & (N - 1)) == 0,
+              "relu_bwd_bias: N must be a power of two in [8, 2048]");
+  auto dx = at::empty_like(dy);
+  const int64_t nvec = M * N / 8;
+  int32_t grid = (int32_t)std::min<int64_t>(
+      1024, std::max<int64_t>(1, nvec / (256 * 8)));
+  auto db_part = at::empty({grid, N}, dy.options().dtype(at::kFloat));
+  if (nvec > 0) {
+    launch_relu_bwd_bias(dy.data_ptr(), y.data_ptr(), dx.data_ptr(),
+                         db_part.data_ptr<float>(), nvec, (int32_t)N, grid,
+                         current_stream());
+  } else {
+    db_part.zero_();
+  }
+  return {dx, db_part.sum(0)};
+}
+
 }  // namespace
 }  // namespace rsdl
 
@@ -422,6 +459,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("num_dests"));
   m.def("wgrad_bf16", &rsdl::wgrad_bf16, py::arg("dy"), py::arg("x"),
         py::arg("with_bias") = true);
+  m.def("relu_bwd_bias", &rsdl::relu_bwd_bias, py::arg("dy"), py::arg("y"));
   m.attr("DT_F32") = (int)rsdl::DT_F32;
   m.attr("DT_F64") = (int)rsdl::DT_F64;
   m.attr("DT_I32") = (int)rsdl::DT_I32;

@@ -432,3 +432,50 @@ def test_collective_path_single_gpu(dev, tmp_path, monkeypatch):
     finally:
         dist.destroy_process_group()
         fabric._shuffle_group = None
+
+
+@pytest.mark.gpu
+def test_relu_bwd_bias_matches_reference():
+    """Fused relu-bwd+bias kernel vs the plain fp32 torch oracle."""
+    from ray_shuffling_data_loader_amd.ops import shuffle_ops
+
+    hip = shuffle_ops._load_hip()
+    torch.manual_seed(7)
+    for m, n in [(250_000, 512), (250_000, 256), (1 << 16, 128),
+                 (12_345, 64), (3, 8)]:
+        y32 = torch.randn(m, n, device="cuda") - 0.3   # mix of dead units
+        dy32 = torch.randn(m, n, device="cuda")
+        y = y32.bfloat16().contiguous()
+        dy = dy32.bfloat16().contiguous()
+        dx, db = hip.relu_bwd_bias(dy, y)
+        ref_dx = torch.ops.aten.threshold_backward(
+            dy.float(), y.float(), 0
+        )
+        ref_db = ref_dx.sum(0)
+        assert torch.equal(dx.float(), ref_dx.bfloat16().float()), (m, n)
+        # db accumulated in fp32 from bf16 inputs — tight tolerance.
+        assert torch.allclose(db, ref_db, rtol=1e-3, atol=1e-1 * m / 1e4), (
+            m, n, (db - ref_db).abs().max().item())
+
+
+@pytest.mark.gpu
+def test_linear_relu_backward_fused_parity():
+    """LinearReLU grads with the fused relu-bwd path vs plain autograd."""
+    from ray_shuffling_data_loader_amd.models.mlp import LinearReLU
+
+    torch.manual_seed(3)
+    lin = LinearReLU(64, 128).cuda()
+    ref = torch.nn.Sequential(
+        torch.nn.Linear(64, 128), torch.nn.ReLU()
+    ).cuda()
+    with torch.no_grad():
+        ref[0].weight.copy_(lin.weight)
+        ref[0].bias.copy_(lin.bias)
+    x = torch.randn(70_000, 64, device="cuda")
+    with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+        out = lin(x).float().square().mean()
+        out_ref = ref(x).float().square().mean()
+    out.backward()
+    out_ref.backward()
+    for p, q in zip(lin.parameters(), ref.parameters()):
+        assert torch.allclose(p.grad, q.grad, rtol=2e-2, atol=2e-3)
