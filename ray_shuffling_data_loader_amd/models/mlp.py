@@ -56,8 +56,15 @@ def _wgrad_bf16_kernel(dy, x, with_bias):
     return dw.to(dy.dtype), (db.to(dy.dtype) if db is not None else None)
 
 
-def _wgrad_chunks(m: int, target: int = 16) -> int:
-    """Largest divisor of m that is <= target (split-K chunk count)."""
+_WGRAD_CHUNK_TARGET = int(os.environ.get("RSDL_BMM_CHUNKS", "16"))
+
+
+def _wgrad_chunks(m: int, target: int = None) -> int:
+    """Largest divisor of m that is <= target (split-K chunk count).
+    Tune with RSDL_BMM_CHUNKS: chunk count x output tiles must exceed the
+    chip's 256 CUs by a healthy margin (profiles/PERF.md)."""
+    if target is None:
+        target = _WGRAD_CHUNK_TARGET
     for c in range(min(target, m), 0, -1):
         if m % c == 0:
             return c
@@ -84,7 +91,7 @@ class _ChunkedLinearFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, weight = ctx.saved_tensors
         dy = dy.contiguous()
-        dx = dy @ weight
+        dx = dy @ weight if ctx.needs_input_grad[0] else None
         m = x.shape[0]
         # Chunking only pays for wide-output, tall-K wgrads; for skinny
         # outputs (e.g. the final [*,1] head) hipBLASLt's plain mm is fine
@@ -144,7 +151,9 @@ class _LinearReLUFn(torch.autograd.Function):
         x, weight, y = ctx.saved_tensors
         # relu' (+ fused bias grad when the HIP kernel applies).
         dy, db = _relu_bwd_bias(dy, y)
-        dx = dy @ weight
+        # First layer: input is the loader's feature batch (no grad) —
+        # skip the [M,N]x[N,K] dgrad GEMM entirely.
+        dx = dy @ weight if ctx.needs_input_grad[0] else None
         m = x.shape[0]
         if (
             _USE_WGRAD_KERNEL
