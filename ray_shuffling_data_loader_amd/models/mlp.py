@@ -56,7 +56,7 @@ def _wgrad_bf16_kernel(dy, x, with_bias):
     return dw.to(dy.dtype), (db.to(dy.dtype) if db is not None else None)
 
 
-_WGRAD_CHUNK_TARGET = int(os.environ.get("RSDL_BMM_CHUNKS", "16"))
+_WGRAD_CHUNK_TARGET = int(os.environ.get("RSDL_BMM_CHUNKS", "50"))
 
 
 def _wgrad_chunks(m: int, target: int = None) -> int:
