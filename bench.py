@@ -39,7 +39,14 @@ if os.environ.get("RSDL_TUNABLEOP", "1") == "1":
         tempfile.gettempdir(), f"rsdl_tunableop_{os.getpid()}.csv"
     )
     if os.path.exists(_tuned_src):
+        # TunableOp resolves FILENAME with the device ordinal inserted
+        # before the extension (e.g. foo.csv -> foo0.csv) on both read and
+        # write; provide every per-ordinal name so each rank's process
+        # actually LOADS the table instead of silently re-tuning in warmup.
         shutil.copyfile(_tuned_src, _tuned_dst)
+        _stem, _ext = os.path.splitext(_tuned_dst)
+        for _ord in range(8):
+            shutil.copyfile(_tuned_src, f"{_stem}{_ord}{_ext}")
     os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
     os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
     os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", _tuned_dst)
