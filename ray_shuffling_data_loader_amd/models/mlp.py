@@ -91,20 +91,30 @@ class _ChunkedLinearFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, weight = ctx.saved_tensors
         dy = dy.contiguous()
-        dx = dy @ weight if ctx.needs_input_grad[0] else None
         m = x.shape[0]
+        n = dy.shape[1]
+        if n == 1:
+            # Scalar-output head: the "GEMMs" degenerate. dgrad is an outer
+            # product dy[M,1] @ W[1,K] == broadcast multiply (hipBLASLt ran
+            # it at 132 us/step vs ~15 us roofline, profiles/PERF.md), and
+            # wgrad is a column-weighted reduce.
+            dx = dy * weight if ctx.needs_input_grad[0] else None
+            dw = (dy * x).sum(0, keepdim=True)
+            db = dy.sum(0) if ctx.has_bias else None
+            return dx, dw, db
+        dx = dy @ weight if ctx.needs_input_grad[0] else None
         # Chunking only pays for wide-output, tall-K wgrads; for skinny
-        # outputs (e.g. the final [*,1] head) hipBLASLt's plain mm is fine
-        # and the batched kernel is pathological (11.7 ms for N=1 vs 0.13).
+        # outputs hipBLASLt's plain mm is fine and the batched kernel is
+        # pathological (11.7 ms for N=1 vs 0.13).
         c = (
             _wgrad_chunks(m)
-            if min(dy.shape[1], x.shape[1]) >= 32 and m >= 1 << 16
+            if min(n, x.shape[1]) >= 32 and m >= 1 << 16
             else 1
         )
         if c > 1:
             dw = (
                 torch.bmm(
-                    dy.view(c, m // c, dy.shape[1]).transpose(1, 2),
+                    dy.view(c, m // c, n).transpose(1, 2),
                     x.view(c, m // c, x.shape[1]),
                 )
                 .sum(0)
