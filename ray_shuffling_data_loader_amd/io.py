@@ -57,9 +57,21 @@ def fuse_schema(base: Schema, feature_matrix: Optional[Tuple[str, List[str]]]
             raise ValueError(
                 "feature-matrix columns are not contiguous in packed layout"
             )
-    fused_specs = [ColumnSpec(name, dt, len(members))]
-    for c in base.columns:
-        if c.name not in members:
+    # Declare the fused schema in the BASE's packed-layout order (which is
+    # non-increasing in dtype size), collapsing the member run into one
+    # vector column in place: the schema's stable (-size, index) sort then
+    # reproduces the base layout byte-for-byte. Declaring the fused column
+    # first breaks when all columns share a dtype size (e.g. float64
+    # features behind an int64 key: the key must stay at offset 0).
+    numel = sum(base.col(m).numel for m in members)
+    fused_specs = []
+    member_set = set(members)
+    for i in base.packed_order:
+        c = base.columns[i]
+        if c.name in member_set:
+            if not any(fs.name == name for fs in fused_specs):
+                fused_specs.append(ColumnSpec(name, dt, numel))
+        else:
             fused_specs.append(c)
     fused = Schema(fused_specs)
     # The fused layout must reinterpret the SAME bytes.

@@ -96,3 +96,30 @@ def test_schema_rejects_duplicates():
 def test_schema_unsupported_dtype():
     with pytest.raises(TypeError):
         Schema([ColumnSpec("c", torch.complex64)])
+
+
+def test_fuse_schema_uniform_dtype_sizes():
+    """Fusing float64 feature columns behind an int64 key keeps the base
+    packed layout (regression: declaring the fused column first moved it
+    to offset 0 when every column shares a dtype size)."""
+    from ray_shuffling_data_loader_amd.io import fuse_schema
+
+    specs = [ColumnSpec("key", torch.int64)]
+    specs += [ColumnSpec(f"f{i}", torch.float64) for i in range(4)]
+    specs.append(ColumnSpec("labels", torch.float64))
+    base = Schema(specs)
+    fused = fuse_schema(base, ("__features__", [f"f{i}" for i in range(4)]))
+    assert fused.offsets["key"] == base.offsets["key"] == 0
+    assert fused.offsets["__features__"] == base.offsets["f0"] == 8
+    assert fused.offsets["labels"] == base.offsets["labels"]
+    assert fused.row_stride == base.row_stride
+    # fp32 features behind an int64 key (the original working case)
+    specs32 = [ColumnSpec("key", torch.int64)]
+    specs32 += [ColumnSpec(f"f{i}", torch.float32) for i in range(4)]
+    specs32.append(ColumnSpec("labels", torch.float32))
+    base32 = Schema(specs32)
+    fused32 = fuse_schema(
+        base32, ("__features__", [f"f{i}" for i in range(4)])
+    )
+    assert fused32.offsets["__features__"] == base32.offsets["f0"] == 8
+    assert fused32.row_stride == base32.row_stride
