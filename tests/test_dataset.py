@@ -318,3 +318,35 @@ def test_out_dtypes_cpu(float_data):
         assert b["__features__"].dtype == torch.bfloat16
         total += len(b)
     assert total == num_rows
+
+
+def test_torch_dataset_feature_matrix_with_key(tmp_path):
+    """feature_matrix over float64 feature columns packed BEHIND an int64
+    key column (the DDP example's data shape) — regression for the
+    fuse_schema layout bug when all columns share a dtype size."""
+    spec = float_data_spec(6)
+    filenames, _ = generate_data(
+        4000, 2, 1, 0.0, str(tmp_path), spec=spec, include_key=True
+    )
+    feature_columns = [f"f{i}" for i in range(6)]
+    ds = TorchShufflingDataset(
+        list(filenames),
+        1,
+        num_trainers=1,
+        batch_size=500,
+        rank=0,
+        num_reducers=2,
+        feature_columns=feature_columns,
+        label_column="labels",
+        feature_matrix=True,
+    )
+    ds.set_epoch(0)
+    total = 0
+    for data, target in ds:
+        assert data[0].shape == (len(target), 6)
+        # float64 source values lie in [0, 1) — a layout bug would pull
+        # int64 key bytes (astronomical floats) into the feature matrix.
+        assert torch.isfinite(data[0]).all()
+        assert data[0].abs().max() <= 1.0
+        total += len(target)
+    assert total == 4000
