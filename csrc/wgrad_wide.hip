@@ -26,7 +26,7 @@ typedef __attribute__((__vector_size__(4 * sizeof(short)))) short w_bf16x4;
 typedef __attribute__((__vector_size__(16 * sizeof(float)))) float w_f32x16;
 
 #define WMT 32
-#define WLDS_STRIDE (WMT + 8)  // 24 halfwords = 48 B per [col] row
+#define WLDS_STRIDE (WMT + 8)  // WMT+8 halfwords per [col] row (pad keeps 16-B reads conflict-free)
 
 // Per-operand stage registers: operand with C columns contributes
 // C*WMT/256 elements per thread = C/32 packed row-pair words per thread
