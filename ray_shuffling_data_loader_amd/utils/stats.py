@@ -458,6 +458,19 @@ def process_stats(
     def open_csv(kind, fields):
         filename = f"{kind}_{hr_num_rows}_rows_{hr_batch_size}_batch_size"
         filename += f"_{now}.csv" if unique_stats else ".csv"
+        if "://" in stats_dir:
+            # Remote stats dir (e.g. s3://bucket/prefix) via fsspec, as the
+            # reference does (reference stats.py: fsspec.open on the
+            # hr_stats_dir). Header handling: remote appends are not
+            # supported, so always write the header.
+            import fsspec
+
+            filename = stats_dir.rstrip("/") + "/" + filename
+            f = fsspec.open(filename, write_mode).open()
+            writer = csv.DictWriter(f, fieldnames=fields)
+            writer.writeheader()
+            print(f"Writing out {kind} to {filename}.")
+            return f, writer
         filename = os.path.join(stats_dir, filename)
         write_header = (
             overwrite_stats
