@@ -42,3 +42,38 @@ def test_bench_stdout_is_one_json_line(tmp_path):
     assert d["n_gpus"] == 1 and d["steps"] == 3 and d["warmup"] == 1
     assert d["value"] > 0 and d["ms_per_step"] > 0
     assert d["higher_is_better"] is True and d["scaling"] == "weak"
+
+
+def test_bench_world2_gloo(tmp_path):
+    """The driver's SCALE tier launches bench.py under torchrun with N>1.
+    Exercise that exact path at world 2 on CPU/gloo: per-rank stripe
+    generation + barrier, MAX-over-ranks timing reduce, and exactly one
+    JSON line on stdout (rank 0 only)."""
+    env = dict(os.environ)
+    env["RSDL_TUNABLEOP"] = "0"
+    proc = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--standalone", "--local-addr", "127.0.0.1",
+            "--nproc-per-node", "2",
+            os.path.join(REPO, "bench.py"),
+            "--gpus", "2", "--steps", "3", "--warmup", "1",
+            "--rows-per-gpu", "80000", "--batch-size", "20000",
+            "--num-cols", "8", "--files-per-gpu", "2",
+            "--reducers-per-gpu", "2", "--device", "cpu",
+            "--dtype", "fp32", "--data-dir", str(tmp_path),
+        ],
+        capture_output=True,
+        text=True,
+        timeout=600,
+        env=env,
+        cwd=REPO,
+    )
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    lines = [ln for ln in proc.stdout.splitlines() if ln.strip()]
+    json_lines = [ln for ln in lines if ln.lstrip().startswith("{")]
+    assert len(json_lines) == 1, f"exactly one JSON line, got: {lines}"
+    d = json.loads(json_lines[0])
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["value"] > 0
