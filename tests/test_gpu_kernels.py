@@ -479,3 +479,32 @@ def test_linear_relu_backward_fused_parity():
     out_ref.backward()
     for p, q in zip(lin.parameters(), ref.parameters()):
         assert torch.allclose(p.grad, q.grad, rtol=2e-2, atol=2e-3)
+
+
+@pytest.mark.gpu
+def test_chunked_linear_scalar_head_grads(dev):
+    """Strict parity for the N==1 head fast path (broadcast-mul dgrad +
+    column-weighted-reduce wgrad) vs plain nn.Linear autograd."""
+    from ray_shuffling_data_loader_amd.models.mlp import ChunkedLinear
+
+    torch.manual_seed(11)
+    for dtype in (torch.float32, torch.bfloat16):
+        x = torch.randn(70_000, 128, device=dev, dtype=dtype,
+                        requires_grad=True)
+        ref = torch.nn.Linear(128, 1).to(dev).to(dtype)
+        chk = ChunkedLinear(128, 1).to(dev).to(dtype)
+        chk.load_state_dict(ref.state_dict())
+        g = torch.randn(70_000, 1, device=dev, dtype=dtype)
+        y_ref = ref(x)
+        y_ref.backward(g)
+        gx, gw, gb = (x.grad.clone(), ref.weight.grad.clone(),
+                      ref.bias.grad.clone())
+        x.grad = None
+        y = chk(x)
+        assert torch.allclose(y, y_ref, atol=1e-3)
+        y.backward(g)
+        tol = dict(atol=1e-4, rtol=1e-4) if dtype == torch.float32 else (
+            dict(atol=2e-1, rtol=2e-2))
+        assert torch.allclose(x.grad, gx, **tol), dtype
+        assert torch.allclose(chk.weight.grad, gw, **tol), dtype
+        assert torch.allclose(chk.bias.grad, gb, **tol), dtype
