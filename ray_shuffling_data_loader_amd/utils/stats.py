@@ -466,7 +466,8 @@ def process_stats(
             import fsspec
 
             filename = stats_dir.rstrip("/") + "/" + filename
-            f = fsspec.open(filename, write_mode).open()
+            # Object stores can't append; always (over)write remote CSVs.
+            f = fsspec.open(filename, "w").open()
             writer = csv.DictWriter(f, fieldnames=fields)
             writer.writeheader()
             print(f"Writing out {kind} to {filename}.")
