@@ -416,3 +416,67 @@ def test_torture_many_producers_consumers():
         assert not t.is_alive()
     assert len(consumed) == produced
     assert len(set(consumed)) == produced
+
+
+def test_queue_concurrent_stress():
+    """Randomized multi-threaded stress of the C++ core: one producer
+    driving the epoch window (max_concurrent_epochs=2, maxsize=4 so puts
+    block on backpressure) against one consumer thread per rank across 6
+    epochs. Asserts exact item conservation per (epoch, rank) and that
+    nothing deadlocks (bounded joins)."""
+    import random
+    import threading
+
+    num_epochs, num_trainers = 6, 3
+    q = BatchQueue(num_epochs, num_trainers, 2, maxsize=4)
+    rng = random.Random(1234)
+    sent = {
+        (e, r): [e * 1000 + r * 100 + i for i in range(rng.randint(0, 17))]
+        for e in range(num_epochs)
+        for r in range(num_trainers)
+    }
+    got = {k: [] for k in sent}
+    errors = []
+
+    def producer():
+        try:
+            for e in range(num_epochs):
+                q.new_epoch(e)
+                items = [
+                    (r, v) for r in range(num_trainers) for v in sent[(e, r)]
+                ]
+                rng.shuffle(items)
+                for r, v in items:
+                    q.put(r, e, v, timeout=30)
+                for r in range(num_trainers):
+                    q.producer_done(r, e)
+        except Exception as exc:  # pragma: no cover
+            errors.append(("producer", exc))
+
+    def consumer(r):
+        try:
+            for e in range(num_epochs):
+                done = False
+                while not done:
+                    batch = q.get_batch(r, e)
+                    if batch and batch[-1] is None:
+                        done = True
+                        batch.pop()
+                    got[(e, r)].extend(batch)
+                    q.task_done(r, e, len(batch) + (1 if done else 0))
+        except Exception as exc:  # pragma: no cover
+            errors.append((f"consumer{r}", exc))
+
+    threads = [threading.Thread(target=producer, daemon=True)]
+    threads += [
+        threading.Thread(target=consumer, args=(r,), daemon=True)
+        for r in range(num_trainers)
+    ]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=60)
+        assert not t.is_alive(), "stress test deadlocked"
+    assert not errors, errors
+    for k in sent:
+        assert sorted(got[k]) == sorted(sent[k]), k
