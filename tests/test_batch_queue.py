@@ -418,6 +418,27 @@ def test_torture_many_producers_consumers():
     assert len(set(consumed)) == produced
 
 
+def test_custom_queue_placement(tmp_path, monkeypatch):
+    """The reference's test_custom_resources pins the queue ACTOR to a node
+    via Ray resources; our placement control is the rendezvous directory
+    (RSDL_QUEUE_DIR) for the named-socket server. Verify the socket lands
+    in the requested dir and a client connects through it."""
+    monkeypatch.setenv("RSDL_QUEUE_DIR", str(tmp_path))
+    import os
+
+    name = "placed_queue"
+    q = BatchQueue(1, 1, 1, name=name)
+    try:
+        socks = [f for f in os.listdir(tmp_path) if name in f]
+        assert socks, f"socket not under RSDL_QUEUE_DIR: {os.listdir(tmp_path)}"
+        c = BatchQueue(1, 1, 1, name=name, connect=True)
+        q.new_epoch(0)
+        c.put(0, 0, 42)
+        assert q.get(0, 0, timeout=10) == 42
+    finally:
+        q.shutdown()
+
+
 def test_queue_concurrent_stress():
     """Randomized multi-threaded stress of the C++ core: one producer
     driving the epoch window (max_concurrent_epochs=2, maxsize=4 so puts
