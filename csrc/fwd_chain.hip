@@ -1,0 +1,219 @@
+// EXPERIMENTAL (round-2 WIP, see docs/MEGAKERNEL_PLAN.md): fused forward
+// chain for the flagship TabularMLP on MI355X (gfx950).
+//
+//   a1 = relu(x0 @ W1^T + b1)   [M,100] -> [M,512]
+//   a2 = relu(a1 @ W2^T + b2)   -> [M,256]
+//   a3 = relu(a2 @ W3^T + b3)   -> [M,128]
+//   out = a3 @ w4^T + b4        -> [M,1]
+//
+// One 256-thread workgroup carries a 64-row slab through the whole chain:
+// the slab's activations live in LDS between layers (64x512 + 64x256 +
+// 64x128 bf16 tiles + padding ~= 125 KB of the 160 KB LDS), weights are
+// read from global (L2-resident, ~0.44 MB total) per MFMA step, and only
+// x0 (read) and the a1/a2/a3/out tiles (written once for backward) touch
+// HBM. Eliminates the inter-layer activation re-reads of the eager path
+// (~448 MB/step at the bench shape; measured eager fwd 0.25 ms vs a
+// ~0.1 ms fused roofline).
+//
+// MFMA orientation (v_mfma_f32_32x32x16_bf16, probe-verified maps in
+// tools/mfma_probe.hip / csrc/wgrad_kernel.hip):
+//   D[mrow][ncol] = A[mrow][k] x B[k][ncol], per lane:
+//     A[m = lane&31][k = (lane>>5)*8 + j]   j = 0..7 of a bf16x8
+//     B[k = (lane>>5)*8 + j][n = lane&31]
+//     D col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+// For y[m,n] = sum_k x[m,k] W[n,k] both fragments are CONTIGUOUS row
+// segments (x row m, W row n) — no transpose staging; the only relayout
+// is D -> next layer's LDS tile (b16 column writes, padded stride).
+//
+// Not wired into any default path: built and bound, exercised only by the
+// RSDL_EXPERIMENTAL=1 GPU test. Known-untuned: weight tiles could be
+// LDS-cached for L2/L3, and the k-loops are not software-pipelined yet.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <cstdint>
+
+namespace rsdl {
+
+typedef __attribute__((__vector_size__(8 * sizeof(short)))) short fc_bf16x8;
+typedef __attribute__((__vector_size__(16 * sizeof(float)))) float fc_f32x16;
+
+#define FC_MT 64            // rows per workgroup slab
+#define FC_K0 100           // input feature count
+#define FC_K0P 112          // padded to a 16-multiple for the k-loop
+#define FC_N1 512
+#define FC_N2 256
+#define FC_N3 128
+// LDS halfword strides (16-B aligned: multiples of 8; +8 pad de-banks the
+// column writes of the D->tile stores).
+#define FC_S0 (FC_K0P + 8)
+#define FC_S1 (FC_N1 + 8)
+#define FC_S2 (FC_N2 + 8)
+#define FC_S3 (FC_N3 + 8)
+
+__device__ __forceinline__ float fc_b2f(short s) {
+  __hip_bfloat16 h;
+  *reinterpret_cast<short*>(&h) = s;
+  return __bfloat162float(h);
+}
+
+__device__ __forceinline__ short fc_f2b(float f) {
+  __hip_bfloat16 h = __float2bfloat16(f);
+  return *reinterpret_cast<short*>(&h);
+}
+
+// One GEMM layer of the chain: src LDS tile [FC_MT][K] (stride SRC_S
+// halfwords) x W[N][K] global -> relu(.+bias) -> dst LDS tile (stride
+// DST_S). N_WAVE = output columns per wave (N/4). Each wave owns m-tiles
+// {0,32} x its n-range; accumulators are static f32x16 arrays.
+template <int K, int N, int SRC_S, int DST_S, bool RELU>
+__device__ void fc_layer(const short* __restrict__ src_lds,
+                         const short* __restrict__ W,
+                         const float* __restrict__ bias,
+                         short* __restrict__ dst_lds, int32_t wave,
+                         int32_t lane) {
+  constexpr int NT = N / 128;  // n-tiles of 32 per wave (4 waves)
+  const int32_t n_base = wave * (N / 4);
+  const int32_t frag_k0 = (lane >> 5) * 8;
+  const int32_t ml = lane & 31;  // A row within m-tile / D col (n)
+
+  fc_f32x16 acc[2][NT] = {};
+  for (int32_t k = 0; k < K; k += 16) {
+    fc_bf16x8 a[2], b[NT];
+    #pragma unroll
+    for (int mt = 0; mt < 2; mt++) {
+      *reinterpret_cast<uint4*>(&a[mt]) =
+          *reinterpret_cast<const uint4*>(
+              &src_lds[(mt * 32 + ml) * SRC_S + k + frag_k0]);
+    }
+    #pragma unroll
+    for (int nt = 0; nt < NT; nt++) {
+      const int32_t n = n_base + nt * 32 + ml;
+      *reinterpret_cast<uint4*>(&b[nt]) =
+          *reinterpret_cast<const uint4*>(&W[n * K + k + frag_k0]);
+    }
+    #pragma unroll
+    for (int mt = 0; mt < 2; mt++) {
+      #pragma unroll
+      for (int nt = 0; nt < NT; nt++) {
+        acc[mt][nt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            a[mt], b[nt], acc[mt][nt], 0, 0, 0);
+      }
+    }
+  }
+  // Epilogue: bias + (relu) + cast + store the D fragments into the dst
+  // tile at [mrow][n] (b16 column writes; DST_S padding spreads banks).
+  #pragma unroll
+  for (int mt = 0; mt < 2; mt++) {
+    #pragma unroll
+    for (int nt = 0; nt < NT; nt++) {
+      const int32_t n = n_base + nt * 32 + ml;
+      const float bv = bias[n];
+      #pragma unroll
+      for (int reg = 0; reg < 16; reg++) {
+        const int32_t mrow =
+            mt * 32 + (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+        float v = acc[mt][nt][reg] + bv;
+        if (RELU) v = v > 0.f ? v : 0.f;
+        dst_lds[mrow * DST_S + n] = fc_f2b(v);
+      }
+    }
+  }
+}
+
+// Copy an LDS activation tile [FC_MT][N] (stride S halfwords) to the
+// global row-major [M, N] tensor, vectorized 16 B.
+template <int N, int S>
+__device__ void fc_store_tile(const short* __restrict__ lds, short* out,
+                              int64_t m0, int64_t M, int32_t tid) {
+  constexpr int VPR = N / 8;  // uint4 vectors per row
+  for (int32_t u = tid; u < FC_MT * VPR; u += 256) {
+    const int32_t m = u / VPR;
+    const int32_t c = (u % VPR) * 8;
+    if (m0 + m < M) {
+      *reinterpret_cast<uint4*>(&out[(m0 + m) * N + c]) =
+          *reinterpret_cast<const uint4*>(&lds[m * S + c]);
+    }
+  }
+}
+
+__global__ void __launch_bounds__(256) fwd_chain_kernel(
+    const short* __restrict__ x0,  // [M, 100] bf16 bits
+    const short* __restrict__ W1, const float* __restrict__ b1,
+    const short* __restrict__ W2, const float* __restrict__ b2,
+    const short* __restrict__ W3, const float* __restrict__ b3,
+    const short* __restrict__ w4, const float* __restrict__ b4,
+    short* __restrict__ a1, short* __restrict__ a2,
+    short* __restrict__ a3, short* __restrict__ out, int64_t M) {
+  __shared__ short t0[FC_MT * FC_S0];
+  __shared__ short t1[FC_MT * FC_S1];
+  __shared__ short t2[FC_MT * FC_S2];
+  __shared__ short t3[FC_MT * FC_S3];
+
+  const int64_t m0 = (int64_t)blockIdx.x * FC_MT;
+  const int32_t tid = threadIdx.x;
+  const int32_t wave = tid >> 6;
+  const int32_t lane = tid & 63;
+
+  // Stage the x0 slab (zero-padding cols 100..111 and rows past M). This
+  // is a one-time per-WG load, not the hot loop — scalar guards are fine.
+  for (int32_t u = tid; u < FC_MT * FC_K0P; u += 256) {
+    const int32_t m = u / FC_K0P;
+    const int32_t c = u % FC_K0P;
+    short v = 0;
+    if (m0 + m < M && c < FC_K0) v = x0[(m0 + m) * FC_K0 + c];
+    t0[m * FC_S0 + c] = v;
+  }
+  __syncthreads();
+
+  fc_layer<FC_K0P, FC_N1, FC_S0, FC_S1, true>(t0, W1, b1, t1, wave, lane);
+  __syncthreads();
+  fc_layer<FC_N1, FC_N2, FC_S1, FC_S2, true>(t1, W2, b2, t2, wave, lane);
+  __syncthreads();
+  fc_layer<FC_N2, FC_N3, FC_S2, FC_S3, true>(t2, W3, b3, t3, wave, lane);
+  __syncthreads();
+
+  // Head: out[m] = sum_k a3[m,k] * w4[k] + b4. 4 threads per row, 32 k
+  // each, pair-wise LDS-free reduce via wave shuffles (partners are
+  // adjacent lanes).
+  {
+    const int32_t m = tid >> 2;
+    const int32_t part = tid & 3;
+    float s = 0.f;
+    #pragma unroll
+    for (int32_t kk = 0; kk < 32; kk++) {
+      const int32_t k = part * 32 + kk;
+      s += fc_b2f(t3[m * FC_S3 + k]) * fc_b2f(w4[k]);
+    }
+    s += __shfl_down(s, 1);
+    s += __shfl_down(s, 2);
+    if (part == 0 && m0 + m < M) {
+      out[m0 + m] = fc_f2b(s + b4[0]);
+    }
+  }
+
+  fc_store_tile<FC_N1, FC_S1>(t1, a1, m0, M, tid);
+  fc_store_tile<FC_N2, FC_S2>(t2, a2, m0, M, tid);
+  fc_store_tile<FC_N3, FC_S3>(t3, a3, m0, M, tid);
+}
+
+void launch_fwd_chain(const void* x0, const void* W1, const float* b1,
+                      const void* W2, const float* b2, const void* W3,
+                      const float* b3, const void* w4, const float* b4,
+                      void* a1, void* a2, void* a3, void* out, int64_t M,
+                      hipStream_t stream) {
+  const int32_t grid = (int32_t)((M + FC_MT - 1) / FC_MT);
+  hipLaunchKernelGGL(fwd_chain_kernel, dim3(grid), dim3(256), 0, stream,
+                     reinterpret_cast<const short*>(x0),
+                     reinterpret_cast<const short*>(W1), b1,
+                     reinterpret_cast<const short*>(W2), b2,
+                     reinterpret_cast<const short*>(W3), b3,
+                     reinterpret_cast<const short*>(w4), b4,
+                     reinterpret_cast<short*>(a1),
+                     reinterpret_cast<short*>(a2),
+                     reinterpret_cast<short*>(a3),
+                     reinterpret_cast<short*>(out), M);
+}
+
+}  // namespace rsdl

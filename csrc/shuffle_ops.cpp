@@ -68,6 +68,11 @@ void launch_wgrad_wide(const void* dy, const void* x, float* dW, float* db,
 void launch_relu_bwd_bias(const void* dy, const void* y, void* dx,
                           float* db_part, int64_t nvec, int32_t N,
                           int32_t grid, hipStream_t stream);
+void launch_fwd_chain(const void* x0, const void* W1, const float* b1,
+                      const void* W2, const float* b2, const void* W3,
+                      const float* b3, const void* w4, const float* b4,
+                      void* a1, void* a2, void* a3, void* out, int64_t M,
+                      hipStream_t stream);
 
 namespace {
 
@@ -437,6 +442,55 @@ std::vector<at::Tensor> relu_bwd_bias(const at::Tensor& dy,
   return {dx, db_part.sum(0)};
 }
 
+
+// EXPERIMENTAL round-2 fused forward chain (csrc/fwd_chain.hip): fixed
+// TabularMLP(100-512-256-128-1) architecture, bf16. Returns
+// (a1, a2, a3, out). Exercised only by the RSDL_EXPERIMENTAL=1 GPU test.
+std::vector<at::Tensor> fwd_chain_bf16(
+    const at::Tensor& x, const at::Tensor& W1, const at::Tensor& b1,
+    const at::Tensor& W2, const at::Tensor& b2, const at::Tensor& W3,
+    const at::Tensor& b3, const at::Tensor& w4, const at::Tensor& b4) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
+                  x.dim() == 2 && x.size(1) == 100 && x.is_contiguous(),
+              "fwd_chain: x must be contiguous bf16 [M,100]");
+  auto chk_w = [](const at::Tensor& W, int64_t n, int64_t k,
+                  const char* nm) {
+    TORCH_CHECK(W.is_cuda() && W.scalar_type() == at::kBFloat16 &&
+                    W.dim() == 2 && W.size(0) == n && W.size(1) == k &&
+                    W.is_contiguous(),
+                "fwd_chain: ", nm, " must be contiguous bf16 [", n, ",", k,
+                "]");
+  };
+  chk_w(W1, 512, 100, "W1");
+  chk_w(W2, 256, 512, "W2");
+  chk_w(W3, 128, 256, "W3");
+  TORCH_CHECK(w4.numel() == 128 && w4.scalar_type() == at::kBFloat16 &&
+                  w4.is_contiguous(),
+              "fwd_chain: w4 must be contiguous bf16 [128]");
+  auto fb = [](const at::Tensor& b, int64_t n) {
+    TORCH_CHECK(b.numel() == n, "fwd_chain: bias size mismatch");
+    return b.to(at::kFloat).contiguous();
+  };
+  auto b1f = fb(b1, 512), b2f = fb(b2, 256), b3f = fb(b3, 128),
+       b4f = fb(b4, 1);
+  const int64_t M = x.size(0);
+  // The kernel's A-fragment loads need W1's k-dim padded 100 -> 112.
+  auto W1p = at::constant_pad_nd(W1, {0, 12}).contiguous();
+  // ...and x0's rows padded likewise, staged in LDS by the kernel itself.
+  auto a1 = at::empty({M, 512}, x.options());
+  auto a2 = at::empty({M, 256}, x.options());
+  auto a3 = at::empty({M, 128}, x.options());
+  auto out = at::empty({M, 1}, x.options());
+  if (M > 0) {
+    launch_fwd_chain(x.data_ptr(), W1p.data_ptr(), b1f.data_ptr<float>(),
+                     W2.data_ptr(), b2f.data_ptr<float>(), W3.data_ptr(),
+                     b3f.data_ptr<float>(), w4.data_ptr(),
+                     b4f.data_ptr<float>(), a1.data_ptr(), a2.data_ptr(),
+                     a3.data_ptr(), out.data_ptr(), M, current_stream());
+  }
+  return {a1, a2, a3, out};
+}
+
 }  // namespace
 }  // namespace rsdl
 
@@ -460,6 +514,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad_bf16", &rsdl::wgrad_bf16, py::arg("dy"), py::arg("x"),
         py::arg("with_bias") = true);
   m.def("relu_bwd_bias", &rsdl::relu_bwd_bias, py::arg("dy"), py::arg("y"));
+  m.def("fwd_chain_bf16", &rsdl::fwd_chain_bf16);
   m.attr("DT_F32") = (int)rsdl::DT_F32;
   m.attr("DT_F64") = (int)rsdl::DT_F64;
   m.attr("DT_I32") = (int)rsdl::DT_I32;

@@ -1,6 +1,8 @@
 """GPU (MI355X) tests: HIP kernels vs the CPU torch/numpy oracle, and the
 end-to-end GPU loader path. All marked @pytest.mark.gpu."""
 
+import os
+
 import numpy as np
 import pytest
 import torch
@@ -515,3 +517,43 @@ def test_chunked_linear_scalar_head_grads(dev):
         assert torch.allclose(x.grad, gx, **dtol), dtype
         assert torch.allclose(chk.weight.grad, gw, **wtol), dtype
         assert torch.allclose(chk.bias.grad, gb, **wtol), dtype
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(
+    os.environ.get("RSDL_EXPERIMENTAL") != "1",
+    reason="round-2 WIP kernel (docs/MEGAKERNEL_PLAN.md); "
+    "opt in with RSDL_EXPERIMENTAL=1",
+)
+def test_fwd_chain_matches_eager(dev):
+    """EXPERIMENTAL fused forward chain vs the eager fp32 reference."""
+    from ray_shuffling_data_loader_amd.ops import shuffle_ops
+
+    hip = shuffle_ops._load_hip()
+    torch.manual_seed(5)
+    M = 4096 + 17  # non-multiple of the 64-row slab: exercises the tail
+    x = torch.randn(M, 100, device=dev).bfloat16()
+    Ws = [torch.randn(n, k, device=dev).bfloat16() / k**0.5
+          for n, k in [(512, 100), (256, 512), (128, 256), (1, 128)]]
+    bs = [torch.randn(n, device=dev).bfloat16() / 8
+          for n in (512, 256, 128, 1)]
+    a1, a2, a3, out = hip.fwd_chain_bf16(
+        x, Ws[0], bs[0], Ws[1], bs[1], Ws[2], bs[2],
+        Ws[3].flatten(), bs[3],
+    )
+    # fp32 eager oracle on the bf16 inputs
+    r = x.float()
+    refs = []
+    for i, (W, b) in enumerate(zip(Ws, bs)):
+        r = r @ W.float().t() + b.float()
+        if i < 3:
+            r = torch.relu(r)
+        refs.append(r)
+    for got, ref, name in [(a1, refs[0], "a1"), (a2, refs[1], "a2"),
+                           (a3, refs[2], "a3"), (out, refs[3], "out")]:
+        ref_b = ref.bfloat16().float()
+        err = (got.float() - ref_b).abs()
+        # bf16 MFMA fp32-accum vs fp32 eager: rounding of inputs dominates
+        scale = ref_b.abs().mean().clamp(min=1.0)
+        assert err.max() <= 0.12 * scale + 0.05, (
+            name, err.max().item(), scale.item())
