@@ -1,0 +1,240 @@
+// EXPERIMENTAL (round-2 WIP, see docs/MEGAKERNEL_PLAN.md): fused backward
+// chain for the flagship TabularMLP on MI355X (gfx950). Companion to
+// csrc/fwd_chain.hip.
+//
+// Given dy = dLoss/dout [M,1] and the saved activations a1/a2/a3:
+//   da3 = dy * w4          (outer product)      dz3 = da3 * (a3 > 0)
+//   da2 = dz3 @ W3         ([M,128] -> [M,256]) dz2 = da2 * (a2 > 0)
+//   da1 = dz2 @ W2         ([M,256] -> [M,512]) dz1 = da1 * (a1 > 0)
+//   db_l = sum_m dz_l (l = 1..3), db4 = sum_m dy
+// One 64-row slab per workgroup; the a_l tiles are loaded once and
+// overwritten IN PLACE by their dz_l (the mask consumes the value it
+// replaces), so LDS peaks at ~118 KB. dz1/dz2/dz3 are written to global
+// for the (separate, reduction-shaped) wgrad kernels; bias partials go to
+// per-workgroup slabs finalized by one at::sum.
+//
+// dgrad orientation: da[m,c] = sum_n dz[m,n] W[n,c] — the contraction is
+// over the layer's OUTPUT index n, so the MFMA B fragment (contiguous
+// over the contraction) needs memory laid out [c][n]: the host passes
+// PRE-TRANSPOSED weights WT = W.t().contiguous(), shape
+// [in_features, out_features] (cheap once per step; both fragments are
+// then contiguous row segments — same A/B/D maps as fwd_chain,
+// probe-verified in tools/mfma_probe.hip). In bc_layer terms WT is
+// [N, K]: N = dgrad output width (in_features), K = contraction width.
+//
+// Exercised only by the RSDL_EXPERIMENTAL=1 GPU test; not on any default
+// path.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <cstdint>
+
+namespace rsdl {
+
+typedef __attribute__((__vector_size__(8 * sizeof(short)))) short bc_bf16x8;
+typedef __attribute__((__vector_size__(16 * sizeof(float)))) float bc_f32x16;
+
+#define BC_MT 64
+#define BC_N1 512
+#define BC_N2 256
+#define BC_N3 128
+#define BC_S1 (BC_N1 + 8)
+#define BC_S2 (BC_N2 + 8)
+#define BC_S3 (BC_N3 + 8)
+
+__device__ __forceinline__ float bc_b2f(short s) {
+  __hip_bfloat16 h;
+  *reinterpret_cast<short*>(&h) = s;
+  return __bfloat162float(h);
+}
+
+__device__ __forceinline__ short bc_f2b(float f) {
+  __hip_bfloat16 h = __float2bfloat16(f);
+  return *reinterpret_cast<short*>(&h);
+}
+
+// Load a [BC_MT][N] global activation tile into LDS (stride S), 16-B
+// vectors, rows past M zero-filled (their dz contributions are masked by
+// the guarded final stores and zero bias partials).
+template <int N, int S>
+__device__ void bc_load_tile(const short* __restrict__ g, short* lds,
+                             int64_t m0, int64_t M, int32_t tid) {
+  constexpr int VPR = N / 8;
+  for (int32_t u = tid; u < BC_MT * VPR; u += 256) {
+    const int32_t m = u / VPR;
+    const int32_t c = (u % VPR) * 8;
+    uint4 v = {0, 0, 0, 0};
+    if (m0 + m < M) {
+      v = *reinterpret_cast<const uint4*>(&g[(m0 + m) * N + c]);
+    }
+    *reinterpret_cast<uint4*>(&lds[m * S + c]) = v;
+  }
+}
+
+template <int N, int S>
+__device__ void bc_store_tile(const short* __restrict__ lds, short* out,
+                              int64_t m0, int64_t M, int32_t tid) {
+  constexpr int VPR = N / 8;
+  for (int32_t u = tid; u < BC_MT * VPR; u += 256) {
+    const int32_t m = u / VPR;
+    const int32_t c = (u % VPR) * 8;
+    if (m0 + m < M) {
+      *reinterpret_cast<uint4*>(&out[(m0 + m) * N + c]) =
+          *reinterpret_cast<const uint4*>(&lds[m * S + c]);
+    }
+  }
+}
+
+// Column sums of a dz tile -> db_part[blockIdx][col] (fp32). 64-row sums,
+// one thread per (col stripe); plain LDS reads, epilogue-only.
+template <int N, int S>
+__device__ void bc_bias_partial(const short* __restrict__ dz,
+                                float* __restrict__ part, int32_t tid) {
+  for (int32_t c = tid; c < N; c += 256) {
+    float s = 0.f;
+    #pragma unroll 4
+    for (int32_t m = 0; m < BC_MT; m++) s += bc_b2f(dz[m * S + c]);
+    part[c] = s;
+  }
+}
+
+// One dgrad layer: da = dz_src @ W (via WT [N,K] contiguous), then
+// dz_dst = da * (a_dst > 0) written IN PLACE over the a_dst tile.
+// K = dz_src width (contraction), N = output width.
+template <int K, int N, int SRC_S, int DST_S>
+__device__ void bc_layer(const short* __restrict__ dz_src,
+                         const short* __restrict__ WT,
+                         short* __restrict__ a_dst, int32_t wave,
+                         int32_t lane) {
+  constexpr int NT = N / 128;
+  const int32_t n_base = wave * (N / 4);
+  const int32_t frag_k0 = (lane >> 5) * 8;
+  const int32_t ml = lane & 31;
+
+  bc_f32x16 acc[2][NT] = {};
+  for (int32_t k = 0; k < K; k += 16) {
+    bc_bf16x8 a[2], b[NT];
+    #pragma unroll
+    for (int mt = 0; mt < 2; mt++) {
+      *reinterpret_cast<uint4*>(&a[mt]) =
+          *reinterpret_cast<const uint4*>(
+              &dz_src[(mt * 32 + ml) * SRC_S + k + frag_k0]);
+    }
+    #pragma unroll
+    for (int nt = 0; nt < NT; nt++) {
+      const int32_t n = n_base + nt * 32 + ml;
+      *reinterpret_cast<uint4*>(&b[nt]) =
+          *reinterpret_cast<const uint4*>(&WT[(int64_t)n * K + k + frag_k0]);
+    }
+    #pragma unroll
+    for (int mt = 0; mt < 2; mt++) {
+      #pragma unroll
+      for (int nt = 0; nt < NT; nt++) {
+        acc[mt][nt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            a[mt], b[nt], acc[mt][nt], 0, 0, 0);
+      }
+    }
+  }
+  // No barrier needed between the k-loop and the epilogue: each wave
+  // reads only dz_src (a different buffer) and masks/writes only its own
+  // n-range of a_dst; inter-layer ordering is handled by the
+  // __syncthreads() between bc_layer calls in the kernel body.
+  #pragma unroll
+  for (int mt = 0; mt < 2; mt++) {
+    #pragma unroll
+    for (int nt = 0; nt < NT; nt++) {
+      const int32_t n = n_base + nt * 32 + ml;
+      #pragma unroll
+      for (int reg = 0; reg < 16; reg++) {
+        const int32_t mrow =
+            mt * 32 + (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+        short* cell = &a_dst[mrow * DST_S + n];
+        const float live = bc_b2f(*cell) > 0.f ? 1.f : 0.f;
+        *cell = bc_f2b(acc[mt][nt][reg] * live);
+      }
+    }
+  }
+}
+
+__global__ void __launch_bounds__(256) bwd_chain_kernel(
+    const short* __restrict__ dy,    // [M,1] bf16 (head grad)
+    const short* __restrict__ a1,    // [M,512] saved activations
+    const short* __restrict__ a2,    // [M,256]
+    const short* __restrict__ a3,    // [M,128]
+    const short* __restrict__ w4,    // [128]
+    const short* __restrict__ W3T,   // [256,128]  (W3^T, contiguous)
+    const short* __restrict__ W2T,   // [512,256]  (W2^T, contiguous)
+    short* __restrict__ dz1, short* __restrict__ dz2,
+    short* __restrict__ dz3,
+    float* __restrict__ db_part,     // [grid][512+256+128+1]
+    int64_t M) {
+  __shared__ short t1[BC_MT * BC_S1];
+  __shared__ short t2[BC_MT * BC_S2];
+  __shared__ short t3[BC_MT * BC_S3];
+  __shared__ float dyf[BC_MT];
+
+  const int64_t m0 = (int64_t)blockIdx.x * BC_MT;
+  const int32_t tid = threadIdx.x;
+  const int32_t wave = tid >> 6;
+  const int32_t lane = tid & 63;
+  float* part = &db_part[(int64_t)blockIdx.x * (BC_N1 + BC_N2 + BC_N3 + 1)];
+
+  bc_load_tile<BC_N1, BC_S1>(a1, t1, m0, M, tid);
+  bc_load_tile<BC_N2, BC_S2>(a2, t2, m0, M, tid);
+  bc_load_tile<BC_N3, BC_S3>(a3, t3, m0, M, tid);
+  if (tid < BC_MT) {
+    dyf[tid] = (m0 + tid < M) ? bc_b2f(dy[m0 + tid]) : 0.f;
+  }
+  __syncthreads();
+
+  // dz3 = (dy * w4) * (a3 > 0), in place over t3; also db4 partial.
+  if (tid == 0) {
+    float s = 0.f;
+    for (int32_t m = 0; m < BC_MT; m++) s += dyf[m];
+    part[BC_N1 + BC_N2 + BC_N3] = s;
+  }
+  for (int32_t u = tid; u < BC_MT * BC_N3; u += 256) {
+    const int32_t m = u / BC_N3;
+    const int32_t k = u % BC_N3;
+    short* cell = &t3[m * BC_S3 + k];
+    const float live = bc_b2f(*cell) > 0.f ? 1.f : 0.f;
+    *cell = bc_f2b(dyf[m] * bc_b2f(w4[k]) * live);
+  }
+  __syncthreads();
+
+  // da2 = dz3 @ W3 (via W3T), mask by a2 -> dz2 in place over t2.
+  bc_layer<BC_N3, BC_N2, BC_S3, BC_S2>(t3, W3T, t2, wave, lane);
+  __syncthreads();
+  // da1 = dz2 @ W2 (via W2T), mask by a1 -> dz1 in place over t1.
+  bc_layer<BC_N2, BC_N1, BC_S2, BC_S1>(t2, W2T, t1, wave, lane);
+  __syncthreads();
+
+  bc_bias_partial<BC_N1, BC_S1>(t1, &part[0], tid);
+  bc_bias_partial<BC_N2, BC_S2>(t2, &part[BC_N1], tid);
+  bc_bias_partial<BC_N3, BC_S3>(t3, &part[BC_N1 + BC_N2], tid);
+
+  bc_store_tile<BC_N1, BC_S1>(t1, dz1, m0, M, tid);
+  bc_store_tile<BC_N2, BC_S2>(t2, dz2, m0, M, tid);
+  bc_store_tile<BC_N3, BC_S3>(t3, dz3, m0, M, tid);
+}
+
+void launch_bwd_chain(const void* dy, const void* a1, const void* a2,
+                      const void* a3, const void* w4, const void* W3T,
+                      const void* W2T, void* dz1, void* dz2, void* dz3,
+                      float* db_part, int64_t M, hipStream_t stream) {
+  const int32_t grid = (int32_t)((M + BC_MT - 1) / BC_MT);
+  hipLaunchKernelGGL(bwd_chain_kernel, dim3(grid), dim3(256), 0, stream,
+                     reinterpret_cast<const short*>(dy),
+                     reinterpret_cast<const short*>(a1),
+                     reinterpret_cast<const short*>(a2),
+                     reinterpret_cast<const short*>(a3),
+                     reinterpret_cast<const short*>(w4),
+                     reinterpret_cast<const short*>(W3T),
+                     reinterpret_cast<const short*>(W2T),
+                     reinterpret_cast<short*>(dz1),
+                     reinterpret_cast<short*>(dz2),
+                     reinterpret_cast<short*>(dz3), db_part, M);
+}
+
+}  // namespace rsdl

@@ -73,6 +73,10 @@ void launch_fwd_chain(const void* x0, const void* W1, const float* b1,
                       const float* b3, const void* w4, const float* b4,
                       void* a1, void* a2, void* a3, void* out, int64_t M,
                       hipStream_t stream);
+void launch_bwd_chain(const void* dy, const void* a1, const void* a2,
+                      const void* a3, const void* w4, const void* W3T,
+                      const void* W2T, void* dz1, void* dz2, void* dz3,
+                      float* db_part, int64_t M, hipStream_t stream);
 
 namespace {
 
@@ -491,6 +495,62 @@ std::vector<at::Tensor> fwd_chain_bf16(
   return {a1, a2, a3, out};
 }
 
+
+// EXPERIMENTAL round-2 fused backward chain (csrc/bwd_chain.hip):
+// dgrad + relu mask + bias partials for the fixed TabularMLP
+// architecture. Returns (dz1, dz2, dz3, db1, db2, db3, db4); the wgrads
+// (dW_l = dz_l^T @ a_{l-1}) remain the caller's streaming kernels.
+std::vector<at::Tensor> bwd_chain_bf16(
+    const at::Tensor& dy, const at::Tensor& a1, const at::Tensor& a2,
+    const at::Tensor& a3, const at::Tensor& w4, const at::Tensor& W3,
+    const at::Tensor& W2) {
+  const int64_t M = dy.size(0);
+  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == at::kBFloat16 &&
+                  dy.numel() == M && dy.is_contiguous(),
+              "bwd_chain: dy must be contiguous bf16 [M] or [M,1]");
+  auto chk_a = [M](const at::Tensor& a, int64_t n, const char* nm) {
+    TORCH_CHECK(a.is_cuda() && a.scalar_type() == at::kBFloat16 &&
+                    a.dim() == 2 && a.size(0) == M && a.size(1) == n &&
+                    a.is_contiguous(),
+                "bwd_chain: ", nm, " must be contiguous bf16 [M,", n, "]");
+  };
+  chk_a(a1, 512, "a1");
+  chk_a(a2, 256, "a2");
+  chk_a(a3, 128, "a3");
+  TORCH_CHECK(w4.numel() == 128 && w4.scalar_type() == at::kBFloat16,
+              "bwd_chain: w4 must be bf16 [128]");
+  TORCH_CHECK(W3.size(0) == 128 && W3.size(1) == 256 &&
+                  W3.scalar_type() == at::kBFloat16,
+              "bwd_chain: W3 must be bf16 [128,256]");
+  TORCH_CHECK(W2.size(0) == 256 && W2.size(1) == 512 &&
+                  W2.scalar_type() == at::kBFloat16,
+              "bwd_chain: W2 must be bf16 [256,512]");
+  auto w4c = w4.contiguous();
+  auto W3T = W3.t().contiguous();  // [256,128]
+  auto W2T = W2.t().contiguous();  // [512,256]
+  auto dz1 = at::empty({M, 512}, dy.options());
+  auto dz2 = at::empty({M, 256}, dy.options());
+  auto dz3 = at::empty({M, 128}, dy.options());
+  const int64_t grid = (M + 63) / 64;
+  auto db_part = at::zeros({std::max<int64_t>(grid, 1), 512 + 256 + 128 + 1},
+                           dy.options().dtype(at::kFloat));
+  if (M > 0) {
+    launch_bwd_chain(dy.data_ptr(), a1.data_ptr(), a2.data_ptr(),
+                     a3.data_ptr(), w4c.data_ptr(), W3T.data_ptr(),
+                     W2T.data_ptr(), dz1.data_ptr(), dz2.data_ptr(),
+                     dz3.data_ptr(), db_part.data_ptr<float>(), M,
+                     current_stream());
+  }
+  auto db = db_part.sum(0);
+  return {dz1,
+          dz2,
+          dz3,
+          db.narrow(0, 0, 512),
+          db.narrow(0, 512, 256),
+          db.narrow(0, 512 + 256, 128),
+          db.narrow(0, 512 + 256 + 128, 1)};
+}
+
 }  // namespace
 }  // namespace rsdl
 
@@ -515,6 +575,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("with_bias") = true);
   m.def("relu_bwd_bias", &rsdl::relu_bwd_bias, py::arg("dy"), py::arg("y"));
   m.def("fwd_chain_bf16", &rsdl::fwd_chain_bf16);
+  m.def("bwd_chain_bf16", &rsdl::bwd_chain_bf16);
   m.attr("DT_F32") = (int)rsdl::DT_F32;
   m.attr("DT_F64") = (int)rsdl::DT_F64;
   m.attr("DT_I32") = (int)rsdl::DT_I32;

@@ -43,6 +43,7 @@ try:
         "csrc/wgrad_kernel.hip",
         "csrc/relu_bwd.hip",
         "csrc/fwd_chain.hip",
+        "csrc/bwd_chain.hip",
         "csrc/wgrad_wide.hip",
     ]
     if all(os.path.exists(s) for s in hip_sources):

@@ -557,3 +557,51 @@ def test_fwd_chain_matches_eager(dev):
         scale = ref_b.abs().mean().clamp(min=1.0)
         assert err.max() <= 0.12 * scale + 0.05, (
             name, err.max().item(), scale.item())
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(
+    os.environ.get("RSDL_EXPERIMENTAL") != "1",
+    reason="round-2 WIP kernel (docs/MEGAKERNEL_PLAN.md); "
+    "opt in with RSDL_EXPERIMENTAL=1",
+)
+def test_bwd_chain_matches_eager(dev):
+    """EXPERIMENTAL fused backward chain vs the eager fp32 reference."""
+    from ray_shuffling_data_loader_amd.ops import shuffle_ops
+
+    hip = shuffle_ops._load_hip()
+    torch.manual_seed(6)
+    M = 4096 + 31
+    a1 = torch.relu(torch.randn(M, 512, device=dev) - 0.2).bfloat16()
+    a2 = torch.relu(torch.randn(M, 256, device=dev) - 0.2).bfloat16()
+    a3 = torch.relu(torch.randn(M, 128, device=dev) - 0.2).bfloat16()
+    dy = torch.randn(M, 1, device=dev).bfloat16()
+    w4 = (torch.randn(128, device=dev) / 11).bfloat16()
+    W3 = (torch.randn(128, 256, device=dev) / 16).bfloat16()
+    W2 = (torch.randn(256, 512, device=dev) / 22).bfloat16()
+    dz1, dz2, dz3, db1, db2, db3, db4 = hip.bwd_chain_bf16(
+        dy, a1, a2, a3, w4, W3, W2
+    )
+    # fp32 eager oracle
+    da3 = dy.float() @ w4.float().unsqueeze(0)
+    rz3 = da3 * (a3.float() > 0)
+    da2 = rz3 @ W3.float()
+    rz2 = da2 * (a2.float() > 0)
+    da1 = rz2 @ W2.float()
+    rz1 = da1 * (a1.float() > 0)
+    for got, ref, name in [(dz3, rz3, "dz3"), (dz2, rz2, "dz2"),
+                           (dz1, rz1, "dz1")]:
+        ref_b = ref.bfloat16().float()
+        err = (got.float() - ref_b).abs()
+        scale = ref_b.abs().mean().clamp(min=1e-3)
+        assert err.max() <= 0.12 * scale + 0.05, (
+            name, err.max().item(), scale.item())
+    # bias grads: fp32-accum column sums of the (bf16-rounded) dz tiles;
+    # tolerance scaled to the reduction (see scalar-head test).
+    for got, refdz, name in [(db1, rz1, "db1"), (db2, rz2, "db2"),
+                             (db3, rz3, "db3")]:
+        ref = refdz.bfloat16().float().sum(0)
+        assert torch.allclose(got, ref, atol=0.5 + 0.02 * ref.abs().max(),
+                              rtol=0.02), name
+    assert torch.allclose(
+        db4, dy.float().sum(0), atol=0.5, rtol=0.02)
