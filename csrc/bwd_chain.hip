@@ -169,9 +169,9 @@ __global__ void __launch_bounds__(256) bwd_chain_kernel(
     short* __restrict__ dz3,
     float* __restrict__ db_part,     // [grid][512+256+128+1]
     int64_t M) {
-  __shared__ short t1[BC_MT * BC_S1];
-  __shared__ short t2[BC_MT * BC_S2];
-  __shared__ short t3[BC_MT * BC_S3];
+  __shared__ __align__(16) short t1[BC_MT * BC_S1];
+  __shared__ __align__(16) short t2[BC_MT * BC_S2];
+  __shared__ __align__(16) short t3[BC_MT * BC_S3];
   __shared__ float dyf[BC_MT];
 
   const int64_t m0 = (int64_t)blockIdx.x * BC_MT;

@@ -146,10 +146,10 @@ __global__ void __launch_bounds__(256) fwd_chain_kernel(
     const short* __restrict__ w4, const float* __restrict__ b4,
     short* __restrict__ a1, short* __restrict__ a2,
     short* __restrict__ a3, short* __restrict__ out, int64_t M) {
-  __shared__ short t0[FC_MT * FC_S0];
-  __shared__ short t1[FC_MT * FC_S1];
-  __shared__ short t2[FC_MT * FC_S2];
-  __shared__ short t3[FC_MT * FC_S3];
+  __shared__ __align__(16) short t0[FC_MT * FC_S0];
+  __shared__ __align__(16) short t1[FC_MT * FC_S1];
+  __shared__ __align__(16) short t2[FC_MT * FC_S2];
+  __shared__ __align__(16) short t3[FC_MT * FC_S3];
 
   const int64_t m0 = (int64_t)blockIdx.x * FC_MT;
   const int32_t tid = threadIdx.x;
