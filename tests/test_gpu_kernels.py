@@ -605,3 +605,42 @@ def test_bwd_chain_matches_eager(dev):
                               rtol=0.02), name
     assert torch.allclose(
         db4, dy.float().sum(0), atol=0.5, rtol=0.02)
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(
+    os.environ.get("RSDL_EXPERIMENTAL") != "1",
+    reason="round-2 WIP fused step (docs/MEGAKERNEL_PLAN.md); "
+    "opt in with RSDL_EXPERIMENTAL=1",
+)
+def test_fused_step_matches_eager(dev):
+    """EXPERIMENTAL whole-step parity: fused_step's loss and param grads
+    vs the eager autocast fwd+bwd on identical weights and batch."""
+    from ray_shuffling_data_loader_amd.models.fused_step import fused_step
+    from ray_shuffling_data_loader_amd.models.mlp import TabularMLP
+
+    torch.manual_seed(9)
+    M = 1 << 17
+    model = TabularMLP(100).to(dev)
+    ref = TabularMLP(100).to(dev)
+    ref.load_state_dict(model.state_dict())
+    x = torch.randn(M, 100, device=dev).bfloat16()
+    t = torch.randn(M, 1, device=dev)
+
+    loss = fused_step(model, x, t)
+
+    with torch.autocast("cuda", torch.bfloat16):
+        out = ref(x)
+        ref_loss = torch.nn.functional.mse_loss(out.float(), t)
+    ref_loss.backward()
+
+    assert abs(loss.item() - ref_loss.item()) <= 0.02 * ref_loss.item() + 1e-3
+    for (n, p), (_, q) in zip(
+        model.named_parameters(), ref.named_parameters()
+    ):
+        assert p.grad is not None and q.grad is not None, n
+        err = (p.grad - q.grad).abs()
+        tol = 0.05 * q.grad.abs().mean().clamp(min=1e-5) + 5e-4
+        assert err.max() <= tol.item() * 20 or torch.allclose(
+            p.grad, q.grad, rtol=0.1, atol=1e-3
+        ), (n, err.max().item(), q.grad.abs().mean().item())
