@@ -1,0 +1,228 @@
+"""Lane-level CPU simulation of the EXPERIMENTAL chain kernels' index
+arithmetic (csrc/fwd_chain.hip / csrc/bwd_chain.hip).
+
+The MFMA fragment maps are probe-verified on hardware
+(tools/mfma_probe.hip); what can still be wrong in the kernels is how the
+maps are USED: fragment addressing, LDS strides, the D->tile epilogue and
+the in-place masking. This test ports those index expressions verbatim to
+numpy, emulates `v_mfma_f32_32x32x16_bf16` from the maps, and checks the
+full chains against plain matmul references. It runs on CPU in CI, so a
+refactor of the kernels' indexing that breaks the math fails here first.
+
+Mirrors (must be kept in sync with the .hip sources):
+  fc_layer / bc_layer loops, FC_S*/BC_S* strides, wave/lane decomposition,
+  the head dot-product, and the dz in-place mask.
+"""
+
+import numpy as np
+import pytest
+
+# ---- constants mirrored from the .hip sources ----
+MT = 64
+K0, K0P = 100, 112
+N1, N2, N3 = 512, 256, 128
+S0, S1, S2, S3 = K0P + 8, N1 + 8, N2 + 8, N3 + 8
+
+
+def mfma_32x32x16(a_frag, b_frag, acc):
+    """Emulate v_mfma_f32_32x32x16_bf16 from the probe-verified maps.
+    a_frag/b_frag: [64, 8] float; acc: [64, 16] float (updated)."""
+    A = np.zeros((32, 16), np.float32)
+    B = np.zeros((16, 32), np.float32)
+    for lane in range(64):
+        for j in range(8):
+            A[lane & 31, (lane >> 5) * 8 + j] = a_frag[lane, j]
+            B[(lane >> 5) * 8 + j, lane & 31] = b_frag[lane, j]
+    D = A @ B
+    for lane in range(64):
+        for reg in range(16):
+            row = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5)
+            acc[lane, reg] += D[row, lane & 31]
+    return acc
+
+
+def fc_layer_sim(src_lds, W, bias, dst_lds, K, N, SRC_S, DST_S, relu):
+    """Port of fc_layer: 4 waves x (2 m-tiles x NT n-tiles), k-loop."""
+    NT = N // 128
+    for wave in range(4):
+        n_base = wave * (N // 4)
+        acc = np.zeros((2, NT, 64, 16), np.float32)
+        for k in range(0, K, 16):
+            a = np.zeros((2, 64, 8), np.float32)
+            b = np.zeros((NT, 64, 8), np.float32)
+            for lane in range(64):
+                frag_k0 = (lane >> 5) * 8
+                ml = lane & 31
+                for mt in range(2):
+                    base = (mt * 32 + ml) * SRC_S + k + frag_k0
+                    a[mt, lane] = src_lds[base:base + 8]
+                for nt in range(NT):
+                    n = n_base + nt * 32 + ml
+                    b[nt, lane] = W[n, k + frag_k0:k + frag_k0 + 8]
+            for mt in range(2):
+                for nt in range(NT):
+                    mfma_32x32x16(a[mt], b[nt], acc[mt, nt])
+        for lane in range(64):
+            ml = lane & 31
+            for mt in range(2):
+                for nt in range(NT):
+                    n = n_base + nt * 32 + ml
+                    bv = bias[n]
+                    for reg in range(16):
+                        mrow = (mt * 32 + (reg & 3) + 8 * (reg >> 2)
+                                + 4 * (lane >> 5))
+                        v = acc[mt, nt, lane, reg] + bv
+                        if relu:
+                            v = max(v, 0.0)
+                        dst_lds[mrow * DST_S + n] = v
+
+
+def bc_layer_sim(dz_src, WT, a_dst, K, N, SRC_S, DST_S):
+    """Port of bc_layer: dgrad + in-place relu mask over a_dst."""
+    NT = N // 128
+    for wave in range(4):
+        n_base = wave * (N // 4)
+        acc = np.zeros((2, NT, 64, 16), np.float32)
+        for k in range(0, K, 16):
+            a = np.zeros((2, 64, 8), np.float32)
+            b = np.zeros((NT, 64, 8), np.float32)
+            for lane in range(64):
+                frag_k0 = (lane >> 5) * 8
+                ml = lane & 31
+                for mt in range(2):
+                    base = (mt * 32 + ml) * SRC_S + k + frag_k0
+                    a[mt, lane] = dz_src[base:base + 8]
+                for nt in range(NT):
+                    n = n_base + nt * 32 + ml
+                    b[nt, lane] = WT[n, k + frag_k0:k + frag_k0 + 8]
+            for mt in range(2):
+                for nt in range(NT):
+                    mfma_32x32x16(a[mt], b[nt], acc[mt, nt])
+        # NOTE: in-place mask per wave over its own n-range (no cross-wave
+        # overlap), as in the kernel.
+        for lane in range(64):
+            ml = lane & 31
+            for mt in range(2):
+                for nt in range(NT):
+                    n = n_base + nt * 32 + ml
+                    for reg in range(16):
+                        mrow = (mt * 32 + (reg & 3) + 8 * (reg >> 2)
+                                + 4 * (lane >> 5))
+                        idx = mrow * DST_S + n
+                        live = 1.0 if a_dst[idx] > 0.0 else 0.0
+                        a_dst[idx] = acc[mt, nt, lane, reg] * live
+
+
+@pytest.fixture(scope="module")
+def rng():
+    return np.random.default_rng(42)
+
+
+def test_mfma_emulation_is_matmul(rng):
+    a = rng.standard_normal((64, 8)).astype(np.float32)
+    b = rng.standard_normal((64, 8)).astype(np.float32)
+    acc = np.zeros((64, 16), np.float32)
+    mfma_32x32x16(a, b, acc)
+    # rebuild A, B and check acc holds A@B exactly per the D map
+    A = np.zeros((32, 16), np.float32)
+    B = np.zeros((16, 32), np.float32)
+    for lane in range(64):
+        for j in range(8):
+            A[lane & 31, (lane >> 5) * 8 + j] = a[lane, j]
+            B[(lane >> 5) * 8 + j, lane & 31] = b[lane, j]
+    D = A @ B
+    for lane in range(64):
+        for reg in range(16):
+            row = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5)
+            assert acc[lane, reg] == D[row, lane & 31]
+
+
+def test_fwd_chain_indexing(rng):
+    """Whole fwd chain on one 64-row slab vs numpy reference."""
+    x0 = rng.standard_normal((MT, K0)).astype(np.float32)
+    Ws = [rng.standard_normal(s).astype(np.float32) * 0.1
+          for s in [(N1, K0), (N2, N1), (N3, N2)]]
+    w4 = rng.standard_normal(N3).astype(np.float32) * 0.1
+    bs = [rng.standard_normal(n).astype(np.float32) * 0.1
+          for n in (N1, N2, N3)]
+    b4 = np.float32(0.3)
+
+    # reference
+    r1 = np.maximum(x0 @ Ws[0].T + bs[0], 0)
+    r2 = np.maximum(r1 @ Ws[1].T + bs[1], 0)
+    r3 = np.maximum(r2 @ Ws[2].T + bs[2], 0)
+    r_out = r3 @ w4 + b4
+
+    # simulate: stage x0 (zero-padded to K0P) like the kernel does
+    t0 = np.zeros(MT * S0, np.float32)
+    for m in range(MT):
+        t0[m * S0:m * S0 + K0] = x0[m]
+    W1p = np.zeros((N1, K0P), np.float32)
+    W1p[:, :K0] = Ws[0]
+    t1 = np.zeros(MT * S1, np.float32)
+    t2 = np.zeros(MT * S2, np.float32)
+    t3 = np.zeros(MT * S3, np.float32)
+    fc_layer_sim(t0, W1p, bs[0], t1, K0P, N1, S0, S1, True)
+    fc_layer_sim(t1, Ws[1], bs[1], t2, N1, N2, S1, S2, True)
+    fc_layer_sim(t2, Ws[2], bs[2], t3, N2, N3, S2, S3, True)
+    a1 = t1.reshape(MT, S1)[:, :N1]
+    a2 = t2.reshape(MT, S2)[:, :N2]
+    a3 = t3.reshape(MT, S3)[:, :N3]
+    np.testing.assert_allclose(a1, r1, rtol=1e-4, atol=1e-4)
+    np.testing.assert_allclose(a2, r2, rtol=1e-4, atol=1e-4)
+    np.testing.assert_allclose(a3, r3, rtol=1e-4, atol=1e-4)
+
+    # head: port of the 4-threads-per-row dot product
+    out = np.zeros(MT, np.float32)
+    for m in range(MT):
+        s = 0.0
+        for part in range(4):
+            for kk in range(32):
+                k = part * 32 + kk
+                s += t3[m * S3 + k] * w4[k]
+        out[m] = s + b4
+    np.testing.assert_allclose(out, r_out, rtol=1e-4, atol=1e-4)
+
+
+def test_bwd_chain_indexing(rng):
+    """Whole bwd chain on one 64-row slab vs numpy reference."""
+    a1 = np.maximum(rng.standard_normal((MT, N1)) - 0.2, 0).astype(
+        np.float32)
+    a2 = np.maximum(rng.standard_normal((MT, N2)) - 0.2, 0).astype(
+        np.float32)
+    a3 = np.maximum(rng.standard_normal((MT, N3)) - 0.2, 0).astype(
+        np.float32)
+    dy = rng.standard_normal(MT).astype(np.float32)
+    w4 = (rng.standard_normal(N3) * 0.1).astype(np.float32)
+    W3 = (rng.standard_normal((N3, N2)) * 0.1).astype(np.float32)
+    W2 = (rng.standard_normal((N2, N1)) * 0.1).astype(np.float32)
+
+    # reference
+    rz3 = np.outer(dy, w4) * (a3 > 0)
+    rz2 = (rz3 @ W3) * (a2 > 0)
+    rz1 = (rz2 @ W2) * (a1 > 0)
+
+    # simulate (tiles staged like bc_load_tile; strides BC_S*)
+    t1 = np.zeros(MT * S1, np.float32)
+    t2 = np.zeros(MT * S2, np.float32)
+    t3 = np.zeros(MT * S3, np.float32)
+    for m in range(MT):
+        t1[m * S1:m * S1 + N1] = a1[m]
+        t2[m * S2:m * S2 + N2] = a2[m]
+        t3[m * S3:m * S3 + N3] = a3[m]
+    # dz3 elementwise stage (port of the u-loop)
+    for u in range(MT * N3):
+        m, k = u // N3, u % N3
+        idx = m * S3 + k
+        live = 1.0 if t3[idx] > 0.0 else 0.0
+        t3[idx] = dy[m] * w4[k] * live
+    # dgrad layers with host-side pre-transposed weights:
+    # WT = W.t() with shape [in_features, out_features]
+    bc_layer_sim(t3, W3.T.copy(), t2, N3, N2, S3, S2)
+    bc_layer_sim(t2, W2.T.copy(), t1, N2, N1, S2, S1)
+    dz3 = t3.reshape(MT, S3)[:, :N3]
+    dz2 = t2.reshape(MT, S2)[:, :N2]
+    dz1 = t1.reshape(MT, S1)[:, :N1]
+    np.testing.assert_allclose(dz3, rz3, rtol=1e-4, atol=1e-5)
+    np.testing.assert_allclose(dz2, rz2, rtol=1e-4, atol=1e-5)
+    np.testing.assert_allclose(dz1, rz1, rtol=1e-4, atol=1e-5)
