@@ -31,6 +31,7 @@ GPU work runs on a dedicated side HIP stream so the trainer's compute stream
 never stalls behind shuffle work.
 """
 
+import contextlib
 import os
 import sys
 import threading
@@ -116,9 +117,7 @@ class ShuffleEngine:
         # RSDL_FORCE_COLLECTIVE=1 runs the full collective path even at
         # world 1 (self-exchange) — lets a single GPU exercise the exact
         # multi-GPU code path end to end.
-        import os as _os
-
-        force = _os.environ.get("RSDL_FORCE_COLLECTIVE") == "1"
+        force = os.environ.get("RSDL_FORCE_COLLECTIVE") == "1"
         self.distributed = initialized and (world > 1 or force)
         if self.distributed:
             if num_trainers != world:
@@ -377,7 +376,7 @@ class ShuffleEngine:
         ctx = (
             torch.cuda.stream(self._stream)
             if self._stream is not None
-            else _nullcontext()
+            else contextlib.nullcontext()
         )
         with ctx:
             for epoch in range(self.start_epoch, self.num_epochs):
@@ -426,10 +425,3 @@ class ShuffleEngine:
                     "shuffle engine worker failed"
                 ) from self._error
 
-
-class _nullcontext:
-    def __enter__(self):
-        return None
-
-    def __exit__(self, *a):
-        return False
