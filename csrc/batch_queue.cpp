@@ -51,6 +51,13 @@ struct QueueEmpty : std::runtime_error {
 struct QueueFull : std::runtime_error {
   using std::runtime_error::runtime_error;
 };
+// Raised by every op (including ones blocked in a CV wait when close() fires)
+// after the queue is closed — the analog of the reference's killed actor
+// surfacing RayActorError to blocked clients
+// (reference batch_queue.py:333-355).
+struct QueueClosed : std::runtime_error {
+  using std::runtime_error::runtime_error;
+};
 
 using Clock = std::chrono::steady_clock;
 
@@ -94,13 +101,15 @@ class BatchQueueCore {
     check_epoch(epoch);
     py::gil_scoped_release nogil;
     std::unique_lock<std::mutex> lk(mu_);
+    throw_if_closed();
     if ((int)curr_epochs_.size() == max_epochs_) {
       int first = curr_epochs_.front();
       curr_epochs_.pop_front();
       // Producers for every rank of the evicted epoch must be done...
-      cv_done_.wait(lk, [&] { return all_producers_done(first); });
+      cv_done_.wait(lk, [&] { return closed_ || all_producers_done(first); });
       // ...and every rank's queue must be fully joined.
-      cv_done_.wait(lk, [&] { return all_joined(first); });
+      cv_done_.wait(lk, [&] { return closed_ || all_joined(first); });
+      throw_if_closed();
     }
     curr_epochs_.push_back(epoch);
   }
@@ -108,9 +117,32 @@ class BatchQueueCore {
   void wait_until_all_epochs_done() {
     py::gil_scoped_release nogil;
     std::unique_lock<std::mutex> lk(mu_);
+    throw_if_closed();
     int last = num_epochs_ - 1;
-    cv_done_.wait(lk, [&] { return all_producers_done(last); });
-    cv_done_.wait(lk, [&] { return all_joined(last); });
+    cv_done_.wait(lk, [&] { return closed_ || all_producers_done(last); });
+    cv_done_.wait(lk, [&] { return closed_ || all_joined(last); });
+    throw_if_closed();
+  }
+
+  // Close the queue: every thread blocked in a CV wait wakes and raises
+  // Closed; every subsequent op raises Closed. Leftover items are dropped at
+  // destruction (the grid keeps its references so close() itself needs no
+  // GIL interaction).
+  void close() {
+    {
+      py::gil_scoped_release nogil;
+      std::lock_guard<std::mutex> lk(mu_);
+      closed_ = true;
+    }
+    cv_items_.notify_all();
+    cv_space_.notify_all();
+    cv_done_.notify_all();
+  }
+
+  bool closed() {
+    py::gil_scoped_release nogil;
+    std::lock_guard<std::mutex> lk(mu_);
+    return closed_;
   }
 
   // ----- producer side -----------------------------------------------------
@@ -118,9 +150,11 @@ class BatchQueueCore {
   void put(int rank, int epoch, py::object item, bool block, double timeout) {
     check(rank, epoch);
     PyObject* raw = item.release().ptr();  // we own one reference now
-    if (!put_raw(rank, epoch, raw, block, timeout)) {
+    int res = put_raw(rank, epoch, raw, block, timeout);
+    if (res != 0) {
       // Re-acquire ownership for proper decref (GIL is held again here).
       py::object steal = py::reinterpret_steal<py::object>(py::handle(raw));
+      if (res == 2) throw QueueClosed("BatchQueue has been shut down");
       throw QueueFull("Full");
     }
   }
@@ -143,13 +177,16 @@ class BatchQueueCore {
       raws.push_back(it.ptr());
     }
     bool ok = false;
+    bool was_closed = false;
     long long cur = 0;
     {
       py::gil_scoped_release nogil;
       std::lock_guard<std::mutex> lk(mu_);
       SubQueue& q = sub(rank, epoch);
       cur = (long long)q.items.size();
-      if (maxsize_ <= 0 || cur + (long long)raws.size() <= maxsize_) {
+      if (closed_) {
+        was_closed = true;
+      } else if (maxsize_ <= 0 || cur + (long long)raws.size() <= maxsize_) {
         for (PyObject* r : raws) {
           q.items.push_back(r);
           q.unfinished++;
@@ -161,6 +198,7 @@ class BatchQueueCore {
       cv_items_.notify_all();
     } else {
       for (PyObject* r : raws) Py_DECREF(r);
+      if (was_closed) throw QueueClosed("BatchQueue has been shut down");
       throw QueueFull("Cannot add " + std::to_string(items.size()) +
                       " items to queue of size " + std::to_string(cur) +
                       " and maxsize " + std::to_string(maxsize_) + ".");
@@ -172,7 +210,10 @@ class BatchQueueCore {
     // Sentinel goes through the blocking put path, exactly like the
     // reference's `await queue.put(None)` (batch_queue.py:420-422).
     Py_INCREF(Py_None);
-    put_raw(rank, epoch, Py_None, /*block=*/true, /*timeout=*/-1.0);
+    if (put_raw(rank, epoch, Py_None, /*block=*/true, /*timeout=*/-1.0) != 0) {
+      Py_DECREF(Py_None);
+      throw QueueClosed("BatchQueue has been shut down");
+    }
     {
       py::gil_scoped_release nogil;
       std::lock_guard<std::mutex> lk(mu_);
@@ -209,7 +250,8 @@ class BatchQueueCore {
       py::gil_scoped_release nogil;
       std::unique_lock<std::mutex> lk(mu_);
       SubQueue& q = sub(rank, epoch);
-      cv_items_.wait(lk, [&] { return !q.items.empty(); });
+      cv_items_.wait(lk, [&] { return closed_ || !q.items.empty(); });
+      if (q.items.empty()) throw_if_closed();
       while (!q.items.empty()) {
         raws.push_back(q.items.front());
         q.items.pop_front();
@@ -324,12 +366,13 @@ class BatchQueueCore {
     return true;
   }
 
-  // Blocking insert of an owned raw reference. Returns false on timeout /
-  // full (caller must decref). Called with the GIL HELD; releases it around
+  // Blocking insert of an owned raw reference. Returns 0 on success, 1 on
+  // timeout / full (caller must decref), 2 on queue closed (caller must
+  // decref and raise Closed). Called with the GIL HELD; releases it around
   // the wait.
-  bool put_raw(int rank, int epoch, PyObject* raw, bool block,
-               double timeout) {
-    bool ok = false;
+  int put_raw(int rank, int epoch, PyObject* raw, bool block,
+              double timeout) {
+    int res = 1;
     {
       py::gil_scoped_release nogil;
       std::unique_lock<std::mutex> lk(mu_);
@@ -337,38 +380,48 @@ class BatchQueueCore {
       auto has_space = [&] {
         return maxsize_ <= 0 || (long long)q.items.size() < maxsize_;
       };
-      if (!has_space()) {
-        if (!block) {
-          // fallthrough: ok stays false
-        } else if (timeout >= 0.0) {
+      auto ready = [&] { return closed_ || has_space(); };
+      if (!closed_ && !has_space() && block) {
+        if (timeout >= 0.0) {
           cv_space_.wait_for(
-              lk, std::chrono::duration<double>(timeout), has_space);
+              lk, std::chrono::duration<double>(timeout), ready);
         } else {
-          cv_space_.wait(lk, has_space);
+          cv_space_.wait(lk, ready);
         }
       }
-      if (has_space()) {
+      if (closed_) {
+        res = 2;
+      } else if (has_space()) {
         q.items.push_back(raw);
         q.unfinished++;
-        ok = true;
+        res = 0;
       }
     }
-    if (ok) cv_items_.notify_all();
-    return ok;
+    if (res == 0) cv_items_.notify_all();
+    return res;
   }
 
-  // Wait until q non-empty; caller holds lk. Returns false on timeout/empty.
+  // Wait until q non-empty; caller holds lk. Returns false on timeout/empty;
+  // throws Closed if the queue closes while waiting (or already was).
   bool wait_nonempty(std::unique_lock<std::mutex>& lk, SubQueue& q, bool block,
                      double timeout) {
     if (!q.items.empty()) return true;
+    throw_if_closed();
     if (!block) return false;
+    auto ready = [&] { return closed_ || !q.items.empty(); };
     if (timeout >= 0.0) {
-      cv_items_.wait_for(lk, std::chrono::duration<double>(timeout),
-                         [&] { return !q.items.empty(); });
-      return !q.items.empty();
+      cv_items_.wait_for(lk, std::chrono::duration<double>(timeout), ready);
+    } else {
+      cv_items_.wait(lk, ready);
     }
-    cv_items_.wait(lk, [&] { return !q.items.empty(); });
-    return true;
+    if (!q.items.empty()) return true;
+    throw_if_closed();
+    return false;
+  }
+
+  // Caller must hold mu_.
+  void throw_if_closed() const {
+    if (closed_) throw QueueClosed("BatchQueue has been shut down");
   }
 
   const int max_epochs_;
@@ -377,6 +430,7 @@ class BatchQueueCore {
   const long long maxsize_;
   std::vector<std::vector<SubQueue>> grid_;
   std::deque<int> curr_epochs_;
+  bool closed_ = false;
   std::mutex mu_;
   std::condition_variable cv_items_;
   std::condition_variable cv_space_;
@@ -392,6 +446,7 @@ PYBIND11_MODULE(_rsdl_cpp, m) {
 
   py::register_exception<QueueEmpty>(m, "Empty");
   py::register_exception<QueueFull>(m, "Full");
+  py::register_exception<QueueClosed>(m, "Closed", PyExc_RuntimeError);
 
   py::class_<BatchQueueCore>(m, "BatchQueueCore")
       .def(py::init<int, int, int, long long>(),
@@ -400,6 +455,8 @@ PYBIND11_MODULE(_rsdl_cpp, m) {
       .def("new_epoch", &BatchQueueCore::new_epoch, py::arg("epoch"))
       .def("wait_until_all_epochs_done",
            &BatchQueueCore::wait_until_all_epochs_done)
+      .def("close", &BatchQueueCore::close)
+      .def_property_readonly("is_closed", &BatchQueueCore::closed)
       .def("put", &BatchQueueCore::put, py::arg("rank"), py::arg("epoch"),
            py::arg("item"), py::arg("block") = true, py::arg("timeout") = -1.0)
       .def("put_batch", &BatchQueueCore::put_batch, py::arg("rank"),

@@ -33,6 +33,7 @@ from typing import Any, Iterable, Optional
 
 from ray_shuffling_data_loader_amd._rsdl_cpp import (  # noqa: F401
     BatchQueueCore,
+    Closed,
     Empty,
     Full,
 )
@@ -43,8 +44,22 @@ _HDR = struct.Struct("<Q")
 
 
 def _queue_socket_path(name: str) -> str:
+    """Socket path inside a per-uid mode-0700 directory: the server handler
+    unpickles requests, so the rendezvous must not be dialable by other
+    local users (a world-writable /tmp socket would hand them arbitrary
+    code execution in the trainer process)."""
     base = os.environ.get("RSDL_QUEUE_DIR", tempfile.gettempdir())
-    return os.path.join(base, f"rsdl_queue_{name}.sock")
+    sockdir = os.path.join(base, f"rsdl_queue_u{os.getuid()}")
+    os.makedirs(sockdir, mode=0o700, exist_ok=True)
+    st = os.stat(sockdir)
+    if st.st_uid != os.getuid():
+        raise RuntimeError(
+            f"queue socket directory {sockdir} is owned by uid {st.st_uid}, "
+            "not us; refusing to rendezvous there (set RSDL_QUEUE_DIR)"
+        )
+    if st.st_mode & 0o077:
+        os.chmod(sockdir, 0o700)
+    return os.path.join(sockdir, f"rsdl_queue_{name}.sock")
 
 
 def _recv_exact(sock: socket.socket, n: int) -> bytes:
@@ -106,6 +121,7 @@ def _recv_msg(sock: socket.socket) -> Any:
 _EXC_BY_NAME = {
     "Empty": Empty,
     "Full": Full,
+    "Closed": Closed,
     "ValueError": ValueError,
     "IndexError": IndexError,
     "RuntimeError": RuntimeError,
@@ -205,6 +221,7 @@ class BatchQueue:
                 self._server = _ThreadedQueueServer(
                     path, _QueueRequestHandler
                 )
+                os.chmod(path, 0o600)  # server unpickles: owner-only
                 self._server.core = self.core
                 self._server_thread = threading.Thread(
                     target=self._server.serve_forever,
@@ -247,8 +264,13 @@ class BatchQueue:
 
     def shutdown(self, force: bool = False, grace_period_s: int = 5) -> None:
         """Tear down the queue (server + core). Parity with reference
-        batch_queue.py:333-355 (graceful terminate then kill)."""
+        batch_queue.py:333-355: like the reference's killed actor, any
+        consumer/producer blocked inside a queue op wakes up and raises
+        (``Closed``, a RuntimeError subclass) instead of hanging forever."""
         del force, grace_period_s  # no child process to kill; kept for parity
+        if self.core is not None:
+            # Wake every thread blocked in get/put/new_epoch/join waits.
+            self.core.close()
         if self._server is not None:
             self._server.shutdown()
             self._server.server_close()

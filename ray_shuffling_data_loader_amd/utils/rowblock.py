@@ -79,6 +79,24 @@ class RowBlock:
     def pin_memory(self) -> "RowBlock":
         return RowBlock({k: t.pin_memory() for k, t in self.columns.items()})
 
+    def record_stream(self, stream: "torch.cuda.Stream") -> None:
+        """Mark every CUDA column as in use on ``stream``. The shuffle engine
+        allocates partitions on its side HIP stream; without this, dropping a
+        batch whose consumer-stream kernels are still in flight lets the
+        caching allocator hand the block back to the side-stream pool, where
+        the pipelined next-epoch shuffle (max_concurrent_epochs=2) could
+        overwrite it mid-read. Columns often share one storage (views mode);
+        record_stream is per-storage underneath, so duplicates are cheap."""
+        seen = set()
+        for t in self.columns.values():
+            if t.device.type != "cuda":
+                continue
+            key = t.untyped_storage().data_ptr()
+            if key in seen:
+                continue
+            seen.add(key)
+            t.record_stream(stream)
+
     def contiguous(self) -> "RowBlock":
         return RowBlock({k: t.contiguous() for k, t in self.columns.items()})
 

@@ -429,8 +429,19 @@ def test_custom_queue_placement(tmp_path, monkeypatch):
     name = "placed_queue"
     q = BatchQueue(1, 1, 1, name=name)
     try:
-        socks = [f for f in os.listdir(tmp_path) if name in f]
+        socks = [
+            os.path.join(root, f)
+            for root, _, files in os.walk(tmp_path)
+            for f in files
+            if name in f
+        ]
         assert socks, f"socket not under RSDL_QUEUE_DIR: {os.listdir(tmp_path)}"
+        # The rendezvous dir must be private (the server unpickles requests).
+        sockdir = os.path.dirname(socks[0])
+        assert os.stat(sockdir).st_mode & 0o077 == 0, oct(
+            os.stat(sockdir).st_mode
+        )
+        assert os.stat(socks[0]).st_mode & 0o077 == 0
         c = BatchQueue(1, 1, 1, name=name, connect=True)
         q.new_epoch(0)
         c.put(0, 0, 42)
@@ -501,3 +512,89 @@ def test_queue_concurrent_stress():
     assert not errors, errors
     for k in sent:
         assert sorted(got[k]) == sorted(sent[k]), k
+
+
+def test_shutdown_unblocks_blocked_get():
+    """A consumer blocked in get() must raise when the queue is shut down
+    from another thread, like the reference's killed actor surfacing an
+    actor error to blocked clients (reference batch_queue.py:333-355)."""
+    from ray_shuffling_data_loader_amd.batch_queue import Closed
+
+    q = make_queue(num_epochs=2, max_concurrent_epochs=2)
+    q.new_epoch(0)
+    errors = []
+
+    def consumer():
+        try:
+            q.core.get(0, 0, True, -1.0)
+        except Exception as e:  # noqa: BLE001
+            errors.append(e)
+
+    t = threading.Thread(target=consumer, daemon=True)
+    t.start()
+    time.sleep(0.2)
+    q.shutdown()
+    t.join(timeout=5)
+    assert not t.is_alive(), "blocked consumer did not wake on shutdown"
+    assert errors and isinstance(errors[0], Closed)
+    assert isinstance(errors[0], RuntimeError)  # Closed subclasses it
+
+
+def test_close_unblocks_blocked_put_and_window():
+    """close() also wakes producers blocked on maxsize backpressure and the
+    driver blocked in the new_epoch eviction gate."""
+    from ray_shuffling_data_loader_amd.batch_queue import Closed
+
+    q = make_queue(num_epochs=3, max_concurrent_epochs=1, maxsize=1)
+    q.new_epoch(0)
+    q.put(0, 0, "a")
+    errors = []
+
+    def producer():
+        try:
+            q.put(0, 0, "b", block=True)
+        except Exception as e:  # noqa: BLE001
+            errors.append(e)
+
+    def driver():
+        try:
+            q.new_epoch(1)  # epoch 0 never joins -> blocks on the window
+        except Exception as e:  # noqa: BLE001
+            errors.append(e)
+
+    threads = [
+        threading.Thread(target=producer, daemon=True),
+        threading.Thread(target=driver, daemon=True),
+    ]
+    for t in threads:
+        t.start()
+    time.sleep(0.2)
+    q.core.close()
+    for t in threads:
+        t.join(timeout=5)
+        assert not t.is_alive(), "blocked thread did not wake on close"
+    assert len(errors) == 2
+    assert all(isinstance(e, Closed) for e in errors)
+
+
+def test_ops_after_close_raise():
+    from ray_shuffling_data_loader_amd.batch_queue import Closed
+
+    q = make_queue(num_epochs=2, max_concurrent_epochs=2)
+    q.new_epoch(0)
+    q.put(0, 0, "x")
+    core = q.core
+    core.close()
+    assert core.is_closed
+    with pytest.raises(Closed):
+        core.put(0, 0, "y", True, -1.0)
+    with pytest.raises(Closed):
+        core.put_nowait_batch(0, 0, ["y"])
+    with pytest.raises(Closed):
+        core.get_batch(0, 1)
+    with pytest.raises(Closed):
+        core.new_epoch(1)
+    with pytest.raises(Closed):
+        core.producer_done(0, 0)
+    with pytest.raises(Closed):
+        core.wait_until_all_epochs_done()
