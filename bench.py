@@ -52,6 +52,15 @@ if os.environ.get("RSDL_TUNABLEOP", "1") == "1":
     os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", _tuned_dst)
     os.environ.setdefault("PYTORCH_TUNABLEOP_VERBOSE", "0")
 
+# First-contact diagnostics for multi-GPU runs: RCCL warnings/errors go to
+# stderr (NCCL_DEBUG must be set before the communicator is created), and a
+# watchdog dumps all ranks' Python stacks if a step wedges (RCCL hangs
+# surface as silent stalls, not exceptions). RSDL_NCCL_DEBUG=INFO for full
+# per-collective logs; RSDL_WATCHDOG_S=0 disables the stack dump.
+os.environ.setdefault(
+    "NCCL_DEBUG", os.environ.get("RSDL_NCCL_DEBUG", "WARN")
+)
+
 import torch
 
 from ray_shuffling_data_loader_amd.data_generation import float_data_spec
@@ -146,7 +155,10 @@ def main():
                         i,
                         i * rows_per_file,
                         rows_per_file,
-                        1,
+                        # ~390k-row row groups: the ingest parallelism unit
+                        # (row-group-granular read tasks) — one giant row
+                        # group per file would serialize uncached decode.
+                        max(1, rows_per_file // 390_625),
                         shard_dir,
                         spec=spec,
                         include_key=False,
@@ -166,8 +178,17 @@ def main():
         torch.distributed.barrier()
 
     # ----- loader + model --------------------------------------------------
+    # The timed window must contain >= 1 epoch rollover so the measured
+    # number includes the full per-epoch reshuffle the metric claims (the
+    # pipelined epoch e+1 shuffle must actually be ready when epoch e runs
+    # out). After warmup we burn extra UNTIMED steps until the epoch
+    # boundary is at most floor(steps/2) steps ahead, positioning the
+    # rollover inside the timed region at any --steps >= 2. This
+    # over-weights reshuffle cost vs true steady state (1 rollover per
+    # steps_per_epoch=50 steps), i.e. it is conservative.
     steps_per_epoch = args.rows_per_gpu // args.batch_size
-    total_steps = args.warmup + args.steps
+    burn_max = steps_per_epoch  # at most one epoch of positioning
+    total_steps = args.warmup + burn_max + args.steps
     num_epochs = (total_steps + steps_per_epoch - 1) // steps_per_epoch + 1
 
     feature_columns = [f"f{i}" for i in range(args.num_cols)]
@@ -209,9 +230,10 @@ def main():
         for epoch in range(num_epochs):
             ds.set_epoch(epoch)
             for item in ds:
-                yield item
+                yield epoch, item
 
     it = batches()
+    cur_epoch = [0]
     is_cuda = device.type == "cuda"
 
     def sync():
@@ -233,7 +255,7 @@ def main():
 
     def one_step():
         t_wait0 = time.perf_counter()
-        data, target = next(it)
+        cur_epoch[0], (data, target) = next(it)
         wait = time.perf_counter() - t_wait0
         x = data[0]
         if x.device != device:
@@ -288,7 +310,7 @@ def main():
 
     def one_step_graphed():
         t_wait0 = time.perf_counter()
-        data, target = next(it)
+        cur_epoch[0], (data, target) = next(it)
         wait = time.perf_counter() - t_wait0
         x = data[0]
         if "g" not in graph_state:
@@ -301,8 +323,60 @@ def main():
     if use_graph:
         one_step = one_step_graphed
 
-    for _ in range(args.warmup):
-        one_step()
+    # Hang watchdog: if the whole job wedges (e.g. an RCCL collective
+    # deadlock at N>1), dump every thread's stack to stderr so the failure
+    # is diagnosable from one run. Armed around the measured region.
+    import faulthandler
+
+    watchdog_s = float(os.environ.get("RSDL_WATCHDOG_S", "300"))
+    if watchdog_s > 0:
+        faulthandler.dump_traceback_later(
+            watchdog_s, repeat=True, exit=False
+        )
+
+    def progress(stage, i):
+        if os.environ.get("RSDL_VERBOSE") == "1":
+            print(
+                f"[bench] rank {rank} {stage} step {i} "
+                f"epoch {cur_epoch[0]}",
+                file=sys.stderr,
+                flush=True,
+            )
+
+    # Track position within the current epoch by observing actual epoch
+    # transitions (robust to the +-1 batch-count variation of the
+    # distributed binomial row split, where nominal arithmetic drifts).
+    last_epoch = [0]
+    in_epoch = [0]
+
+    def step_tracked():
+        w = one_step()
+        if cur_epoch[0] != last_epoch[0]:
+            last_epoch[0] = cur_epoch[0]
+            in_epoch[0] = 1
+        else:
+            in_epoch[0] += 1
+        return w
+
+    try:
+        for i in range(args.warmup):
+            step_tracked()
+            progress("warmup", i)
+        # Position the epoch boundary inside the timed window (see above).
+        burned = 0
+        target_pos = steps_per_epoch - max(1, args.steps // 2)
+        while in_epoch[0] < target_pos and burned < burn_max:
+            step_tracked()
+            burned += 1
+        progress("burn", burned)
+    except Exception:
+        print(
+            f"[bench] rank {rank} FAILED in warmup/burn at epoch "
+            f"{cur_epoch[0]}",
+            file=sys.stderr,
+            flush=True,
+        )
+        raise
 
     sync()
     barrier()
@@ -319,21 +393,36 @@ def main():
         prof.__enter__()
     t0 = time.perf_counter()
     waits = []
-    for i in range(args.steps):
-        waits.append(one_step())
-        if prof is not None and i == 2:
-            prof.__exit__(None, None, None)
-            print(
-                prof.key_averages().table(
-                    sort_by="self_cpu_time_total", row_limit=18
-                ),
-                flush=True,
-            )
-            prof = None
+    rollovers = 0
+    try:
+        for i in range(args.steps):
+            e_before = cur_epoch[0]
+            waits.append(step_tracked())
+            if cur_epoch[0] != e_before:
+                rollovers += 1
+            if prof is not None and i == 2:
+                prof.__exit__(None, None, None)
+                print(
+                    prof.key_averages().table(
+                        sort_by="self_cpu_time_total", row_limit=18
+                    ),
+                    flush=True,
+                )
+                prof = None
+    except Exception:
+        print(
+            f"[bench] rank {rank} FAILED in timed loop at step "
+            f"{len(waits)} epoch {cur_epoch[0]}",
+            file=sys.stderr,
+            flush=True,
+        )
+        raise
     sync()
     barrier()
     sync()
     elapsed = time.perf_counter() - t0
+    if watchdog_s > 0:
+        faulthandler.cancel_dump_traceback_later()
     if os.environ.get("RSDL_DEBUG_WAITS") == "1" and rank == 0:
         top = sorted(enumerate(waits), key=lambda kv: -kv[1])[:5]
         print(
@@ -398,6 +487,12 @@ def main():
                 "num_reducers": args.reducers_per_gpu * world,
                 "max_concurrent_epochs": args.max_concurrent_epochs,
                 "parallelism": f"dp{n_gpus}",
+                # Proof the timed region crossed >= 1 epoch boundary, i.e.
+                # the number includes a full per-epoch reshuffle: untimed
+                # positioning steps burned after warmup, and the count of
+                # epoch rollovers observed inside the timed window.
+                "burn_steps": burned,
+                "epoch_rollovers_in_timed_window": rollovers,
                 "p50_batch_wait_ms": p50_wait_ms,
                 "mean_batch_wait_ms": (
                     sum(waits) / len(waits) * 1000 if waits else None

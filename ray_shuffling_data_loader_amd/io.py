@@ -7,8 +7,9 @@ exchange layout — on GPU via the pack_columns HIP kernel when a device is
 given, else via numpy strided views.
 """
 
+import os
 import warnings
-from concurrent.futures import ThreadPoolExecutor
+from concurrent.futures import FIRST_COMPLETED, ThreadPoolExecutor, wait
 from typing import Dict, List, Optional, Sequence, Tuple
 
 import numpy as np
@@ -113,6 +114,36 @@ def read_file_columns(
     return out
 
 
+def _read_row_groups_columns(
+    filename: str,
+    groups: List[int],
+    schema: Schema,
+    pin: bool,
+) -> Dict[str, torch.Tensor]:
+    """Decode a run of row groups of one file into column tensors (single
+    Arrow copy into pinned memory when ``pin``). use_threads=False: task
+    granularity already saturates the reader pool; nested Arrow threads
+    only oversubscribe."""
+    pf = pq.ParquetFile(filename)
+    table = pf.read_row_groups(
+        groups, columns=schema.names, use_threads=False
+    )
+    out = {}
+    for spec in schema.columns:
+        arr = table.column(spec.name).to_numpy(zero_copy_only=False)
+        if not arr.flags["C_CONTIGUOUS"]:
+            arr = np.ascontiguousarray(arr)
+        with warnings.catch_warnings():
+            warnings.simplefilter("ignore")
+            t = torch.from_numpy(arr)
+        if pin:
+            p = torch.empty_like(t, pin_memory=True)
+            p.copy_(t)
+            t = p
+        out[spec.name] = t
+    return out
+
+
 def read_files_packed(
     filenames: Sequence[str],
     schema: Schema,
@@ -120,40 +151,112 @@ def read_files_packed(
     reader_threads: int = 8,
 ) -> torch.Tensor:
     """Read + pack many Parquet files into one [N, row_stride] uint8 tensor
-    on ``device``. Reads are threaded (Arrow releases the GIL); packing runs
-    on the GPU (pack_columns kernel) when device is cuda."""
+    on ``device``.
+
+    Tasks are **row-group-granular** (the reference's parallelism unit is
+    one map task per file, reference shuffle.py:111-114; file-level tasks
+    capped effective reader parallelism at files-per-rank threads — the
+    round-1 uncached-ingest bottleneck). Each task decodes one run of row
+    groups; destination offsets are precomputed from Parquet metadata so
+    tasks complete and pack OUT OF ORDER into disjoint slices of the
+    preallocated block (no torch.cat, no 2x transient memory).
+
+    In-flight tasks are bounded to ~2x reader_threads (RSDL_READ_WINDOW):
+    decoded pinned column sets can never accumulate beyond the window even
+    when the consumer side falls behind the readers.
+
+    GPU path: worker threads do Arrow decode + one copy into pinned
+    buffers; the caller thread issues async H2D DMAs + pack kernels as each
+    task lands (torch's caching host allocator keeps each pinned block
+    alive until its copy's stream work completes). CPU path: workers pack
+    directly into their disjoint output slices — the pack itself
+    parallelizes.
+    """
     if not filenames:
         return torch.empty(0, schema.row_stride, dtype=torch.uint8,
                            device=device)
     use_gpu = device.type == "cuda"
 
-    # Reader threads do Arrow decode + the single copy into pinned-host
-    # arenas in parallel; the caller thread only issues async H2D DMAs and
-    # pack kernels, which overlap the remaining reads (torch's caching host
-    # allocator keeps each pinned block alive until its copy's stream work
-    # completes).
-    def load(fn):
-        return read_file_columns(fn, schema, pin=use_gpu)
+    # Task list: (filename, row-group run, dest row offset, n_rows). Runs
+    # are split so ~>= 2 tasks per reader thread exist when possible, but
+    # never below one row group (Parquet's decode granularity).
+    per_file_groups: List[List[int]] = []
+    per_file_sizes: List[List[int]] = []
+    total = 0
+    for fn in filenames:
+        md = pq.ParquetFile(fn).metadata
+        sizes = [md.row_group(g).num_rows for g in range(md.num_row_groups)]
+        per_file_groups.append(list(range(md.num_row_groups)))
+        per_file_sizes.append(sizes)
+        total += sum(sizes)
 
-    # Preallocate the full source block and pack each file's rows into its
-    # slice — no torch.cat, no 2x transient memory at ingest.
-    counts = [pq.ParquetFile(fn).metadata.num_rows for fn in filenames]
-    total = sum(counts)
+    tasks = []  # (filename, [group_ids], dest_off, n_rows)
+    off = 0
+    for fn, groups, sizes in zip(filenames, per_file_groups, per_file_sizes):
+        for g, n in zip(groups, sizes):
+            tasks.append((fn, [g], off, n))
+            off += n
+    # Coalesce adjacent tiny row groups of the same file so task overhead
+    # stays negligible (>= ~250k rows per task unless the file is smaller).
+    min_rows = int(os.environ.get("RSDL_MIN_TASK_ROWS", "250000"))
+    coalesced = []
+    for t in tasks:
+        if (
+            coalesced
+            and coalesced[-1][0] == t[0]
+            and coalesced[-1][3] < min_rows
+            and coalesced[-1][2] + coalesced[-1][3] == t[2]
+        ):
+            prev = coalesced[-1]
+            coalesced[-1] = (prev[0], prev[1] + t[1], prev[2], prev[3] + t[3])
+        else:
+            coalesced.append(t)
+    tasks = coalesced
+
     packed = torch.empty(
         total, schema.row_stride, dtype=torch.uint8, device=device
     )
     if device.type == "cpu":
         packed.zero_()  # CPU path packs payload only; keep padding defined
-    off = 0
+
+    window = int(
+        os.environ.get("RSDL_READ_WINDOW", str(2 * max(1, reader_threads)))
+    )
+
+    if not use_gpu:
+        # CPU: decode AND pack inside the workers (disjoint output slices).
+        def load_pack_cpu(task):
+            fn, groups, dst, n_rows = task
+            cols = _read_row_groups_columns(fn, groups, schema, pin=False)
+            pack_columns(cols, schema, out=packed[dst : dst + n_rows])
+
+        with ThreadPoolExecutor(max_workers=max(1, reader_threads)) as pool:
+            list(pool.map(load_pack_cpu, tasks))
+        return packed
+
+    def load(task):
+        fn, groups, dst, n_rows = task
+        return dst, n_rows, _read_row_groups_columns(
+            fn, groups, schema, pin=True
+        )
+
     with ThreadPoolExecutor(max_workers=max(1, reader_threads)) as pool:
-        for n_rows, cols_host in zip(counts, pool.map(load, filenames)):
-            if use_gpu:
+        pending = set()
+        it = iter(tasks)
+        for task in it:
+            pending.add(pool.submit(load, task))
+            if len(pending) >= window:
+                break
+        while pending:
+            done, pending = wait(pending, return_when=FIRST_COMPLETED)
+            for fut in done:
+                dst, n_rows, cols_host = fut.result()
                 cols = {
                     name: t.to(device, non_blocking=True)
                     for name, t in cols_host.items()
                 }
-            else:
-                cols = cols_host
-            pack_columns(cols, schema, out=packed[off : off + n_rows])
-            off += n_rows
+                pack_columns(cols, schema, out=packed[dst : dst + n_rows])
+                nxt = next(it, None)
+                if nxt is not None:
+                    pending.add(pool.submit(load, nxt))
     return packed
