@@ -72,15 +72,17 @@ def _recv_exact(sock: socket.socket, n: int) -> bytes:
     return buf
 
 
-# RSDL_SHM_QUEUE=1: serialize messages with torch.multiprocessing's
-# ForkingPickler so tensor PAYLOADS cross the socket as shared-memory /
-# device-IPC handles instead of bytes (the socket carries only metadata).
-# CPU storages move to shm files (sharing strategy "file_system" — the
-# "file_descriptor" default cannot pass fds between UNRELATED processes,
-# which is exactly what the named-queue rendezvous connects); CUDA
-# storages travel as dmabuf IPC handles (HSA_ENABLE_IPC_MODE_LEGACY=0).
-# Default off: plain pickle is simpler and the collective mode (the
-# production path) never serializes batches at all.
+# RSDL_SHM_QUEUE (default ON): serialize messages with
+# torch.multiprocessing's ForkingPickler so tensor PAYLOADS cross the
+# socket as shared-memory / device-IPC handles instead of bytes (the
+# socket carries only metadata) — the analog of the reference's plasma
+# zero-copy get (reference dataset.py:136-139). CPU storages move to shm
+# files (sharing strategy "file_system" — the "file_descriptor" default
+# cannot pass fds between UNRELATED processes, which is exactly what the
+# named-queue rendezvous connects); CUDA storages travel as dmabuf IPC
+# handles (HSA_ENABLE_IPC_MODE_LEGACY=0). Set RSDL_SHM_QUEUE=0 for plain
+# pickle. The collective mode (the production path) never serializes
+# batches at all.
 _FP = None
 
 
@@ -88,7 +90,7 @@ def _shm_pickler():
     """Lazy setup of the ForkingPickler path (checked per call so tests can
     toggle RSDL_SHM_QUEUE without re-importing the module)."""
     global _FP
-    if os.environ.get("RSDL_SHM_QUEUE", "0") != "1":
+    if os.environ.get("RSDL_SHM_QUEUE", "1") != "1":
         return None
     if _FP is None:
         import torch.multiprocessing as _tmp  # registers torch reducers

@@ -66,6 +66,9 @@ def shuffle(
     return engine.run()
 
 
+_EPOCH_ENGINES = {}
+
+
 def shuffle_epoch(
     epoch: int,
     filenames: Sequence[str],
@@ -73,18 +76,34 @@ def shuffle_epoch(
     num_reducers: int,
     num_trainers: int,
     stats_collector=None,
+    engine: ShuffleEngine = None,
     **engine_kwargs,
 ) -> None:
-    """Shuffle one epoch into the consumer (reference shuffle.py:89-126)."""
-    engine = ShuffleEngine(
-        filenames,
-        batch_consumer,
-        num_epochs=epoch + 1,
-        num_reducers=num_reducers,
-        num_trainers=num_trainers,
-        stats_collector=stats_collector,
-        **engine_kwargs,
-    )
+    """Shuffle one epoch into the consumer (reference shuffle.py:89-126).
+
+    Calls with the same (filenames, reducers, trainers) reuse one cached
+    engine, so the source stays resident (HBM/host per ``source_cache``)
+    instead of being re-read and re-packed per invocation — calling this in
+    a loop costs one ingest total, like :class:`ShuffleEngine` itself.
+    Pass ``engine=`` to manage the instance explicitly.
+    """
+    if engine is None:
+        key = (tuple(filenames), num_reducers, num_trainers)
+        engine = _EPOCH_ENGINES.get(key)
+        if engine is None:
+            engine = ShuffleEngine(
+                filenames,
+                batch_consumer,
+                num_epochs=epoch + 1,
+                num_reducers=num_reducers,
+                num_trainers=num_trainers,
+                stats_collector=stats_collector,
+                **engine_kwargs,
+            )
+            _EPOCH_ENGINES[key] = engine
+        engine.num_epochs = max(engine.num_epochs, epoch + 1)
+        engine.consumer = batch_consumer
+        engine.stats = stats_collector
     engine._shuffle_epoch(epoch)
 
 

@@ -384,11 +384,22 @@ def test_wgrad_kernel_matches_reference(dev):
         dw, db = wgrad(dy, x, with_bias=True)
         ref_dw = dy.t().float() @ x.float()
         ref_db = dy.float().sum(0)
-        # fp32 accumulation in both; ordering differs -> tiny tolerance
-        assert torch.allclose(dw, ref_dw, atol=2.0, rtol=1e-2), (
-            m, n, k, (dw - ref_dw).abs().max().item(),
-        )
-        assert torch.allclose(db, ref_db, atol=2.0, rtol=1e-2), (m, n, k)
+        # Both sides use the SAME bf16 inputs with fp32 accumulation, so
+        # the only discrepancy is accumulation ORDER. For two tree-ish fp32
+        # reductions of M products of standard normals, the rounding
+        # discrepancy is bounded by ~2*log2(M)*eps32*sum|terms| with
+        # sum|terms| ~= E|ab|*M = (2/pi)*M; allow 4x slop for split-K
+        # partials. This scales with the reduction length instead of a
+        # flat atol, so regressions at small M can't hide.
+        import math
+
+        eps32 = 2.0 ** -24
+        tol_dw = 8 * math.log2(m) * eps32 * (2 / math.pi) * m
+        tol_db = 8 * math.log2(m) * eps32 * 0.8 * m
+        err_dw = (dw - ref_dw).abs().max().item()
+        err_db = (db - ref_db).abs().max().item()
+        assert err_dw <= tol_dw, (m, n, k, err_dw, tol_dw)
+        assert err_db <= tol_db, (m, n, k, err_db, tol_db)
 
 
 def test_collective_path_single_gpu(dev, tmp_path, monkeypatch):
