@@ -108,31 +108,57 @@ __device__ void bc_layer(const short* __restrict__ dz_src,
                          short* __restrict__ a_dst, int32_t wave,
                          int32_t lane) {
   constexpr int NT = N / 128;
+  constexpr int ITERS = K / 16;
   const int32_t n_base = wave * (N / 4);
   const int32_t frag_k0 = (lane >> 5) * 8;
   const int32_t ml = lane & 31;
 
+  // Software-pipelined k-loop (same rationale as fc_layer: 1 wave/SIMD —
+  // the next fragments must be in flight during the current MFMAs).
+  const short* srcA[2] = {&dz_src[ml * SRC_S + frag_k0],
+                          &dz_src[(32 + ml) * SRC_S + frag_k0]};
+  const short* srcB[NT];
+  #pragma unroll
+  for (int nt = 0; nt < NT; nt++) {
+    srcB[nt] = &WT[(int64_t)(n_base + nt * 32 + ml) * K + frag_k0];
+  }
+
   bc_f32x16 acc[2][NT] = {};
-  for (int32_t k = 0; k < K; k += 16) {
-    bc_bf16x8 a[2], b[NT];
-    #pragma unroll
-    for (int mt = 0; mt < 2; mt++) {
-      *reinterpret_cast<uint4*>(&a[mt]) =
-          *reinterpret_cast<const uint4*>(
-              &dz_src[(mt * 32 + ml) * SRC_S + k + frag_k0]);
-    }
-    #pragma unroll
-    for (int nt = 0; nt < NT; nt++) {
-      const int32_t n = n_base + nt * 32 + ml;
-      *reinterpret_cast<uint4*>(&b[nt]) =
-          *reinterpret_cast<const uint4*>(&WT[(int64_t)n * K + k + frag_k0]);
+  bc_bf16x8 a[2][2], b[2][NT];
+
+  #pragma unroll
+  for (int mt = 0; mt < 2; mt++) {
+    *reinterpret_cast<uint4*>(&a[0][mt]) =
+        *reinterpret_cast<const uint4*>(srcA[mt]);
+  }
+  #pragma unroll
+  for (int nt = 0; nt < NT; nt++) {
+    *reinterpret_cast<uint4*>(&b[0][nt]) =
+        *reinterpret_cast<const uint4*>(srcB[nt]);
+  }
+  #pragma unroll
+  for (int i = 0; i < ITERS; i++) {
+    const int cur = i & 1;
+    const int nxt = cur ^ 1;
+    if (i + 1 < ITERS) {
+      const int32_t k = (i + 1) * 16;
+      #pragma unroll
+      for (int mt = 0; mt < 2; mt++) {
+        *reinterpret_cast<uint4*>(&a[nxt][mt]) =
+            *reinterpret_cast<const uint4*>(&srcA[mt][k]);
+      }
+      #pragma unroll
+      for (int nt = 0; nt < NT; nt++) {
+        *reinterpret_cast<uint4*>(&b[nxt][nt]) =
+            *reinterpret_cast<const uint4*>(&srcB[nt][k]);
+      }
     }
     #pragma unroll
     for (int mt = 0; mt < 2; mt++) {
       #pragma unroll
       for (int nt = 0; nt < NT; nt++) {
         acc[mt][nt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            a[mt], b[nt], acc[mt][nt], 0, 0, 0);
+            a[cur][mt], b[cur][nt], acc[mt][nt], 0, 0, 0);
       }
     }
   }

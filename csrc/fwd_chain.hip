@@ -74,31 +74,61 @@ __device__ void fc_layer(const short* __restrict__ src_lds,
                          short* __restrict__ dst_lds, int32_t wave,
                          int32_t lane) {
   constexpr int NT = N / 128;  // n-tiles of 32 per wave (4 waves)
+  constexpr int ITERS = K / 16;
   const int32_t n_base = wave * (N / 4);
   const int32_t frag_k0 = (lane >> 5) * 8;
   const int32_t ml = lane & 31;  // A row within m-tile / D col (n)
 
+  // Software-pipelined k-loop: LDS occupancy pins this kernel at 1
+  // wave/SIMD, so cross-wave latency hiding does not exist — the next
+  // iteration's A (LDS) and B (global/L2) fragments must be IN FLIGHT
+  // while the current MFMAs run. Fully unrolled (compile-time ITERS)
+  // double-buffered prefetch: buf indices become constants after unroll,
+  // so the fragment arrays stay in registers.
+  const short* srcA[2] = {&src_lds[ml * SRC_S + frag_k0],
+                          &src_lds[(32 + ml) * SRC_S + frag_k0]};
+  const short* srcB[NT];
+  #pragma unroll
+  for (int nt = 0; nt < NT; nt++) {
+    srcB[nt] = &W[(int64_t)(n_base + nt * 32 + ml) * K + frag_k0];
+  }
+
   fc_f32x16 acc[2][NT] = {};
-  for (int32_t k = 0; k < K; k += 16) {
-    fc_bf16x8 a[2], b[NT];
-    #pragma unroll
-    for (int mt = 0; mt < 2; mt++) {
-      *reinterpret_cast<uint4*>(&a[mt]) =
-          *reinterpret_cast<const uint4*>(
-              &src_lds[(mt * 32 + ml) * SRC_S + k + frag_k0]);
-    }
-    #pragma unroll
-    for (int nt = 0; nt < NT; nt++) {
-      const int32_t n = n_base + nt * 32 + ml;
-      *reinterpret_cast<uint4*>(&b[nt]) =
-          *reinterpret_cast<const uint4*>(&W[n * K + k + frag_k0]);
+  fc_bf16x8 a[2][2], b[2][NT];
+
+  #pragma unroll
+  for (int mt = 0; mt < 2; mt++) {
+    *reinterpret_cast<uint4*>(&a[0][mt]) =
+        *reinterpret_cast<const uint4*>(srcA[mt]);
+  }
+  #pragma unroll
+  for (int nt = 0; nt < NT; nt++) {
+    *reinterpret_cast<uint4*>(&b[0][nt]) =
+        *reinterpret_cast<const uint4*>(srcB[nt]);
+  }
+  #pragma unroll
+  for (int i = 0; i < ITERS; i++) {
+    const int cur = i & 1;
+    const int nxt = cur ^ 1;
+    if (i + 1 < ITERS) {
+      const int32_t k = (i + 1) * 16;
+      #pragma unroll
+      for (int mt = 0; mt < 2; mt++) {
+        *reinterpret_cast<uint4*>(&a[nxt][mt]) =
+            *reinterpret_cast<const uint4*>(&srcA[mt][k]);
+      }
+      #pragma unroll
+      for (int nt = 0; nt < NT; nt++) {
+        *reinterpret_cast<uint4*>(&b[nxt][nt]) =
+            *reinterpret_cast<const uint4*>(&srcB[nt][k]);
+      }
     }
     #pragma unroll
     for (int mt = 0; mt < 2; mt++) {
       #pragma unroll
       for (int nt = 0; nt < NT; nt++) {
         acc[mt][nt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            a[mt], b[nt], acc[mt][nt], 0, 0, 0);
+            a[cur][mt], b[cur][nt], acc[mt][nt], 0, 0, 0);
       }
     }
   }

@@ -465,7 +465,10 @@ std::vector<at::Tensor> fwd_chain_bf16(
                 "fwd_chain: ", nm, " must be contiguous bf16 [", n, ",", k,
                 "]");
   };
-  chk_w(W1, 512, 100, "W1");
+  // W1 may arrive pre-padded [512,112] (cached by the fused-step driver)
+  // or raw [512,100] (padded here, one kernel per call).
+  const bool w1_padded = W1.size(1) == 112;
+  chk_w(W1, 512, w1_padded ? 112 : 100, "W1");
   chk_w(W2, 256, 512, "W2");
   chk_w(W3, 128, 256, "W3");
   TORCH_CHECK(w4.numel() == 128 && w4.scalar_type() == at::kBFloat16 &&
@@ -479,7 +482,7 @@ std::vector<at::Tensor> fwd_chain_bf16(
        b4f = fb(b4, 1);
   const int64_t M = x.size(0);
   // The kernel's A-fragment loads need W1's k-dim padded 100 -> 112.
-  auto W1p = at::constant_pad_nd(W1, {0, 12}).contiguous();
+  auto W1p = w1_padded ? W1 : at::constant_pad_nd(W1, {0, 12}).contiguous();
   // ...and x0's rows padded likewise, staged in LDS by the kernel itself.
   auto a1 = at::empty({M, 512}, x.options());
   auto a2 = at::empty({M, 256}, x.options());
@@ -519,21 +522,31 @@ std::vector<at::Tensor> bwd_chain_bf16(
   chk_a(a3, 128, "a3");
   TORCH_CHECK(w4.numel() == 128 && w4.scalar_type() == at::kBFloat16,
               "bwd_chain: w4 must be bf16 [128]");
-  TORCH_CHECK(W3.size(0) == 128 && W3.size(1) == 256 &&
-                  W3.scalar_type() == at::kBFloat16,
-              "bwd_chain: W3 must be bf16 [128,256]");
-  TORCH_CHECK(W2.size(0) == 256 && W2.size(1) == 512 &&
-                  W2.scalar_type() == at::kBFloat16,
-              "bwd_chain: W2 must be bf16 [256,512]");
+  // Weights may arrive pre-transposed ([256,128] / [512,256], cached by
+  // the fused-step driver) or in model layout (transposed here per call).
+  // The shapes are unambiguous between the two layouts.
+  TORCH_CHECK(W3.scalar_type() == at::kBFloat16 &&
+                  ((W3.size(0) == 128 && W3.size(1) == 256) ||
+                   (W3.size(0) == 256 && W3.size(1) == 128)),
+              "bwd_chain: W3 must be bf16 [128,256] or W3^T [256,128]");
+  TORCH_CHECK(W2.scalar_type() == at::kBFloat16 &&
+                  ((W2.size(0) == 256 && W2.size(1) == 512) ||
+                   (W2.size(0) == 512 && W2.size(1) == 256)),
+              "bwd_chain: W2 must be bf16 [256,512] or W2^T [512,256]");
   auto w4c = w4.contiguous();
-  auto W3T = W3.t().contiguous();  // [256,128]
-  auto W2T = W2.t().contiguous();  // [512,256]
+  auto W3T = (W3.size(0) == 256) ? W3.contiguous()
+                                 : W3.t().contiguous();  // [256,128]
+  auto W2T = (W2.size(0) == 512) ? W2.contiguous()
+                                 : W2.t().contiguous();  // [512,256]
   auto dz1 = at::empty({M, 512}, dy.options());
   auto dz2 = at::empty({M, 256}, dy.options());
   auto dz3 = at::empty({M, 128}, dy.options());
   const int64_t grid = (M + 63) / 64;
-  auto db_part = at::zeros({std::max<int64_t>(grid, 1), 512 + 256 + 128 + 1},
+  // Every block writes its whole db_part row (bc_bias_partial covers all
+  // columns incl. db4), so empty() is safe — no memset kernel per call.
+  auto db_part = at::empty({std::max<int64_t>(grid, 1), 512 + 256 + 128 + 1},
                            dy.options().dtype(at::kFloat));
+  if (M == 0) db_part.zero_();
   if (M > 0) {
     launch_bwd_chain(dy.data_ptr(), a1.data_ptr(), a2.data_ptr(),
                      a3.data_ptr(), w4c.data_ptr(), W3T.data_ptr(),

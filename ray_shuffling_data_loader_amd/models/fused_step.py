@@ -45,6 +45,33 @@ def _wgrad_bmm(dz: torch.Tensor, src: torch.Tensor) -> torch.Tensor:
     return dz.t() @ src
 
 
+def _weight_buffers(model: TabularMLP, lin):
+    """Persistent bf16 weight staging for the chain kernels: W1 pre-padded
+    to [512,112], W2/W3/w4 in model layout, W2T/W3T pre-transposed for the
+    dgrad B fragments. Allocated once; refreshed per step with plain
+    cast-copies (the params change every optimizer step)."""
+    buf = getattr(model, "_fused_buf", None)
+    dev = lin[0].weight.device
+    if buf is None or buf["W1p"].device != dev:
+        bf = dict(dtype=torch.bfloat16, device=dev)
+        buf = {
+            "W1p": torch.zeros(512, 112, **bf),
+            "W2": torch.empty(256, 512, **bf),
+            "W3": torch.empty(128, 256, **bf),
+            "w4": torch.empty(128, **bf),
+            "W2T": torch.empty(512, 256, **bf),
+            "W3T": torch.empty(256, 128, **bf),
+        }
+        model._fused_buf = buf
+    buf["W1p"][:, :100].copy_(lin[0].weight.detach())
+    buf["W2"].copy_(lin[1].weight.detach())
+    buf["W3"].copy_(lin[2].weight.detach())
+    buf["w4"].copy_(lin[3].weight.detach().view(-1))
+    buf["W2T"].copy_(buf["W2"].t())
+    buf["W3T"].copy_(buf["W3"].t())
+    return buf
+
+
 def fused_step(
     model: TabularMLP, x: torch.Tensor, target: torch.Tensor
 ) -> Tuple[torch.Tensor, None]:
@@ -57,23 +84,23 @@ def fused_step(
     hip = _load_hip()
     lin = _layers(model)
     M = x.shape[0]
-    wb = [(m.weight.detach().bfloat16(), m.bias.detach()) for m in lin]
-    (W1, b1), (W2, b2), (W3, b3), (W4, b4) = wb
+    buf = _weight_buffers(model, lin)
+    b1, b2, b3, b4 = (m.bias.detach() for m in lin)
     a1, a2, a3, out = hip.fwd_chain_bf16(
-        x, W1, b1, W2, b2, W3, b3, W4.flatten(), b4
+        x, buf["W1p"], b1, buf["W2"], b2, buf["W3"], b3, buf["w4"], b4
     )
     diff = out.float() - target.float().reshape(-1, 1)
     loss = diff.square().mean()
     dy = (2.0 / (M * 1.0)) * diff  # d(mean((out-t)^2))/d out
     dyb = dy.bfloat16().contiguous()
     dz1, dz2, dz3, db1, db2, db3, db4 = hip.bwd_chain_bf16(
-        dyb, a1, a2, a3, W4.flatten(), W3, W2
+        dyb, a1, a2, a3, buf["w4"], buf["W3T"], buf["W2T"]
     )
     grads = [
-        (_wgrad_bmm(dz1, x).float(), db1),
-        (_wgrad_bmm(dz2, a1).float(), db2),
-        (_wgrad_bmm(dz3, a2).float(), db3),
-        ((dyb.t().float() @ a3.float()), db4),
+        (_wgrad_bmm(dz1, x), db1),
+        (_wgrad_bmm(dz2, a1), db2),
+        (_wgrad_bmm(dz3, a2), db3),
+        (_wgrad_bmm(dyb, a3), db4),
     ]
     for m, (gw, gb) in zip(lin, grads):
         m.weight.grad = gw.to(m.weight.dtype)

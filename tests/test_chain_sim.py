@@ -243,6 +243,10 @@ def test_fused_step_glue_cpu(monkeypatch):
     class FakeHip:
         @staticmethod
         def fwd_chain_bf16(x, W1, b1, W2, b2, W3, b3, w4, b4):
+            # The binding accepts W1 pre-padded [512,112] (zeros in cols
+            # 100..111) or raw [512,100] — mirror that dispatch.
+            if W1.shape[1] == 112:
+                W1 = W1[:, :100]
             a1 = torch.relu(x.float() @ W1.float().t() + b1.float())
             a2 = torch.relu(a1 @ W2.float().t() + b2.float())
             a3 = torch.relu(a2 @ W3.float().t() + b3.float())
@@ -252,6 +256,12 @@ def test_fused_step_glue_cpu(monkeypatch):
 
         @staticmethod
         def bwd_chain_bf16(dy, a1, a2, a3, w4, W3, W2):
+            # Accept model-layout or pre-transposed weights (binding
+            # dispatches on shape: W3T [256,128], W2T [512,256]).
+            if W3.shape[0] == 256:
+                W3 = W3.t()
+            if W2.shape[0] == 512:
+                W2 = W2.t()
             da3 = dy.float() @ w4.float().unsqueeze(0)
             dz3 = (da3 * (a3.float() > 0)).bfloat16()
             da2 = dz3.float() @ W3.float()
