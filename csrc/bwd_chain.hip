@@ -35,7 +35,8 @@ namespace rsdl {
 typedef __attribute__((__vector_size__(8 * sizeof(short)))) short bc_bf16x8;
 typedef __attribute__((__vector_size__(16 * sizeof(float)))) float bc_f32x16;
 
-#define BC_MT 64
+#define BC_MT 32
+#define BC_MTILES (BC_MT / 32)
 #define BC_N1 512
 #define BC_N2 256
 #define BC_N3 128
@@ -115,19 +116,22 @@ __device__ void bc_layer(const short* __restrict__ dz_src,
 
   // Software-pipelined k-loop (same rationale as fc_layer: 1 wave/SIMD —
   // the next fragments must be in flight during the current MFMAs).
-  const short* srcA[2] = {&dz_src[ml * SRC_S + frag_k0],
-                          &dz_src[(32 + ml) * SRC_S + frag_k0]};
+  const short* srcA[BC_MTILES];
+  #pragma unroll
+  for (int mt = 0; mt < BC_MTILES; mt++) {
+    srcA[mt] = &dz_src[(mt * 32 + ml) * SRC_S + frag_k0];
+  }
   const short* srcB[NT];
   #pragma unroll
   for (int nt = 0; nt < NT; nt++) {
     srcB[nt] = &WT[(int64_t)(n_base + nt * 32 + ml) * K + frag_k0];
   }
 
-  bc_f32x16 acc[2][NT] = {};
-  bc_bf16x8 a[2][2], b[2][NT];
+  bc_f32x16 acc[BC_MTILES][NT] = {};
+  bc_bf16x8 a[2][BC_MTILES], b[2][NT];
 
   #pragma unroll
-  for (int mt = 0; mt < 2; mt++) {
+  for (int mt = 0; mt < BC_MTILES; mt++) {
     *reinterpret_cast<uint4*>(&a[0][mt]) =
         *reinterpret_cast<const uint4*>(srcA[mt]);
   }
@@ -143,7 +147,7 @@ __device__ void bc_layer(const short* __restrict__ dz_src,
     if (i + 1 < ITERS) {
       const int32_t k = (i + 1) * 16;
       #pragma unroll
-      for (int mt = 0; mt < 2; mt++) {
+      for (int mt = 0; mt < BC_MTILES; mt++) {
         *reinterpret_cast<uint4*>(&a[nxt][mt]) =
             *reinterpret_cast<const uint4*>(&srcA[mt][k]);
       }
@@ -154,7 +158,7 @@ __device__ void bc_layer(const short* __restrict__ dz_src,
       }
     }
     #pragma unroll
-    for (int mt = 0; mt < 2; mt++) {
+    for (int mt = 0; mt < BC_MTILES; mt++) {
       #pragma unroll
       for (int nt = 0; nt < NT; nt++) {
         acc[mt][nt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
@@ -167,7 +171,7 @@ __device__ void bc_layer(const short* __restrict__ dz_src,
   // n-range of a_dst; inter-layer ordering is handled by the
   // __syncthreads() between bc_layer calls in the kernel body.
   #pragma unroll
-  for (int mt = 0; mt < 2; mt++) {
+  for (int mt = 0; mt < BC_MTILES; mt++) {
     #pragma unroll
     for (int nt = 0; nt < NT; nt++) {
       const int32_t n = n_base + nt * 32 + ml;
@@ -244,6 +248,8 @@ __global__ void __launch_bounds__(256) bwd_chain_kernel(
   bc_store_tile<BC_N2, BC_S2>(t2, dz2, m0, M, tid);
   bc_store_tile<BC_N3, BC_S3>(t3, dz3, m0, M, tid);
 }
+
+int64_t bwd_chain_grid(int64_t M) { return (M + BC_MT - 1) / BC_MT; }
 
 void launch_bwd_chain(const void* dy, const void* a1, const void* a2,
                       const void* a3, const void* w4, const void* W3T,

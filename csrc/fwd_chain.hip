@@ -6,14 +6,19 @@
 //   a3 = relu(a2 @ W3^T + b3)   -> [M,128]
 //   out = a3 @ w4^T + b4        -> [M,1]
 //
-// One 256-thread workgroup carries a 64-row slab through the whole chain:
-// the slab's activations live in LDS between layers (64x512 + 64x256 +
-// 64x128 bf16 tiles + padding ~= 125 KB of the 160 KB LDS), weights are
+// One 256-thread workgroup carries an FC_MT-row slab through the whole
+// chain: the slab's activations live in LDS between layers, weights are
 // read from global (L2-resident, ~0.44 MB total) per MFMA step, and only
 // x0 (read) and the a1/a2/a3/out tiles (written once for backward) touch
 // HBM. Eliminates the inter-layer activation re-reads of the eager path
 // (~448 MB/step at the bench shape; measured eager fwd 0.25 ms vs a
 // ~0.1 ms fused roofline).
+//
+// FC_MT = 32 (round-2 tuning): the 64-row variant used ~133 KB LDS ->
+// 1 workgroup/CU = 1 wave/SIMD, leaving the L2 weight-fragment latency
+// of the k-loops unhidden even with software pipelining (measured
+// 512 us vs the ~100 us roofline). 32-row tiles take ~67 KB -> 2
+// workgroups/CU, so one workgroup's MFMAs cover the other's loads.
 //
 // MFMA orientation (v_mfma_f32_32x32x16_bf16, probe-verified maps in
 // tools/mfma_probe.hip / csrc/wgrad_kernel.hip):
@@ -26,8 +31,7 @@
 // is D -> next layer's LDS tile (b16 column writes, padded stride).
 //
 // Not wired into any default path: built and bound, exercised only by the
-// RSDL_EXPERIMENTAL=1 GPU test. Known-untuned: weight tiles could be
-// LDS-cached for L2/L3, and the k-loops are not software-pipelined yet.
+// RSDL_EXPERIMENTAL=1 GPU test.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
@@ -39,7 +43,8 @@ namespace rsdl {
 typedef __attribute__((__vector_size__(8 * sizeof(short)))) short fc_bf16x8;
 typedef __attribute__((__vector_size__(16 * sizeof(float)))) float fc_f32x16;
 
-#define FC_MT 64            // rows per workgroup slab
+#define FC_MT 32            // rows per workgroup slab
+#define FC_MTILES (FC_MT / 32)  // 32-row MFMA m-tiles per slab
 #define FC_K0 100           // input feature count
 #define FC_K0P 112          // padded to a 16-multiple for the k-loop
 #define FC_N1 512
@@ -85,19 +90,22 @@ __device__ void fc_layer(const short* __restrict__ src_lds,
   // while the current MFMAs run. Fully unrolled (compile-time ITERS)
   // double-buffered prefetch: buf indices become constants after unroll,
   // so the fragment arrays stay in registers.
-  const short* srcA[2] = {&src_lds[ml * SRC_S + frag_k0],
-                          &src_lds[(32 + ml) * SRC_S + frag_k0]};
+  const short* srcA[FC_MTILES];
+  #pragma unroll
+  for (int mt = 0; mt < FC_MTILES; mt++) {
+    srcA[mt] = &src_lds[(mt * 32 + ml) * SRC_S + frag_k0];
+  }
   const short* srcB[NT];
   #pragma unroll
   for (int nt = 0; nt < NT; nt++) {
     srcB[nt] = &W[(int64_t)(n_base + nt * 32 + ml) * K + frag_k0];
   }
 
-  fc_f32x16 acc[2][NT] = {};
-  fc_bf16x8 a[2][2], b[2][NT];
+  fc_f32x16 acc[FC_MTILES][NT] = {};
+  fc_bf16x8 a[2][FC_MTILES], b[2][NT];
 
   #pragma unroll
-  for (int mt = 0; mt < 2; mt++) {
+  for (int mt = 0; mt < FC_MTILES; mt++) {
     *reinterpret_cast<uint4*>(&a[0][mt]) =
         *reinterpret_cast<const uint4*>(srcA[mt]);
   }
@@ -113,7 +121,7 @@ __device__ void fc_layer(const short* __restrict__ src_lds,
     if (i + 1 < ITERS) {
       const int32_t k = (i + 1) * 16;
       #pragma unroll
-      for (int mt = 0; mt < 2; mt++) {
+      for (int mt = 0; mt < FC_MTILES; mt++) {
         *reinterpret_cast<uint4*>(&a[nxt][mt]) =
             *reinterpret_cast<const uint4*>(&srcA[mt][k]);
       }
@@ -124,7 +132,7 @@ __device__ void fc_layer(const short* __restrict__ src_lds,
       }
     }
     #pragma unroll
-    for (int mt = 0; mt < 2; mt++) {
+    for (int mt = 0; mt < FC_MTILES; mt++) {
       #pragma unroll
       for (int nt = 0; nt < NT; nt++) {
         acc[mt][nt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
@@ -135,7 +143,7 @@ __device__ void fc_layer(const short* __restrict__ src_lds,
   // Epilogue: bias + (relu) + cast + store the D fragments into the dst
   // tile at [mrow][n] (b16 column writes; DST_S padding spreads banks).
   #pragma unroll
-  for (int mt = 0; mt < 2; mt++) {
+  for (int mt = 0; mt < FC_MTILES; mt++) {
     #pragma unroll
     for (int nt = 0; nt < NT; nt++) {
       const int32_t n = n_base + nt * 32 + ml;
@@ -204,20 +212,24 @@ __global__ void __launch_bounds__(256) fwd_chain_kernel(
   fc_layer<FC_N2, FC_N3, FC_S2, FC_S3, true>(t2, W3, b3, t3, wave, lane);
   __syncthreads();
 
-  // Head: out[m] = sum_k a3[m,k] * w4[k] + b4. 4 threads per row, 32 k
-  // each, pair-wise LDS-free reduce via wave shuffles (partners are
-  // adjacent lanes).
+  // Head: out[m] = sum_k a3[m,k] * w4[k] + b4. 256/FC_MT threads per row,
+  // pair-wise LDS-free reduce via wave shuffles (partners are adjacent
+  // lanes).
   {
-    const int32_t m = tid >> 2;
-    const int32_t part = tid & 3;
+    constexpr int TPR = 256 / FC_MT;       // threads per row
+    constexpr int KPT = FC_N3 / TPR;       // k per thread
+    const int32_t m = tid / TPR;
+    const int32_t part = tid % TPR;
     float s = 0.f;
     #pragma unroll
-    for (int32_t kk = 0; kk < 32; kk++) {
-      const int32_t k = part * 32 + kk;
+    for (int32_t kk = 0; kk < KPT; kk++) {
+      const int32_t k = part * KPT + kk;
       s += fc_b2f(t3[m * FC_S3 + k]) * fc_b2f(w4[k]);
     }
-    s += __shfl_down(s, 1);
-    s += __shfl_down(s, 2);
+    #pragma unroll
+    for (int32_t d = 1; d < TPR; d <<= 1) {
+      s += __shfl_down(s, d);
+    }
     if (part == 0 && m0 + m < M) {
       out[m0 + m] = fc_f2b(s + b4[0]);
     }

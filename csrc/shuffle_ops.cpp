@@ -73,6 +73,7 @@ void launch_fwd_chain(const void* x0, const void* W1, const float* b1,
                       const float* b3, const void* w4, const float* b4,
                       void* a1, void* a2, void* a3, void* out, int64_t M,
                       hipStream_t stream);
+int64_t bwd_chain_grid(int64_t M);
 void launch_bwd_chain(const void* dy, const void* a1, const void* a2,
                       const void* a3, const void* w4, const void* W3T,
                       const void* W2T, void* dz1, void* dz2, void* dz3,
@@ -541,7 +542,7 @@ std::vector<at::Tensor> bwd_chain_bf16(
   auto dz1 = at::empty({M, 512}, dy.options());
   auto dz2 = at::empty({M, 256}, dy.options());
   auto dz3 = at::empty({M, 128}, dy.options());
-  const int64_t grid = (M + 63) / 64;
+  const int64_t grid = bwd_chain_grid(M);
   // Every block writes its whole db_part row (bc_bias_partial covers all
   // columns incl. db4), so empty() is safe — no memset kernel per call.
   auto db_part = at::empty({std::max<int64_t>(grid, 1), 512 + 256 + 128 + 1},

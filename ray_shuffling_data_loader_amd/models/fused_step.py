@@ -100,7 +100,9 @@ def fused_step(
         (_wgrad_bmm(dz1, x), db1),
         (_wgrad_bmm(dz2, a1), db2),
         (_wgrad_bmm(dz3, a2), db3),
-        (_wgrad_bmm(dyb, a3), db4),
+        # dW4 [1,128]: plain bf16 GEMM — the chunked bmm at batch=50/N=1
+        # hits a pathological ~10 ms host path in hipBLASLt dispatch.
+        (dyb.t() @ a3, db4),
     ]
     for m, (gw, gb) in zip(lin, grads):
         m.weight.grad = gw.to(m.weight.dtype)
