@@ -34,6 +34,7 @@ namespace rsdl {
 
 typedef __attribute__((__vector_size__(8 * sizeof(short)))) short bc_bf16x8;
 typedef __attribute__((__vector_size__(16 * sizeof(float)))) float bc_f32x16;
+typedef __attribute__((__vector_size__(4 * sizeof(unsigned int)))) unsigned int bc_u32x4;
 
 #define BC_MT 32
 #define BC_MTILES (BC_MT / 32)
@@ -65,11 +66,14 @@ __device__ void bc_load_tile(const short* __restrict__ g, short* lds,
   for (int32_t u = tid; u < BC_MT * VPR; u += 256) {
     const int32_t m = u / VPR;
     const int32_t c = (u % VPR) * 8;
-    uint4 v = {0, 0, 0, 0};
+    bc_u32x4 v = {0, 0, 0, 0};
     if (m0 + m < M) {
-      v = *reinterpret_cast<const uint4*>(&g[(m0 + m) * N + c]);
+      // Non-temporal: the activation stream (449 MB/step) must not evict
+      // the L2-resident transposed weights every workgroup re-reads.
+      v = __builtin_nontemporal_load(
+          reinterpret_cast<const bc_u32x4*>(&g[(m0 + m) * N + c]));
     }
-    *reinterpret_cast<uint4*>(&lds[m * S + c]) = v;
+    *reinterpret_cast<bc_u32x4*>(&lds[m * S + c]) = v;
   }
 }
 
@@ -81,8 +85,9 @@ __device__ void bc_store_tile(const short* __restrict__ lds, short* out,
     const int32_t m = u / VPR;
     const int32_t c = (u % VPR) * 8;
     if (m0 + m < M) {
-      *reinterpret_cast<uint4*>(&out[(m0 + m) * N + c]) =
-          *reinterpret_cast<const uint4*>(&lds[m * S + c]);
+      __builtin_nontemporal_store(
+          *reinterpret_cast<const bc_u32x4*>(&lds[m * S + c]),
+          reinterpret_cast<bc_u32x4*>(&out[(m0 + m) * N + c]));
     }
   }
 }

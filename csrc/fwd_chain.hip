@@ -42,6 +42,7 @@ namespace rsdl {
 
 typedef __attribute__((__vector_size__(8 * sizeof(short)))) short fc_bf16x8;
 typedef __attribute__((__vector_size__(16 * sizeof(float)))) float fc_f32x16;
+typedef __attribute__((__vector_size__(4 * sizeof(unsigned int)))) unsigned int fc_u32x4;
 
 #define FC_MT 32            // rows per workgroup slab
 #define FC_MTILES (FC_MT / 32)  // 32-row MFMA m-tiles per slab
@@ -170,8 +171,12 @@ __device__ void fc_store_tile(const short* __restrict__ lds, short* out,
     const int32_t m = u / VPR;
     const int32_t c = (u % VPR) * 8;
     if (m0 + m < M) {
-      *reinterpret_cast<uint4*>(&out[(m0 + m) * N + c]) =
-          *reinterpret_cast<const uint4*>(&lds[m * S + c]);
+      // Non-temporal: activations are written once and re-read only by the
+      // backward kernel much later — keeping them OUT of L2 preserves the
+      // weight working set (which every workgroup re-reads).
+      __builtin_nontemporal_store(
+          *reinterpret_cast<const fc_u32x4*>(&lds[m * S + c]),
+          reinterpret_cast<fc_u32x4*>(&out[(m0 + m) * N + c]));
     }
   }
 }
@@ -200,7 +205,9 @@ __global__ void __launch_bounds__(256) fwd_chain_kernel(
     const int32_t m = u / FC_K0P;
     const int32_t c = u % FC_K0P;
     short v = 0;
-    if (m0 + m < M && c < FC_K0) v = x0[(m0 + m) * FC_K0 + c];
+    if (m0 + m < M && c < FC_K0) {
+      v = __builtin_nontemporal_load(&x0[(m0 + m) * FC_K0 + c]);
+    }
     t0[m * FC_S0 + c] = v;
   }
   __syncthreads();

@@ -72,6 +72,13 @@ def _weight_buffers(model: TabularMLP, lin):
     return buf
 
 
+def _head_wgrad(dyb: torch.Tensor, a3: torch.Tensor) -> torch.Tensor:
+    from ray_shuffling_data_loader_amd.ops.shuffle_ops import wgrad
+
+    dw, _ = wgrad(dyb, a3, with_bias=False)
+    return dw  # [1,128] fp32
+
+
 def fused_step(
     model: TabularMLP, x: torch.Tensor, target: torch.Tensor
 ) -> Tuple[torch.Tensor, None]:
@@ -100,9 +107,11 @@ def fused_step(
         (_wgrad_bmm(dz1, x), db1),
         (_wgrad_bmm(dz2, a1), db2),
         (_wgrad_bmm(dz3, a2), db3),
-        # dW4 [1,128]: plain bf16 GEMM — the chunked bmm at batch=50/N=1
-        # hits a pathological ~10 ms host path in hipBLASLt dispatch.
-        (dyb.t() @ a3, db4),
+        # dW4 [1,128]: hand-written MFMA wgrad — beats the library on the
+        # degenerate head shape (profiles/PERF.md wgrad zoo); the chunked
+        # bmm at batch=50/N=1 hit a ~10 ms host stall in hipBLASLt
+        # dispatch, and the plain bf16 GEMM picked a 170 us kernel.
+        (_head_wgrad(dyb, a3), db4),
     ]
     for m, (gw, gb) in zip(lin, grads):
         m.weight.grad = gw.to(m.weight.dtype)
