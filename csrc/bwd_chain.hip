@@ -126,10 +126,12 @@ __device__ void bc_layer(const short* __restrict__ dz_src,
   for (int mt = 0; mt < BC_MTILES; mt++) {
     srcA[mt] = &dz_src[(mt * 32 + ml) * SRC_S + frag_k0];
   }
+  // Fragment-major swizzled weights (see fwd_chain.hip): block
+  // (ntile, kc) = 512 contiguous halfwords, lane slice at lane*8.
   const short* srcB[NT];
   #pragma unroll
   for (int nt = 0; nt < NT; nt++) {
-    srcB[nt] = &WT[(int64_t)(n_base + nt * 32 + ml) * K + frag_k0];
+    srcB[nt] = &WT[((int64_t)(wave * NT + nt) * ITERS) * 512 + lane * 8];
   }
 
   bc_f32x16 acc[BC_MTILES][NT] = {};
@@ -159,7 +161,7 @@ __device__ void bc_layer(const short* __restrict__ dz_src,
       #pragma unroll
       for (int nt = 0; nt < NT; nt++) {
         *reinterpret_cast<uint4*>(&b[nxt][nt]) =
-            *reinterpret_cast<const uint4*>(&srcB[nt][k]);
+            *reinterpret_cast<const uint4*>(&srcB[nt][(i + 1) * 512]);
       }
     }
     #pragma unroll

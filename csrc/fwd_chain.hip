@@ -26,9 +26,17 @@
 //     A[m = lane&31][k = (lane>>5)*8 + j]   j = 0..7 of a bf16x8
 //     B[k = (lane>>5)*8 + j][n = lane&31]
 //     D col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
-// For y[m,n] = sum_k x[m,k] W[n,k] both fragments are CONTIGUOUS row
-// segments (x row m, W row n) — no transpose staging; the only relayout
-// is D -> next layer's LDS tile (b16 column writes, padded stride).
+// For y[m,n] = sum_k x[m,k] W[n,k] the A fragment is a contiguous row
+// segment of the LDS tile. The B (weight) fragment is served from a
+// FRAGMENT-MAJOR swizzled weight layout
+//   SW[n_tile][k_chunk][h][ml][8]  (n_tile = n/32, k_chunk = k/16,
+//   h = lane>>5, ml = lane&31)
+// so one wave's 64 16-B fragment loads are a single contiguous 1 KB
+// block. In the natural [N][K] layout each B load touched 32 cache
+// lines 1 KB apart (rows), which bound the kernel on L1/TCP line
+// processing (~16 GB of line traffic per launch) — the swizzle makes
+// weight traffic coalesced and line-minimal. The host binding performs
+// the swizzle per call (weights are 0.44 MB total; a few us).
 //
 // Not wired into any default path: built and bound, exercised only by the
 // RSDL_EXPERIMENTAL=1 GPU test.
@@ -96,10 +104,12 @@ __device__ void fc_layer(const short* __restrict__ src_lds,
   for (int mt = 0; mt < FC_MTILES; mt++) {
     srcA[mt] = &src_lds[(mt * 32 + ml) * SRC_S + frag_k0];
   }
+  // Swizzled weight base for this wave's n-tiles: block (ntile, kc) is
+  // 512 contiguous halfwords; lane's 16-B slice at lane*8.
   const short* srcB[NT];
   #pragma unroll
   for (int nt = 0; nt < NT; nt++) {
-    srcB[nt] = &W[(int64_t)(n_base + nt * 32 + ml) * K + frag_k0];
+    srcB[nt] = &W[((int64_t)(wave * NT + nt) * ITERS) * 512 + lane * 8];
   }
 
   fc_f32x16 acc[FC_MTILES][NT] = {};
@@ -129,7 +139,7 @@ __device__ void fc_layer(const short* __restrict__ src_lds,
       #pragma unroll
       for (int nt = 0; nt < NT; nt++) {
         *reinterpret_cast<uint4*>(&b[nxt][nt]) =
-            *reinterpret_cast<const uint4*>(&srcB[nt][k]);
+            *reinterpret_cast<const uint4*>(&srcB[nt][(i + 1) * 512]);
       }
     }
     #pragma unroll

@@ -451,6 +451,18 @@ std::vector<at::Tensor> relu_bwd_bias(const at::Tensor& dy,
 // EXPERIMENTAL round-2 fused forward chain (csrc/fwd_chain.hip): fixed
 // TabularMLP(100-512-256-128-1) architecture, bf16. Returns
 // (a1, a2, a3, out). Exercised only by the RSDL_EXPERIMENTAL=1 GPU test.
+// Fragment-major weight swizzle for the chain kernels (see
+// csrc/fwd_chain.hip): [N,K] -> [N/32][K/16][2][32][8] flat, so one
+// wave's B-fragment load is a contiguous 1 KB block.
+static at::Tensor swizzle_frag(const at::Tensor& W) {
+  int64_t N = W.size(0), K = W.size(1);
+  TORCH_CHECK(N % 32 == 0 && K % 16 == 0, "swizzle_frag: bad shape");
+  return W.view({N / 32, 32, K / 16, 2, 8})
+      .permute({0, 2, 3, 1, 4})
+      .contiguous()
+      .view({-1});
+}
+
 std::vector<at::Tensor> fwd_chain_bf16(
     const at::Tensor& x, const at::Tensor& W1, const at::Tensor& b1,
     const at::Tensor& W2, const at::Tensor& b2, const at::Tensor& W3,
@@ -490,8 +502,11 @@ std::vector<at::Tensor> fwd_chain_bf16(
   auto a3 = at::empty({M, 128}, x.options());
   auto out = at::empty({M, 1}, x.options());
   if (M > 0) {
-    launch_fwd_chain(x.data_ptr(), W1p.data_ptr(), b1f.data_ptr<float>(),
-                     W2.data_ptr(), b2f.data_ptr<float>(), W3.data_ptr(),
+    auto W1s = swizzle_frag(W1p);
+    auto W2s = swizzle_frag(W2);
+    auto W3s = swizzle_frag(W3);
+    launch_fwd_chain(x.data_ptr(), W1s.data_ptr(), b1f.data_ptr<float>(),
+                     W2s.data_ptr(), b2f.data_ptr<float>(), W3s.data_ptr(),
                      b3f.data_ptr<float>(), w4.data_ptr(),
                      b4f.data_ptr<float>(), a1.data_ptr(), a2.data_ptr(),
                      a3.data_ptr(), out.data_ptr(), M, current_stream());
@@ -549,9 +564,11 @@ std::vector<at::Tensor> bwd_chain_bf16(
                            dy.options().dtype(at::kFloat));
   if (M == 0) db_part.zero_();
   if (M > 0) {
+    auto W3Ts = swizzle_frag(W3T);
+    auto W2Ts = swizzle_frag(W2T);
     launch_bwd_chain(dy.data_ptr(), a1.data_ptr(), a2.data_ptr(),
-                     a3.data_ptr(), w4c.data_ptr(), W3T.data_ptr(),
-                     W2T.data_ptr(), dz1.data_ptr(), dz2.data_ptr(),
+                     a3.data_ptr(), w4c.data_ptr(), W3Ts.data_ptr(),
+                     W2Ts.data_ptr(), dz1.data_ptr(), dz2.data_ptr(),
                      dz3.data_ptr(), db_part.data_ptr<float>(), M,
                      current_stream());
   }
