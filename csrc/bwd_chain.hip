@@ -204,7 +204,11 @@ __global__ void __launch_bounds__(256) bwd_chain_kernel(
     const short* __restrict__ W2T,   // [512,256]  (W2^T, contiguous)
     short* __restrict__ dz1, short* __restrict__ dz2,
     short* __restrict__ dz3,
-    float* __restrict__ db_part,     // [grid][512+256+128+1]
+    // [grid][512+256+128+1+256]: db1|db2|db3|db4|dW4 partials. dW4[k] =
+    // sum_m dy[m]*a3[m,k] folded into the dz3 seed loop (a3 is already
+    // in LDS there) — replaces a separate 54 us head-wgrad kernel; the
+    // two per-k thread partials (even/odd m) are summed host-side.
+    float* __restrict__ db_part,
     int64_t M) {
   __shared__ __align__(16) short t1[BC_MT * BC_S1];
   __shared__ __align__(16) short t2[BC_MT * BC_S2];
@@ -215,7 +219,8 @@ __global__ void __launch_bounds__(256) bwd_chain_kernel(
   const int32_t tid = threadIdx.x;
   const int32_t wave = tid >> 6;
   const int32_t lane = tid & 63;
-  float* part = &db_part[(int64_t)blockIdx.x * (BC_N1 + BC_N2 + BC_N3 + 1)];
+  float* part =
+      &db_part[(int64_t)blockIdx.x * (BC_N1 + BC_N2 + BC_N3 + 1 + 256)];
 
   bc_load_tile<BC_N1, BC_S1>(a1, t1, m0, M, tid);
   bc_load_tile<BC_N2, BC_S2>(a2, t2, m0, M, tid);
@@ -231,12 +236,18 @@ __global__ void __launch_bounds__(256) bwd_chain_kernel(
     for (int32_t m = 0; m < BC_MT; m++) s += dyf[m];
     part[BC_N1 + BC_N2 + BC_N3] = s;
   }
-  for (int32_t u = tid; u < BC_MT * BC_N3; u += 256) {
-    const int32_t m = u / BC_N3;
-    const int32_t k = u % BC_N3;
-    short* cell = &t3[m * BC_S3 + k];
-    const float live = bc_b2f(*cell) > 0.f ? 1.f : 0.f;
-    *cell = bc_f2b(dyf[m] * bc_b2f(w4[k]) * live);
+  {
+    float s4 = 0.f;  // dW4 partial: this thread's k is constant (tid%128)
+    for (int32_t u = tid; u < BC_MT * BC_N3; u += 256) {
+      const int32_t m = u / BC_N3;
+      const int32_t k = u % BC_N3;
+      short* cell = &t3[m * BC_S3 + k];
+      const float a3v = bc_b2f(*cell);
+      s4 += dyf[m] * a3v;
+      const float live = a3v > 0.f ? 1.f : 0.f;
+      *cell = bc_f2b(dyf[m] * bc_b2f(w4[k]) * live);
+    }
+    part[BC_N1 + BC_N2 + BC_N3 + 1 + tid] = s4;
   }
   __syncthreads();
 

@@ -198,11 +198,18 @@ __global__ void __launch_bounds__(256) fwd_chain_kernel(
     const short* __restrict__ W3, const float* __restrict__ b3,
     const short* __restrict__ w4, const float* __restrict__ b4,
     short* __restrict__ a1, short* __restrict__ a2,
-    short* __restrict__ a3, short* __restrict__ out, int64_t M) {
+    short* __restrict__ a3, short* __restrict__ out,
+    // Optional fused MSE epilogue (target != nullptr): dyb[m] =
+    // (2/M)*(out[m]-target[m]) in bf16 and loss_part[blockIdx] =
+    // sum_m (out[m]-target[m])^2 — removes the eager loss/grad kernel
+    // chain from the fused train step.
+    const float* __restrict__ target, short* __restrict__ dyb,
+    float* __restrict__ loss_part, float inv_m, int64_t M) {
   __shared__ __align__(16) short t0[FC_MT * FC_S0];
   __shared__ __align__(16) short t1[FC_MT * FC_S1];
   __shared__ __align__(16) short t2[FC_MT * FC_S2];
   __shared__ __align__(16) short t3[FC_MT * FC_S3];
+  __shared__ float lsum[FC_MT];
 
   const int64_t m0 = (int64_t)blockIdx.x * FC_MT;
   const int32_t tid = threadIdx.x;
@@ -247,8 +254,27 @@ __global__ void __launch_bounds__(256) fwd_chain_kernel(
     for (int32_t d = 1; d < TPR; d <<= 1) {
       s += __shfl_down(s, d);
     }
-    if (part == 0 && m0 + m < M) {
-      out[m0 + m] = fc_f2b(s + b4[0]);
+    if (part == 0) {
+      const float o = s + b4[0];
+      float d2 = 0.f;
+      if (m0 + m < M) {
+        out[m0 + m] = fc_f2b(o);
+        if (target != nullptr) {
+          const float diff = o - target[m0 + m];
+          dyb[m0 + m] = fc_f2b(2.f * inv_m * diff);
+          d2 = diff * diff;
+        }
+      }
+      if (target != nullptr) lsum[m] = d2;
+    }
+  }
+  if (target != nullptr) {
+    __syncthreads();
+    if (tid == 0) {
+      float s = 0.f;
+      #pragma unroll
+      for (int32_t m = 0; m < FC_MT; m++) s += lsum[m];
+      loss_part[blockIdx.x] = s;
     }
   }
 
@@ -257,11 +283,14 @@ __global__ void __launch_bounds__(256) fwd_chain_kernel(
   fc_store_tile<FC_N3, FC_S3>(t3, a3, m0, M, tid);
 }
 
+int64_t fwd_chain_grid(int64_t M) { return (M + FC_MT - 1) / FC_MT; }
+
 void launch_fwd_chain(const void* x0, const void* W1, const float* b1,
                       const void* W2, const float* b2, const void* W3,
                       const float* b3, const void* w4, const float* b4,
-                      void* a1, void* a2, void* a3, void* out, int64_t M,
-                      hipStream_t stream) {
+                      void* a1, void* a2, void* a3, void* out,
+                      const float* target, void* dyb, float* loss_part,
+                      int64_t M, hipStream_t stream) {
   const int32_t grid = (int32_t)((M + FC_MT - 1) / FC_MT);
   hipLaunchKernelGGL(fwd_chain_kernel, dim3(grid), dim3(256), 0, stream,
                      reinterpret_cast<const short*>(x0),
@@ -272,7 +301,9 @@ void launch_fwd_chain(const void* x0, const void* W1, const float* b1,
                      reinterpret_cast<short*>(a1),
                      reinterpret_cast<short*>(a2),
                      reinterpret_cast<short*>(a3),
-                     reinterpret_cast<short*>(out), M);
+                     reinterpret_cast<short*>(out), target,
+                     reinterpret_cast<short*>(dyb), loss_part,
+                     target ? 1.f / (float)M : 0.f, M);
 }
 
 }  // namespace rsdl

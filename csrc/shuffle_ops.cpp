@@ -68,11 +68,13 @@ void launch_wgrad_wide(const void* dy, const void* x, float* dW, float* db,
 void launch_relu_bwd_bias(const void* dy, const void* y, void* dx,
                           float* db_part, int64_t nvec, int32_t N,
                           int32_t grid, hipStream_t stream);
+int64_t fwd_chain_grid(int64_t M);
 void launch_fwd_chain(const void* x0, const void* W1, const float* b1,
                       const void* W2, const float* b2, const void* W3,
                       const float* b3, const void* w4, const float* b4,
-                      void* a1, void* a2, void* a3, void* out, int64_t M,
-                      hipStream_t stream);
+                      void* a1, void* a2, void* a3, void* out,
+                      const float* target, void* dyb, float* loss_part,
+                      int64_t M, hipStream_t stream);
 int64_t bwd_chain_grid(int64_t M);
 void launch_bwd_chain(const void* dy, const void* a1, const void* a2,
                       const void* a3, const void* w4, const void* W3T,
@@ -466,7 +468,8 @@ static at::Tensor swizzle_frag(const at::Tensor& W) {
 std::vector<at::Tensor> fwd_chain_bf16(
     const at::Tensor& x, const at::Tensor& W1, const at::Tensor& b1,
     const at::Tensor& W2, const at::Tensor& b2, const at::Tensor& W3,
-    const at::Tensor& b3, const at::Tensor& w4, const at::Tensor& b4) {
+    const at::Tensor& b3, const at::Tensor& w4, const at::Tensor& b4,
+    const c10::optional<at::Tensor>& target) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
                   x.dim() == 2 && x.size(1) == 100 && x.is_contiguous(),
               "fwd_chain: x must be contiguous bf16 [M,100]");
@@ -501,6 +504,24 @@ std::vector<at::Tensor> fwd_chain_bf16(
   auto a2 = at::empty({M, 256}, x.options());
   auto a3 = at::empty({M, 128}, x.options());
   auto out = at::empty({M, 1}, x.options());
+  // Fused MSE epilogue: with a target, the kernel also emits the bf16
+  // loss gradient dyb = (2/M)(out - target) and per-block squared-error
+  // partials (loss = loss_part.sum()/M) — no eager loss/grad kernels.
+  const bool with_loss = target.has_value();
+  at::Tensor tgt, dyb, loss_part;
+  const float* tgt_ptr = nullptr;
+  void* dyb_ptr = nullptr;
+  float* lp_ptr = nullptr;
+  if (with_loss) {
+    tgt = target->reshape({-1}).to(at::kFloat).contiguous();
+    TORCH_CHECK(tgt.numel() == M, "fwd_chain: target size mismatch");
+    dyb = at::empty({M, 1}, x.options());
+    loss_part = at::empty({std::max<int64_t>(fwd_chain_grid(M), 1)},
+                          x.options().dtype(at::kFloat));
+    tgt_ptr = tgt.data_ptr<float>();
+    dyb_ptr = dyb.data_ptr();
+    lp_ptr = loss_part.data_ptr<float>();
+  }
   if (M > 0) {
     auto W1s = swizzle_frag(W1p);
     auto W2s = swizzle_frag(W2);
@@ -509,8 +530,10 @@ std::vector<at::Tensor> fwd_chain_bf16(
                      W2s.data_ptr(), b2f.data_ptr<float>(), W3s.data_ptr(),
                      b3f.data_ptr<float>(), w4.data_ptr(),
                      b4f.data_ptr<float>(), a1.data_ptr(), a2.data_ptr(),
-                     a3.data_ptr(), out.data_ptr(), M, current_stream());
+                     a3.data_ptr(), out.data_ptr(), tgt_ptr, dyb_ptr,
+                     lp_ptr, M, current_stream());
   }
+  if (with_loss) return {a1, a2, a3, out, dyb, loss_part};
   return {a1, a2, a3, out};
 }
 
@@ -519,6 +542,17 @@ std::vector<at::Tensor> fwd_chain_bf16(
 // dgrad + relu mask + bias partials for the fixed TabularMLP
 // architecture. Returns (dz1, dz2, dz3, db1, db2, db3, db4); the wgrads
 // (dW_l = dz_l^T @ a_{l-1}) remain the caller's streaming kernels.
+// Swizzle of W^T computed directly from model-layout W [K,N] (single
+// strided copy, no intermediate transpose materialization).
+static at::Tensor swizzle_frag_T(const at::Tensor& W) {
+  int64_t K = W.size(0), N = W.size(1);
+  TORCH_CHECK(K % 16 == 0 && N % 32 == 0, "swizzle_frag_T: bad shape");
+  return W.view({K / 16, 2, 8, N / 32, 32})
+      .permute({3, 0, 1, 4, 2})
+      .contiguous()
+      .view({-1});
+}
+
 std::vector<at::Tensor> bwd_chain_bf16(
     const at::Tensor& dy, const at::Tensor& a1, const at::Tensor& a2,
     const at::Tensor& a3, const at::Tensor& w4, const at::Tensor& W3,
@@ -550,36 +584,46 @@ std::vector<at::Tensor> bwd_chain_bf16(
                    (W2.size(0) == 512 && W2.size(1) == 256)),
               "bwd_chain: W2 must be bf16 [256,512] or W2^T [512,256]");
   auto w4c = w4.contiguous();
-  auto W3T = (W3.size(0) == 256) ? W3.contiguous()
-                                 : W3.t().contiguous();  // [256,128]
-  auto W2T = (W2.size(0) == 512) ? W2.contiguous()
-                                 : W2.t().contiguous();  // [512,256]
+  // Swizzled W^T fragments, computed directly from whichever layout was
+  // passed (model [out,in] or pre-transposed [in,out]).
+  auto W3Ts = (W3.size(0) == 128) ? swizzle_frag_T(W3.contiguous())
+                                  : swizzle_frag(W3.contiguous());
+  auto W2Ts = (W2.size(0) == 256) ? swizzle_frag_T(W2.contiguous())
+                                  : swizzle_frag(W2.contiguous());
   auto dz1 = at::empty({M, 512}, dy.options());
   auto dz2 = at::empty({M, 256}, dy.options());
   auto dz3 = at::empty({M, 128}, dy.options());
   const int64_t grid = bwd_chain_grid(M);
+  constexpr int64_t kPartW = 512 + 256 + 128 + 1 + 256;
   // Every block writes its whole db_part row (bc_bias_partial covers all
-  // columns incl. db4), so empty() is safe — no memset kernel per call.
-  auto db_part = at::empty({std::max<int64_t>(grid, 1), 512 + 256 + 128 + 1},
+  // columns incl. db4 and the dW4 partials), so empty() is safe — no
+  // memset kernel per call.
+  auto db_part = at::empty({std::max<int64_t>(grid, 1), kPartW},
                            dy.options().dtype(at::kFloat));
   if (M == 0) db_part.zero_();
   if (M > 0) {
-    auto W3Ts = swizzle_frag(W3T);
-    auto W2Ts = swizzle_frag(W2T);
     launch_bwd_chain(dy.data_ptr(), a1.data_ptr(), a2.data_ptr(),
                      a3.data_ptr(), w4c.data_ptr(), W3Ts.data_ptr(),
                      W2Ts.data_ptr(), dz1.data_ptr(), dz2.data_ptr(),
                      dz3.data_ptr(), db_part.data_ptr<float>(), M,
                      current_stream());
   }
-  auto db = db_part.sum(0);
+  // Column reduction as a GEMV (ones^T @ db_part): hipBLASLt streams the
+  // 28 MB slab near roofline where the generic [rows,1153] column-reduce
+  // kernel ran at ~0.3 TB/s.
+  auto ones =
+      at::ones({1, db_part.size(0)}, dy.options().dtype(at::kFloat));
+  auto db = at::mm(ones, db_part).reshape({kPartW});
+  auto dw4 = (db.narrow(0, 897, 128) + db.narrow(0, 897 + 128, 128))
+                 .reshape({1, 128});
   return {dz1,
           dz2,
           dz3,
           db.narrow(0, 0, 512),
           db.narrow(0, 512, 256),
           db.narrow(0, 512 + 256, 128),
-          db.narrow(0, 512 + 256 + 128, 1)};
+          db.narrow(0, 512 + 256 + 128, 1),
+          dw4};
 }
 
 }  // namespace
@@ -605,7 +649,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad_bf16", &rsdl::wgrad_bf16, py::arg("dy"), py::arg("x"),
         py::arg("with_bias") = true);
   m.def("relu_bwd_bias", &rsdl::relu_bwd_bias, py::arg("dy"), py::arg("y"));
-  m.def("fwd_chain_bf16", &rsdl::fwd_chain_bf16);
+  m.def("fwd_chain_bf16", &rsdl::fwd_chain_bf16, py::arg("x"),
+        py::arg("W1"), py::arg("b1"), py::arg("W2"), py::arg("b2"),
+        py::arg("W3"), py::arg("b3"), py::arg("w4"), py::arg("b4"),
+        py::arg("target") = c10::nullopt);
   m.def("bwd_chain_bf16", &rsdl::bwd_chain_bf16);
   m.attr("DT_F32") = (int)rsdl::DT_F32;
   m.attr("DT_F64") = (int)rsdl::DT_F64;

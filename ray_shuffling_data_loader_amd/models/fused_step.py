@@ -59,24 +59,13 @@ def _weight_buffers(model: TabularMLP, lin):
             "W2": torch.empty(256, 512, **bf),
             "W3": torch.empty(128, 256, **bf),
             "w4": torch.empty(128, **bf),
-            "W2T": torch.empty(512, 256, **bf),
-            "W3T": torch.empty(256, 128, **bf),
         }
         model._fused_buf = buf
     buf["W1p"][:, :100].copy_(lin[0].weight.detach())
     buf["W2"].copy_(lin[1].weight.detach())
     buf["W3"].copy_(lin[2].weight.detach())
     buf["w4"].copy_(lin[3].weight.detach().view(-1))
-    buf["W2T"].copy_(buf["W2"].t())
-    buf["W3T"].copy_(buf["W3"].t())
     return buf
-
-
-def _head_wgrad(dyb: torch.Tensor, a3: torch.Tensor) -> torch.Tensor:
-    from ray_shuffling_data_loader_amd.ops.shuffle_ops import wgrad
-
-    dw, _ = wgrad(dyb, a3, with_bias=False)
-    return dw  # [1,128] fp32
 
 
 def fused_step(
@@ -93,25 +82,21 @@ def fused_step(
     M = x.shape[0]
     buf = _weight_buffers(model, lin)
     b1, b2, b3, b4 = (m.bias.detach() for m in lin)
-    a1, a2, a3, out = hip.fwd_chain_bf16(
-        x, buf["W1p"], b1, buf["W2"], b2, buf["W3"], b3, buf["w4"], b4
+    # Loss + dy are fused into the forward kernel's head epilogue.
+    a1, a2, a3, out, dyb, loss_part = hip.fwd_chain_bf16(
+        x, buf["W1p"], b1, buf["W2"], b2, buf["W3"], b3, buf["w4"], b4,
+        target=target,
     )
-    diff = out.float() - target.float().reshape(-1, 1)
-    loss = diff.square().mean()
-    dy = (2.0 / (M * 1.0)) * diff  # d(mean((out-t)^2))/d out
-    dyb = dy.bfloat16().contiguous()
-    dz1, dz2, dz3, db1, db2, db3, db4 = hip.bwd_chain_bf16(
-        dyb, a1, a2, a3, buf["w4"], buf["W3T"], buf["W2T"]
+    loss = loss_part.sum() / M
+    # dW4/db4 partials are folded into the backward kernel's seed loop.
+    dz1, dz2, dz3, db1, db2, db3, db4, dw4 = hip.bwd_chain_bf16(
+        dyb, a1, a2, a3, buf["w4"], buf["W3"], buf["W2"]
     )
     grads = [
         (_wgrad_bmm(dz1, x), db1),
         (_wgrad_bmm(dz2, a1), db2),
         (_wgrad_bmm(dz3, a2), db3),
-        # dW4 [1,128]: hand-written MFMA wgrad — beats the library on the
-        # degenerate head shape (profiles/PERF.md wgrad zoo); the chunked
-        # bmm at batch=50/N=1 hit a ~10 ms host stall in hipBLASLt
-        # dispatch, and the plain bf16 GEMM picked a 170 us kernel.
-        (_head_wgrad(dyb, a3), db4),
+        (dw4, db4),
     ]
     for m, (gw, gb) in zip(lin, grads):
         m.weight.grad = gw.to(m.weight.dtype)

@@ -569,6 +569,23 @@ def test_fwd_chain_matches_eager(dev):
         assert err.max() <= 0.12 * scale + 0.05, (
             name, err.max().item(), scale.item())
 
+    # Fused MSE epilogue (target given): dyb + loss partials.
+    tgt = torch.randn(M, 1, device=dev)
+    a1b, a2b, a3b, outb, dyb, loss_part = hip.fwd_chain_bf16(
+        x, Ws[0], bs[0], Ws[1], bs[1], Ws[2], bs[2],
+        Ws[3].flatten(), bs[3], target=tgt,
+    )
+    assert torch.equal(outb, out)
+    diff = outb.float() - tgt
+    ref_dy = (2.0 / M) * diff
+    assert torch.allclose(
+        dyb.float(), ref_dy.bfloat16().float(), atol=1e-6, rtol=0.02
+    )
+    ref_loss = diff.square().mean()
+    got_loss = loss_part.sum() / M
+    assert torch.allclose(got_loss, ref_loss, rtol=1e-3), (
+        got_loss.item(), ref_loss.item())
+
 
 @pytest.mark.gpu
 @pytest.mark.skipif(
@@ -590,7 +607,7 @@ def test_bwd_chain_matches_eager(dev):
     w4 = (torch.randn(128, device=dev) / 11).bfloat16()
     W3 = (torch.randn(128, 256, device=dev) / 16).bfloat16()
     W2 = (torch.randn(256, 512, device=dev) / 22).bfloat16()
-    dz1, dz2, dz3, db1, db2, db3, db4 = hip.bwd_chain_bf16(
+    dz1, dz2, dz3, db1, db2, db3, db4, dw4 = hip.bwd_chain_bf16(
         dy, a1, a2, a3, w4, W3, W2
     )
     # fp32 eager oracle
@@ -616,6 +633,11 @@ def test_bwd_chain_matches_eager(dev):
                               rtol=0.02), name
     assert torch.allclose(
         db4, dy.float().sum(0), atol=0.5, rtol=0.02)
+    # dW4 partials folded into the seed loop: dW4 = dy^T @ a3.
+    ref_dw4 = dy.float().t() @ a3.float()
+    assert torch.allclose(
+        dw4, ref_dw4, atol=0.5 + 0.02 * ref_dw4.abs().max(), rtol=0.02
+    ), (dw4 - ref_dw4).abs().max().item()
 
 
 @pytest.mark.gpu
