@@ -213,8 +213,31 @@ def main():
         output=args.loader_output,
     )
 
+    # Default train step: the fused chain-kernel path (fwd+loss+bwd in two
+    # hand-written MFMA kernels + split-K wgrads — measured faster than
+    # eager autocast; RSDL_FUSED_STEP=0 reverts). At world>1 the model is
+    # NOT DDP-wrapped on the fused path: gradients land in views of one
+    # flat buffer and are all-reduced with a single collective (DDP's
+    # autograd-hook bucketing never fires on a manual backward).
     model = TabularMLP(args.num_cols).to(device)
-    if world > 1:
+    use_fused = (
+        os.environ.get("RSDL_FUSED_STEP", "1") == "1"
+        and device.type == "cuda"
+        and args.dtype == "bf16"
+        and args.num_cols == 100
+    )
+    flat_grad = None
+    if use_fused:
+        if world > 1:
+            params = list(model.parameters())
+            flat_grad = torch.zeros(
+                sum(p.numel() for p in params), device=device
+            )
+            off = 0
+            for p in params:
+                p.grad = flat_grad[off : off + p.numel()].view_as(p)
+                off += p.numel()
+    elif world > 1:
         model = torch.nn.parallel.DistributedDataParallel(model)
     try:
         # Single fused multi-tensor update kernel (falls back where the
@@ -253,6 +276,11 @@ def main():
             )
         return contextlib.nullcontext()
 
+    if use_fused:
+        from ray_shuffling_data_loader_amd.models.fused_step import (
+            fused_step,
+        )
+
     def one_step():
         t_wait0 = time.perf_counter()
         cur_epoch[0], (data, target) = next(it)
@@ -261,6 +289,13 @@ def main():
         if x.device != device:
             x = x.to(device, non_blocking=True)
             target = target.to(device, non_blocking=True)
+        if use_fused:
+            fused_step(model, x, target)
+            if flat_grad is not None:
+                flat_grad.div_(world)
+                torch.distributed.all_reduce(flat_grad)
+            opt.step()
+            return wait
         opt.zero_grad(set_to_none=True)
         with amp():
             out = model(x)

@@ -99,6 +99,16 @@ def fused_step(
         (dw4, db4),
     ]
     for m, (gw, gb) in zip(lin, grads):
-        m.weight.grad = gw.to(m.weight.dtype)
-        m.bias.grad = gb.to(m.bias.dtype).reshape(m.bias.shape)
+        # Copy into pre-existing .grad buffers when present (the DP bench
+        # pre-creates them as views of one flat buffer so the gradient
+        # all-reduce is a single collective); otherwise assign.
+        if m.weight.grad is not None and m.weight.grad.shape == gw.shape:
+            m.weight.grad.copy_(gw)
+        else:
+            m.weight.grad = gw.to(m.weight.dtype)
+        gb = gb.reshape(m.bias.shape)
+        if m.bias.grad is not None:
+            m.bias.grad.copy_(gb)
+        else:
+            m.bias.grad = gb.to(m.bias.dtype)
     return loss
