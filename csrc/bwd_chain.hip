@@ -92,26 +92,18 @@ __device__ void bc_store_tile(const short* __restrict__ lds, short* out,
   }
 }
 
-// Column sums of a dz tile -> db_part[blockIdx][col] (fp32). 64-row sums,
-// one thread per (col stripe); plain LDS reads, epilogue-only.
-template <int N, int S>
-__device__ void bc_bias_partial(const short* __restrict__ dz,
-                                float* __restrict__ part, int32_t tid) {
-  for (int32_t c = tid; c < N; c += 256) {
-    float s = 0.f;
-    #pragma unroll 4
-    for (int32_t m = 0; m < BC_MT; m++) s += bc_b2f(dz[m * S + c]);
-    part[c] = s;
-  }
-}
-
 // One dgrad layer: da = dz_src @ W (via WT [N,K] contiguous), then
 // dz_dst = da * (a_dst > 0) written IN PLACE over the a_dst tile.
-// K = dz_src width (contraction), N = output width.
+// K = dz_src width (contraction), N = output width. The layer's bias
+// partials db[n] = sum_m dz[m,n] are folded into the epilogue (the
+// masked fp32 values are already in registers; lanes l/l+32 share a
+// column, one shfl_xor combines them) — the old per-column scalar-LDS
+// reduction pass (~112 LDS reads/thread) is gone.
 template <int K, int N, int SRC_S, int DST_S>
 __device__ void bc_layer(const short* __restrict__ dz_src,
                          const short* __restrict__ WT,
-                         short* __restrict__ a_dst, int32_t wave,
+                         short* __restrict__ a_dst,
+                         float* __restrict__ db_out, int32_t wave,
                          int32_t lane) {
   constexpr int NT = N / 128;
   constexpr int ITERS = K / 16;
@@ -178,19 +170,24 @@ __device__ void bc_layer(const short* __restrict__ dz_src,
   // n-range of a_dst; inter-layer ordering is handled by the
   // __syncthreads() between bc_layer calls in the kernel body.
   #pragma unroll
-  for (int mt = 0; mt < BC_MTILES; mt++) {
+  for (int nt = 0; nt < NT; nt++) {
+    const int32_t n = n_base + nt * 32 + ml;
+    float colsum = 0.f;
     #pragma unroll
-    for (int nt = 0; nt < NT; nt++) {
-      const int32_t n = n_base + nt * 32 + ml;
+    for (int mt = 0; mt < BC_MTILES; mt++) {
       #pragma unroll
       for (int reg = 0; reg < 16; reg++) {
         const int32_t mrow =
             mt * 32 + (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
         short* cell = &a_dst[mrow * DST_S + n];
         const float live = bc_b2f(*cell) > 0.f ? 1.f : 0.f;
-        *cell = bc_f2b(acc[mt][nt][reg] * live);
+        const float v = acc[mt][nt][reg] * live;
+        *cell = bc_f2b(v);
+        colsum += v;
       }
     }
+    colsum += __shfl_xor(colsum, 32);
+    if (lane < 32) db_out[n] = colsum;
   }
 }
 
@@ -214,6 +211,7 @@ __global__ void __launch_bounds__(256) bwd_chain_kernel(
   __shared__ __align__(16) short t2[BC_MT * BC_S2];
   __shared__ __align__(16) short t3[BC_MT * BC_S3];
   __shared__ float dyf[BC_MT];
+  __shared__ float s3sh[256];
 
   const int64_t m0 = (int64_t)blockIdx.x * BC_MT;
   const int32_t tid = threadIdx.x;
@@ -230,7 +228,8 @@ __global__ void __launch_bounds__(256) bwd_chain_kernel(
   }
   __syncthreads();
 
-  // dz3 = (dy * w4) * (a3 > 0), in place over t3; also db4 partial.
+  // dz3 = (dy * w4) * (a3 > 0), in place over t3; also db4 + dW4 + db3
+  // partials (dy and a3 are at hand here).
   if (tid == 0) {
     float s = 0.f;
     for (int32_t m = 0; m < BC_MT; m++) s += dyf[m];
@@ -238,6 +237,7 @@ __global__ void __launch_bounds__(256) bwd_chain_kernel(
   }
   {
     float s4 = 0.f;  // dW4 partial: this thread's k is constant (tid%128)
+    float s3 = 0.f;  // db3 partial for the same k
     for (int32_t u = tid; u < BC_MT * BC_N3; u += 256) {
       const int32_t m = u / BC_N3;
       const int32_t k = u % BC_N3;
@@ -245,22 +245,27 @@ __global__ void __launch_bounds__(256) bwd_chain_kernel(
       const float a3v = bc_b2f(*cell);
       s4 += dyf[m] * a3v;
       const float live = a3v > 0.f ? 1.f : 0.f;
-      *cell = bc_f2b(dyf[m] * bc_b2f(w4[k]) * live);
+      const float dz3v = dyf[m] * bc_b2f(w4[k]) * live;
+      *cell = bc_f2b(dz3v);
+      s3 += dz3v;
     }
     part[BC_N1 + BC_N2 + BC_N3 + 1 + tid] = s4;
+    s3sh[tid] = s3;
   }
   __syncthreads();
+  if (tid < BC_N3) {
+    part[BC_N1 + BC_N2 + tid] = s3sh[tid] + s3sh[tid + BC_N3];
+  }
 
-  // da2 = dz3 @ W3 (via W3T), mask by a2 -> dz2 in place over t2.
-  bc_layer<BC_N3, BC_N2, BC_S3, BC_S2>(t3, W3T, t2, wave, lane);
+  // da2 = dz3 @ W3 (via W3T), mask by a2 -> dz2 in place over t2;
+  // db2 partials written by the epilogue.
+  bc_layer<BC_N3, BC_N2, BC_S3, BC_S2>(t3, W3T, t2, &part[BC_N1], wave,
+                                       lane);
   __syncthreads();
-  // da1 = dz2 @ W2 (via W2T), mask by a1 -> dz1 in place over t1.
-  bc_layer<BC_N2, BC_N1, BC_S2, BC_S1>(t2, W2T, t1, wave, lane);
+  // da1 = dz2 @ W2 (via W2T), mask by a1 -> dz1 in place over t1; db1.
+  bc_layer<BC_N2, BC_N1, BC_S2, BC_S1>(t2, W2T, t1, &part[0], wave,
+                                       lane);
   __syncthreads();
-
-  bc_bias_partial<BC_N1, BC_S1>(t1, &part[0], tid);
-  bc_bias_partial<BC_N2, BC_S2>(t2, &part[BC_N1], tid);
-  bc_bias_partial<BC_N3, BC_S3>(t3, &part[BC_N1 + BC_N2], tid);
 
   bc_store_tile<BC_N1, BC_S1>(t1, dz1, m0, M, tid);
   bc_store_tile<BC_N2, BC_S2>(t2, dz2, m0, M, tid);

@@ -51,6 +51,7 @@ namespace rsdl {
 typedef __attribute__((__vector_size__(8 * sizeof(short)))) short fc_bf16x8;
 typedef __attribute__((__vector_size__(16 * sizeof(float)))) float fc_f32x16;
 typedef __attribute__((__vector_size__(4 * sizeof(unsigned int)))) unsigned int fc_u32x4;
+typedef __attribute__((__vector_size__(2 * sizeof(unsigned int)))) unsigned int fc_u32x2;
 
 #define FC_MT 32            // rows per workgroup slab
 #define FC_MTILES (FC_MT / 32)  // 32-row MFMA m-tiles per slab
@@ -216,16 +217,20 @@ __global__ void __launch_bounds__(256) fwd_chain_kernel(
   const int32_t wave = tid >> 6;
   const int32_t lane = tid & 63;
 
-  // Stage the x0 slab (zero-padding cols 100..111 and rows past M). This
-  // is a one-time per-WG load, not the hot loop — scalar guards are fine.
-  for (int32_t u = tid; u < FC_MT * FC_K0P; u += 256) {
-    const int32_t m = u / FC_K0P;
-    const int32_t c = u % FC_K0P;
-    short v = 0;
-    if (m0 + m < M && c < FC_K0) {
-      v = __builtin_nontemporal_load(&x0[(m0 + m) * FC_K0 + c]);
+  // Stage the x0 slab (zero-padding cols 100..111 and rows past M),
+  // 8-byte vectors (row stride 200 B is 8-B aligned).
+  {
+    constexpr int VPR = FC_K0P / 4;  // uint2 slots per padded row
+    for (int32_t u = tid; u < FC_MT * VPR; u += 256) {
+      const int32_t m = u / VPR;
+      const int32_t c = (u % VPR) * 4;
+      fc_u32x2 v = {0, 0};
+      if (m0 + m < M && c < FC_K0) {
+        v = __builtin_nontemporal_load(
+            reinterpret_cast<const fc_u32x2*>(&x0[(m0 + m) * FC_K0 + c]));
+      }
+      *reinterpret_cast<fc_u32x2*>(&t0[m * FC_S0 + c]) = v;
     }
-    t0[m * FC_S0 + c] = v;
   }
   __syncthreads();
 
