@@ -82,7 +82,12 @@ __device__ __forceinline__ short fc_f2b(float f) {
 // halfwords) x W[N][K] global -> relu(.+bias) -> dst LDS tile (stride
 // DST_S). N_WAVE = output columns per wave (N/4). Each wave owns m-tiles
 // {0,32} x its n-range; accumulators are static f32x16 arrays.
-template <int K, int N, int SRC_S, int DST_S, bool RELU>
+// A_FRAGMAJOR: src is a GLOBAL fragment-major m-tile block (layout as
+// the swizzled weights: [kc][h][ml][8], 512 halfwords per k-chunk) —
+// used for layer 1, whose input x is pre-swizzled host-side; SRC_S is
+// ignored. Otherwise src is the LDS tile of the previous layer.
+template <int K, int N, int SRC_S, int DST_S, bool RELU,
+          bool A_FRAGMAJOR = false>
 __device__ void fc_layer(const short* __restrict__ src_lds,
                          const short* __restrict__ W,
                          const float* __restrict__ bias,
@@ -103,7 +108,9 @@ __device__ void fc_layer(const short* __restrict__ src_lds,
   const short* srcA[FC_MTILES];
   #pragma unroll
   for (int mt = 0; mt < FC_MTILES; mt++) {
-    srcA[mt] = &src_lds[(mt * 32 + ml) * SRC_S + frag_k0];
+    srcA[mt] = A_FRAGMAJOR
+                   ? &src_lds[(int64_t)mt * ITERS * 512 + lane * 8]
+                   : &src_lds[(mt * 32 + ml) * SRC_S + frag_k0];
   }
   // Swizzled weight base for this wave's n-tiles: block (ntile, kc) is
   // 512 contiguous halfwords; lane's 16-B slice at lane*8.
@@ -135,7 +142,8 @@ __device__ void fc_layer(const short* __restrict__ src_lds,
       #pragma unroll
       for (int mt = 0; mt < FC_MTILES; mt++) {
         *reinterpret_cast<uint4*>(&a[nxt][mt]) =
-            *reinterpret_cast<const uint4*>(&srcA[mt][k]);
+            *reinterpret_cast<const uint4*>(
+                &srcA[mt][A_FRAGMAJOR ? (i + 1) * 512 : k]);
       }
       #pragma unroll
       for (int nt = 0; nt < NT; nt++) {
@@ -193,7 +201,7 @@ __device__ void fc_store_tile(const short* __restrict__ lds, short* out,
 }
 
 __global__ void __launch_bounds__(256) fwd_chain_kernel(
-    const short* __restrict__ x0,  // [M, 100] bf16 bits
+    const short* __restrict__ x0s,  // fragment-major swizzled x (see note)
     const short* __restrict__ W1, const float* __restrict__ b1,
     const short* __restrict__ W2, const float* __restrict__ b2,
     const short* __restrict__ W3, const float* __restrict__ b3,
@@ -206,37 +214,33 @@ __global__ void __launch_bounds__(256) fwd_chain_kernel(
     // chain from the fused train step.
     const float* __restrict__ target, short* __restrict__ dyb,
     float* __restrict__ loss_part, float inv_m, int64_t M) {
-  __shared__ __align__(16) short t0[FC_MT * FC_S0];
-  __shared__ __align__(16) short t1[FC_MT * FC_S1];
-  __shared__ __align__(16) short t2[FC_MT * FC_S2];
-  __shared__ __align__(16) short t3[FC_MT * FC_S3];
-  __shared__ float lsum[FC_MT];
+  // LDS budget is the occupancy lever: t1 + t2 + lsum = ~50 KB -> 3
+  // workgroups/CU. x needs no tile (pre-swizzled fragment-major in
+  // global, coalesced A loads), and t3 ALIASES t1's storage — a1 is
+  // stored to global right after layer 2 consumes it, before layer 3's
+  // epilogue writes t3.
+  __shared__ __align__(16) char smem[FC_MT * FC_S1 * 2 + FC_MT * FC_S2 * 2 +
+                                     FC_MT * 4];
+  short* t1 = reinterpret_cast<short*>(smem);
+  short* t2 = reinterpret_cast<short*>(smem + FC_MT * FC_S1 * 2);
+  short* t3 = t1;  // aliased: live ranges are disjoint (barrier-ordered)
+  float* lsum = reinterpret_cast<float*>(smem + FC_MT * FC_S1 * 2 +
+                                         FC_MT * FC_S2 * 2);
 
   const int64_t m0 = (int64_t)blockIdx.x * FC_MT;
   const int32_t tid = threadIdx.x;
   const int32_t wave = tid >> 6;
   const int32_t lane = tid & 63;
 
-  // Stage the x0 slab (zero-padding cols 100..111 and rows past M),
-  // 8-byte vectors (row stride 200 B is 8-B aligned).
-  {
-    constexpr int VPR = FC_K0P / 4;  // uint2 slots per padded row
-    for (int32_t u = tid; u < FC_MT * VPR; u += 256) {
-      const int32_t m = u / VPR;
-      const int32_t c = (u % VPR) * 4;
-      fc_u32x2 v = {0, 0};
-      if (m0 + m < M && c < FC_K0) {
-        v = __builtin_nontemporal_load(
-            reinterpret_cast<const fc_u32x2*>(&x0[(m0 + m) * FC_K0 + c]));
-      }
-      *reinterpret_cast<fc_u32x2*>(&t0[m * FC_S0 + c]) = v;
-    }
-  }
-  __syncthreads();
-
-  fc_layer<FC_K0P, FC_N1, FC_S0, FC_S1, true>(t0, W1, b1, t1, wave, lane);
+  // Layer 1: A fragments straight from the swizzled global x block.
+  const short* xblk = &x0s[(int64_t)blockIdx.x * (FC_K0P / 16) * 512];
+  fc_layer<FC_K0P, FC_N1, 0, FC_S1, true, true>(xblk, W1, b1, t1, wave,
+                                                lane);
   __syncthreads();
   fc_layer<FC_N1, FC_N2, FC_S1, FC_S2, true>(t1, W2, b2, t2, wave, lane);
+  __syncthreads();
+  // a1 leaves LDS now so layer 3 can reuse t1's storage for t3.
+  fc_store_tile<FC_N1, FC_S1>(t1, a1, m0, M, tid);
   __syncthreads();
   fc_layer<FC_N2, FC_N3, FC_S2, FC_S3, true>(t2, W3, b3, t3, wave, lane);
   __syncthreads();
@@ -283,14 +287,13 @@ __global__ void __launch_bounds__(256) fwd_chain_kernel(
     }
   }
 
-  fc_store_tile<FC_N1, FC_S1>(t1, a1, m0, M, tid);
   fc_store_tile<FC_N2, FC_S2>(t2, a2, m0, M, tid);
   fc_store_tile<FC_N3, FC_S3>(t3, a3, m0, M, tid);
 }
 
 int64_t fwd_chain_grid(int64_t M) { return (M + FC_MT - 1) / FC_MT; }
 
-void launch_fwd_chain(const void* x0, const void* W1, const float* b1,
+void launch_fwd_chain(const void* x0s, const void* W1, const float* b1,
                       const void* W2, const float* b2, const void* W3,
                       const float* b3, const void* w4, const float* b4,
                       void* a1, void* a2, void* a3, void* out,
@@ -298,7 +301,7 @@ void launch_fwd_chain(const void* x0, const void* W1, const float* b1,
                       int64_t M, hipStream_t stream) {
   const int32_t grid = (int32_t)((M + FC_MT - 1) / FC_MT);
   hipLaunchKernelGGL(fwd_chain_kernel, dim3(grid), dim3(256), 0, stream,
-                     reinterpret_cast<const short*>(x0),
+                     reinterpret_cast<const short*>(x0s),
                      reinterpret_cast<const short*>(W1), b1,
                      reinterpret_cast<const short*>(W2), b2,
                      reinterpret_cast<const short*>(W3), b3,

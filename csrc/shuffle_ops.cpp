@@ -526,7 +526,18 @@ std::vector<at::Tensor> fwd_chain_bf16(
     auto W1s = swizzle_frag(W1p);
     auto W2s = swizzle_frag(W2);
     auto W3s = swizzle_frag(W3);
-    launch_fwd_chain(x.data_ptr(), W1s.data_ptr(), b1f.data_ptr<float>(),
+    // x pre-swizzled to the same fragment-major layout (per 32-row
+    // m-tile): layer 1's A fragments then come straight from global,
+    // coalesced, and the kernel needs no x LDS tile (occupancy 2 -> 3
+    // workgroups/CU). Pad cols must be ZERO (uninitialized bf16 could be
+    // NaN; NaN * W1pad(0) would poison real rows) — constant_pad_nd
+    // provides that.
+    const int64_t Mp = (M + 31) / 32 * 32;
+    auto xp = at::constant_pad_nd(x, {0, 12, 0, Mp - M});
+    auto xs = xp.view({Mp / 32, 32, 7, 2, 8})
+                  .permute({0, 2, 3, 1, 4})
+                  .contiguous();
+    launch_fwd_chain(xs.data_ptr(), W1s.data_ptr(), b1f.data_ptr<float>(),
                      W2s.data_ptr(), b2f.data_ptr<float>(), W3s.data_ptr(),
                      b3f.data_ptr<float>(), w4.data_ptr(),
                      b4f.data_ptr<float>(), a1.data_ptr(), a2.data_ptr(),

@@ -531,11 +531,6 @@ def test_chunked_linear_scalar_head_grads(dev):
 
 
 @pytest.mark.gpu
-@pytest.mark.skipif(
-    os.environ.get("RSDL_EXPERIMENTAL") != "1",
-    reason="round-2 WIP kernel (docs/MEGAKERNEL_PLAN.md); "
-    "opt in with RSDL_EXPERIMENTAL=1",
-)
 def test_fwd_chain_matches_eager(dev):
     """EXPERIMENTAL fused forward chain vs the eager fp32 reference."""
     from ray_shuffling_data_loader_amd.ops import shuffle_ops
@@ -588,11 +583,6 @@ def test_fwd_chain_matches_eager(dev):
 
 
 @pytest.mark.gpu
-@pytest.mark.skipif(
-    os.environ.get("RSDL_EXPERIMENTAL") != "1",
-    reason="round-2 WIP kernel (docs/MEGAKERNEL_PLAN.md); "
-    "opt in with RSDL_EXPERIMENTAL=1",
-)
 def test_bwd_chain_matches_eager(dev):
     """EXPERIMENTAL fused backward chain vs the eager fp32 reference."""
     from ray_shuffling_data_loader_amd.ops import shuffle_ops
@@ -641,11 +631,6 @@ def test_bwd_chain_matches_eager(dev):
 
 
 @pytest.mark.gpu
-@pytest.mark.skipif(
-    os.environ.get("RSDL_EXPERIMENTAL") != "1",
-    reason="round-2 WIP fused step (docs/MEGAKERNEL_PLAN.md); "
-    "opt in with RSDL_EXPERIMENTAL=1",
-)
 def test_fused_step_matches_eager(dev):
     """EXPERIMENTAL whole-step parity: fused_step's loss and param grads
     vs the eager autocast fwd+bwd on identical weights and batch."""
