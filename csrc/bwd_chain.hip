@@ -197,21 +197,28 @@ __global__ void __launch_bounds__(256) bwd_chain_kernel(
     const short* __restrict__ a2,    // [M,256]
     const short* __restrict__ a3,    // [M,128]
     const short* __restrict__ w4,    // [128]
-    const short* __restrict__ W3T,   // [256,128]  (W3^T, contiguous)
-    const short* __restrict__ W2T,   // [512,256]  (W2^T, contiguous)
+    const short* __restrict__ W3T,   // swizzled W3^T fragments
+    const short* __restrict__ W2T,   // swizzled W2^T fragments
     short* __restrict__ dz1, short* __restrict__ dz2,
     short* __restrict__ dz3,
     // [grid][512+256+128+1+256]: db1|db2|db3|db4|dW4 partials. dW4[k] =
     // sum_m dy[m]*a3[m,k] folded into the dz3 seed loop (a3 is already
-    // in LDS there) — replaces a separate 54 us head-wgrad kernel; the
-    // two per-k thread partials (even/odd m) are summed host-side.
+    // in LDS there); the two per-k thread partials (even/odd m) are
+    // summed host-side.
     float* __restrict__ db_part,
     int64_t M) {
-  __shared__ __align__(16) short t1[BC_MT * BC_S1];
-  __shared__ __align__(16) short t2[BC_MT * BC_S2];
-  __shared__ __align__(16) short t3[BC_MT * BC_S3];
-  __shared__ float dyf[BC_MT];
-  __shared__ float s3sh[256];
+  // LDS is the occupancy lever: t3 (dz3) ALIASES t1's storage — a1 is
+  // loaded only after dz3 has been stored to global, so t1+t2+epsilon
+  // = ~50 KB -> 3 workgroups/CU (the monolithic t1+t2+t3 layout was
+  // 59 KB -> 2).
+  __shared__ __align__(16) char smem[BC_MT * BC_S1 * 2 + BC_MT * BC_S2 * 2 +
+                                     BC_MT * 4 + 256 * 4];
+  short* t1 = reinterpret_cast<short*>(smem);
+  short* t3 = t1;  // aliased: live ranges disjoint (barrier-ordered)
+  short* t2 = reinterpret_cast<short*>(smem + BC_MT * BC_S1 * 2);
+  float* dyf = reinterpret_cast<float*>(smem + BC_MT * BC_S1 * 2 +
+                                        BC_MT * BC_S2 * 2);
+  float* s3sh = dyf + BC_MT;
 
   const int64_t m0 = (int64_t)blockIdx.x * BC_MT;
   const int32_t tid = threadIdx.x;
@@ -220,7 +227,6 @@ __global__ void __launch_bounds__(256) bwd_chain_kernel(
   float* part =
       &db_part[(int64_t)blockIdx.x * (BC_N1 + BC_N2 + BC_N3 + 1 + 256)];
 
-  bc_load_tile<BC_N1, BC_S1>(a1, t1, m0, M, tid);
   bc_load_tile<BC_N2, BC_S2>(a2, t2, m0, M, tid);
   bc_load_tile<BC_N3, BC_S3>(a3, t3, m0, M, tid);
   if (tid < BC_MT) {
@@ -262,14 +268,16 @@ __global__ void __launch_bounds__(256) bwd_chain_kernel(
   bc_layer<BC_N3, BC_N2, BC_S3, BC_S2>(t3, W3T, t2, &part[BC_N1], wave,
                                        lane);
   __syncthreads();
+  bc_store_tile<BC_N3, BC_S3>(t3, dz3, m0, M, tid);
+  bc_store_tile<BC_N2, BC_S2>(t2, dz2, m0, M, tid);
+  __syncthreads();  // t3 fully read before a1 overwrites its storage
+  bc_load_tile<BC_N1, BC_S1>(a1, t1, m0, M, tid);
+  __syncthreads();
   // da1 = dz2 @ W2 (via W2T), mask by a1 -> dz1 in place over t1; db1.
   bc_layer<BC_N2, BC_N1, BC_S2, BC_S1>(t2, W2T, t1, &part[0], wave,
                                        lane);
   __syncthreads();
-
   bc_store_tile<BC_N1, BC_S1>(t1, dz1, m0, M, tid);
-  bc_store_tile<BC_N2, BC_S2>(t2, dz2, m0, M, tid);
-  bc_store_tile<BC_N3, BC_S3>(t3, dz3, m0, M, tid);
 }
 
 int64_t bwd_chain_grid(int64_t M) { return (M + BC_MT - 1) / BC_MT; }

@@ -291,6 +291,54 @@ __global__ void __launch_bounds__(256) fwd_chain_kernel(
   fc_store_tile<FC_N3, FC_S3>(t3, a3, m0, M, tid);
 }
 
+// x [M,100] bf16 -> fragment-major [ceil(M/32)][7][2][32][8] with zero
+// padding (cols 100..111 and rows past M). One thread per 8-halfword
+// fragment slice; replaces a 3-kernel pad+permute+contiguous chain.
+__global__ void __launch_bounds__(256) swizzle_x_kernel(
+    const short* __restrict__ x, short* __restrict__ out, int64_t M,
+    int64_t total_blocks) {
+  const int64_t b = (int64_t)blockIdx.x * 256 + threadIdx.x;
+  if (b >= total_blocks) return;
+  // b = ((mt*7 + kc)*2 + h)*32 + ml
+  const int32_t ml = (int32_t)(b & 31);
+  int64_t r = b >> 5;
+  const int32_t h = (int32_t)(r & 1);
+  r >>= 1;
+  const int32_t kc = (int32_t)(r % 7);
+  const int64_t mt = r / 7;
+  const int64_t m = mt * 32 + ml;
+  const int32_t k0 = kc * 16 + h * 8;
+  short v[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (m < M) {
+    const short* row = &x[m * FC_K0];
+    if (k0 + 8 <= FC_K0) {
+      // rows are 8-B aligned (100 * 2 B stride), not 16-B: two 8-B loads
+      *reinterpret_cast<fc_u32x2*>(&v[0]) = __builtin_nontemporal_load(
+          reinterpret_cast<const fc_u32x2*>(&row[k0]));
+      *reinterpret_cast<fc_u32x2*>(&v[4]) = __builtin_nontemporal_load(
+          reinterpret_cast<const fc_u32x2*>(&row[k0 + 4]));
+    } else {
+      #pragma unroll
+      for (int j = 0; j < 8; j++) {
+        if (k0 + j < FC_K0) v[j] = row[k0 + j];
+      }
+    }
+  }
+  __builtin_nontemporal_store(*reinterpret_cast<fc_u32x4*>(v),
+                              reinterpret_cast<fc_u32x4*>(&out[b * 8]));
+}
+
+void launch_swizzle_x(const void* x, void* out, int64_t M,
+                      hipStream_t stream) {
+  const int64_t mtiles = (M + FC_MT - 1) / FC_MT;
+  // fragment blocks per 32-row m-tile: 7 kc * 2 h * 32 ml
+  const int64_t total = mtiles * (FC_K0P / 16) * 2 * 32;
+  const int64_t grid = (total + 255) / 256;
+  hipLaunchKernelGGL(swizzle_x_kernel, dim3((uint32_t)grid), dim3(256), 0,
+                     stream, reinterpret_cast<const short*>(x),
+                     reinterpret_cast<short*>(out), M, total);
+}
+
 int64_t fwd_chain_grid(int64_t M) { return (M + FC_MT - 1) / FC_MT; }
 
 void launch_fwd_chain(const void* x0s, const void* W1, const float* b1,

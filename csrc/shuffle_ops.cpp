@@ -69,6 +69,8 @@ void launch_relu_bwd_bias(const void* dy, const void* y, void* dx,
                           float* db_part, int64_t nvec, int32_t N,
                           int32_t grid, hipStream_t stream);
 int64_t fwd_chain_grid(int64_t M);
+void launch_swizzle_x(const void* x, void* out, int64_t M,
+                      hipStream_t stream);
 void launch_fwd_chain(const void* x0, const void* W1, const float* b1,
                       const void* W2, const float* b2, const void* W3,
                       const float* b3, const void* w4, const float* b4,
@@ -529,14 +531,12 @@ std::vector<at::Tensor> fwd_chain_bf16(
     // x pre-swizzled to the same fragment-major layout (per 32-row
     // m-tile): layer 1's A fragments then come straight from global,
     // coalesced, and the kernel needs no x LDS tile (occupancy 2 -> 3
-    // workgroups/CU). Pad cols must be ZERO (uninitialized bf16 could be
-    // NaN; NaN * W1pad(0) would poison real rows) — constant_pad_nd
-    // provides that.
+    // workgroups/CU). Pads are zeroed by the swizzle kernel
+    // (uninitialized bf16 could be NaN; NaN * W1pad(0) would poison
+    // real rows).
     const int64_t Mp = (M + 31) / 32 * 32;
-    auto xp = at::constant_pad_nd(x, {0, 12, 0, Mp - M});
-    auto xs = xp.view({Mp / 32, 32, 7, 2, 8})
-                  .permute({0, 2, 3, 1, 4})
-                  .contiguous();
+    auto xs = at::empty({Mp * 112}, x.options());
+    launch_swizzle_x(x.data_ptr(), xs.data_ptr(), M, current_stream());
     launch_fwd_chain(xs.data_ptr(), W1s.data_ptr(), b1f.data_ptr<float>(),
                      W2s.data_ptr(), b2f.data_ptr<float>(), W3s.data_ptr(),
                      b3f.data_ptr<float>(), w4.data_ptr(),
