@@ -230,6 +230,9 @@ def main():
     if use_fused:
         if world > 1:
             params = list(model.parameters())
+            # DDP would broadcast at wrap time; do it explicitly here.
+            for p in params:
+                torch.distributed.broadcast(p.data, src=0)
             flat_grad = torch.zeros(
                 sum(p.numel() for p in params), device=device
             )
