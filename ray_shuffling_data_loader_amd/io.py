@@ -121,26 +121,52 @@ def _read_row_groups_columns(
     pin: bool,
 ) -> Dict[str, torch.Tensor]:
     """Decode a run of row groups of one file into column tensors (single
-    Arrow copy into pinned memory when ``pin``). use_threads=False: task
+    Arrow copy into ONE pinned arena when ``pin`` — per-column pinned
+    allocations at row-group task granularity meant thousands of
+    hipHostMalloc/H2D calls per ingest). use_threads=False: task
     granularity already saturates the reader pool; nested Arrow threads
     only oversubscribe."""
     pf = pq.ParquetFile(filename)
     table = pf.read_row_groups(
         groups, columns=schema.names, use_threads=False
     )
-    out = {}
+    arrs = {}
+    metas = []  # (name, byte offset, nbytes)
+    total = 0
     for spec in schema.columns:
         arr = table.column(spec.name).to_numpy(zero_copy_only=False)
         if not arr.flags["C_CONTIGUOUS"]:
             arr = np.ascontiguousarray(arr)
+        arrs[spec.name] = arr
+        total = (total + 15) & ~15  # keep every column slice 16-B aligned
+        metas.append((spec.name, total, arr.nbytes))
+        total += arr.nbytes
+    if not pin:
+        out = {}
+        for name, _, _ in metas:
+            with warnings.catch_warnings():
+                warnings.simplefilter("ignore")
+                out[name] = torch.from_numpy(arrs[name])
+        return out
+    arena = torch.empty(total, dtype=torch.uint8, pin_memory=True)
+    for name, off, nb in metas:
         with warnings.catch_warnings():
             warnings.simplefilter("ignore")
-            t = torch.from_numpy(arr)
-        if pin:
-            p = torch.empty_like(t, pin_memory=True)
-            p.copy_(t)
-            t = p
-        out[spec.name] = t
+            arena[off : off + nb].copy_(
+                torch.from_numpy(arrs[name].view(np.uint8).ravel())
+            )
+    return {"__arena__": arena, "__metas__": metas}
+
+
+def _arena_views(
+    arena: torch.Tensor, metas, schema: Schema
+) -> Dict[str, torch.Tensor]:
+    """Typed column views into a (host or device) ingest arena."""
+    out = {}
+    for name, off, nb in metas:
+        spec = schema.col(name)
+        t = arena[off : off + nb].view(spec.dtype)
+        out[name] = t if spec.numel == 1 else t.view(-1, spec.numel)
     return out
 
 
@@ -250,11 +276,11 @@ def read_files_packed(
         while pending:
             done, pending = wait(pending, return_when=FIRST_COMPLETED)
             for fut in done:
-                dst, n_rows, cols_host = fut.result()
-                cols = {
-                    name: t.to(device, non_blocking=True)
-                    for name, t in cols_host.items()
-                }
+                dst, n_rows, host = fut.result()
+                # ONE async H2D per task (the pinned arena), then typed
+                # device views feed the pack kernel.
+                d_arena = host["__arena__"].to(device, non_blocking=True)
+                cols = _arena_views(d_arena, host["__metas__"], schema)
                 pack_columns(cols, schema, out=packed[dst : dst + n_rows])
                 nxt = next(it, None)
                 if nxt is not None:
