@@ -78,6 +78,9 @@ void launch_fwd_chain(const void* x0, const void* W1, const float* b1,
                       const float* target, void* dyb, float* loss_part,
                       int64_t M, hipStream_t stream);
 int64_t bwd_chain_grid(int64_t M);
+void launch_wgrad_frag(const void* AT, const void* BT, float* dW, int32_t N,
+                       int32_t K, int64_t mchunks, int32_t nt_w,
+                       int32_t kt_w, hipStream_t stream);
 void launch_bwd_chain(const void* dy, const void* a1, const void* a2,
                       const void* a3, const void* w4, const void* W3T,
                       const void* W2T, void* dz1, void* dz2, void* dz3,
@@ -638,6 +641,32 @@ std::vector<at::Tensor> bwd_chain_bf16(
 }
 
 }  // namespace
+// Fragment-major wgrad (csrc/wgrad_frag.hip): dW = dz^T @ src with both
+// inputs pre-swizzled to [C/32][Mp/16][2][32][8] fragment layout.
+at::Tensor wgrad_frag_bf16(const at::Tensor& AT, const at::Tensor& BT,
+                           int64_t N, int64_t K, int64_t mchunks,
+                           int64_t nt_w, int64_t kt_w) {
+  TORCH_CHECK(AT.is_cuda() && AT.scalar_type() == at::kBFloat16 &&
+                  AT.is_contiguous() && BT.is_cuda() &&
+                  BT.scalar_type() == at::kBFloat16 && BT.is_contiguous(),
+              "wgrad_frag: inputs must be contiguous bf16 on GPU");
+  TORCH_CHECK(AT.numel() == (N / 32) * mchunks * 512,
+              "wgrad_frag: AT numel mismatch");
+  TORCH_CHECK(BT.numel() == (K / 32) * mchunks * 512,
+              "wgrad_frag: BT numel mismatch");
+  TORCH_CHECK((nt_w == 2 && kt_w == 4) || (nt_w == 1 && kt_w == 8),
+              "wgrad_frag: unsupported tile config");
+  TORCH_CHECK(N % (nt_w * 128) == 0 && K % (kt_w * 32) == 0,
+              "wgrad_frag: N/K not divisible by block shape");
+  auto dW = at::zeros({N, K}, AT.options().dtype(at::kFloat));
+  if (mchunks > 0) {
+    launch_wgrad_frag(AT.data_ptr(), BT.data_ptr(), dW.data_ptr<float>(),
+                      (int32_t)N, (int32_t)K, mchunks, (int32_t)nt_w,
+                      (int32_t)kt_w, current_stream());
+  }
+  return dW;
+}
+
 }  // namespace rsdl
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -660,6 +689,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad_bf16", &rsdl::wgrad_bf16, py::arg("dy"), py::arg("x"),
         py::arg("with_bias") = true);
   m.def("relu_bwd_bias", &rsdl::relu_bwd_bias, py::arg("dy"), py::arg("y"));
+  m.def("wgrad_frag_bf16", &rsdl::wgrad_frag_bf16, py::arg("AT"),
+        py::arg("BT"), py::arg("N"), py::arg("K"), py::arg("mchunks"),
+        py::arg("nt_w"), py::arg("kt_w"));
   m.def("fwd_chain_bf16", &rsdl::fwd_chain_bf16, py::arg("x"),
         py::arg("W1"), py::arg("b1"), py::arg("W2"), py::arg("b2"),
         py::arg("W3"), py::arg("b3"), py::arg("w4"), py::arg("b4"),

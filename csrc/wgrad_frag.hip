@@ -1,0 +1,159 @@
+// Fragment-major weight-gradient kernel for MI355X (gfx950).
+//
+//   dW[n][k] = sum_m dz[m][n] * src[m][k]
+//
+// Both inputs arrive TRANSPOSED in the fragment-major layout the MFMA
+// wants (same block structure as the chain kernels' swizzled weights,
+// csrc/fwd_chain.hip, but over the M contraction):
+//
+//   AT = dz^T  fragments: [N/32][Mp/16][2][32][8]   (Mp = M padded to 16)
+//   BT = src^T fragments: [K/32][Mp/16][2][32][8]
+//
+// so every wave's fragment load is a contiguous 1 KB block — the round-1
+// wgrad kernel (csrc/wgrad_kernel.hip) read m-strided fragments and was
+// bound on L1/TCC line processing at ~1.3-1.6 TB/s; the library split-K
+// bmm ran ~3x off the stream roofline. The producers emit these layouts
+// for free: bwd_chain writes dz^T straight from its epilogue registers
+// and fwd_chain/swizzle_x emit src^T (round-2 stage 2/3).
+//
+// Structure: NO LDS, no barriers — per m-chunk (16 rows) each wave loads
+// its (NT_W + KT_W) fragments directly from global (coalesced; waves of
+// one workgroup share B blocks -> L1 hits) and issues NT_W*KT_W MFMAs,
+// double-buffered. The workgroup owns an output block of
+// [128*NT_W n x 32*KT_W k] and one m-slab; fp32 partials land in dW with
+// unsafeAtomicAdd (coalesced 32x4B rows).
+//
+// XCD-aware decode: consecutive blockIdx round-robin across the 8 XCDs,
+// so the grid is decoded as (xcd, seq) with all output-blocks of one
+// m-slab given to ONE xcd — their shared input slab stays in that XCD's
+// L2 instead of being refetched from HBM per block.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <cstdint>
+
+namespace rsdl {
+
+typedef __attribute__((__vector_size__(8 * sizeof(short)))) short wf_bf16x8;
+typedef __attribute__((__vector_size__(16 * sizeof(float)))) float wf_f32x16;
+
+template <int NT_W, int KT_W>
+__global__ void __launch_bounds__(256, 2) wgrad_frag_kernel(
+    const short* __restrict__ AT, const short* __restrict__ BT,
+    float* __restrict__ dW,  // [N,K] fp32, pre-zeroed
+    int32_t N, int32_t K, int64_t mchunks, int32_t nblk_n, int32_t nblk_k,
+    int64_t nslabs, int64_t chunks_per_slab) {
+  const int32_t NBLK = nblk_n * nblk_k;
+  const int32_t xcd = blockIdx.x & 7;
+  const int32_t g = blockIdx.x >> 3;
+  const int32_t slab_local = g / NBLK;
+  const int32_t blk = g % NBLK;
+  const int64_t slab = (int64_t)slab_local * 8 + xcd;
+  if (slab >= nslabs) return;
+  const int32_t bn = blk / nblk_k;
+  const int32_t bk = blk % nblk_k;
+  const int64_t c0 = slab * chunks_per_slab;
+  const int64_t cend = min(c0 + chunks_per_slab, mchunks);
+  if (c0 >= cend) return;
+  const int64_t iters = cend - c0;
+
+  const int32_t wave = threadIdx.x >> 6;
+  const int32_t lane = threadIdx.x & 63;
+  const int32_t nt0 = bn * (4 * NT_W) + wave * NT_W;
+  const int32_t kt0 = bk * KT_W;
+
+  const short* aptr[NT_W];
+  const short* bptr[KT_W];
+  #pragma unroll
+  for (int nt = 0; nt < NT_W; nt++) {
+    aptr[nt] = AT + ((int64_t)(nt0 + nt) * mchunks + c0) * 512 + lane * 8;
+  }
+  #pragma unroll
+  for (int kt = 0; kt < KT_W; kt++) {
+    bptr[kt] = BT + ((int64_t)(kt0 + kt) * mchunks + c0) * 512 + lane * 8;
+  }
+
+  wf_f32x16 acc[NT_W][KT_W] = {};
+  wf_bf16x8 a0[NT_W], b0[KT_W], a1[NT_W], b1[KT_W];
+
+#define WF_LOAD(abuf, bbuf, i)                                             \
+  {                                                                        \
+    _Pragma("unroll") for (int nt = 0; nt < NT_W; nt++) {                  \
+      *reinterpret_cast<uint4*>(&abuf[nt]) =                               \
+          *reinterpret_cast<const uint4*>(&aptr[nt][(i)*512]);             \
+    }                                                                      \
+    _Pragma("unroll") for (int kt = 0; kt < KT_W; kt++) {                  \
+      *reinterpret_cast<uint4*>(&bbuf[kt]) =                               \
+          *reinterpret_cast<const uint4*>(&bptr[kt][(i)*512]);             \
+    }                                                                      \
+  }
+#define WF_MFMA(abuf, bbuf)                                                \
+  {                                                                        \
+    _Pragma("unroll") for (int nt = 0; nt < NT_W; nt++) {                  \
+      _Pragma("unroll") for (int kt = 0; kt < KT_W; kt++) {                \
+        acc[nt][kt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(             \
+            abuf[nt], bbuf[kt], acc[nt][kt], 0, 0, 0);                     \
+      }                                                                    \
+    }                                                                      \
+  }
+
+  WF_LOAD(a0, b0, 0);
+  int64_t i = 0;
+  while (i + 2 <= iters) {
+    WF_LOAD(a1, b1, i + 1);
+    WF_MFMA(a0, b0);
+    if (i + 2 < iters) WF_LOAD(a0, b0, i + 2);
+    WF_MFMA(a1, b1);
+    i += 2;
+  }
+  if (i < iters) WF_MFMA(a0, b0);
+#undef WF_LOAD
+#undef WF_MFMA
+
+  // Epilogue: D[row = n-in-tile][col = k-in-tile]; 32 lanes write 32
+  // consecutive k's -> coalesced 128-B atomic rows.
+  #pragma unroll
+  for (int nt = 0; nt < NT_W; nt++) {
+    #pragma unroll
+    for (int kt = 0; kt < KT_W; kt++) {
+      const int32_t k = (kt0 + kt) * 32 + (lane & 31);
+      #pragma unroll
+      for (int reg = 0; reg < 16; reg++) {
+        const int32_t n = (nt0 + nt) * 32 + (reg & 3) + 8 * (reg >> 2) +
+                          4 * (lane >> 5);
+        unsafeAtomicAdd(&dW[(int64_t)n * K + k], acc[nt][kt][reg]);
+      }
+    }
+  }
+}
+
+// layer configs: 1 -> NT_W=2,KT_W=4 (dW1 [512,128pad]); 2 -> 1,8
+// (dW2 [256,512]); 3 -> 1,8 (dW3 [128,256]).
+void launch_wgrad_frag(const void* AT, const void* BT, float* dW, int32_t N,
+                       int32_t K, int64_t mchunks, int32_t nt_w,
+                       int32_t kt_w, hipStream_t stream) {
+  const int32_t nblk_n = N / (nt_w * 128);
+  const int32_t nblk_k = K / (kt_w * 32);
+  const int32_t nblk = nblk_n * nblk_k;
+  // target ~2048 workgroups for 256 CUs at 2-3 WGs each
+  int64_t nslabs = (2048 + nblk - 1) / nblk;
+  if (nslabs > mchunks) nslabs = mchunks;
+  const int64_t chunks_per_slab = (mchunks + nslabs - 1) / nslabs;
+  const int64_t grid = ((nslabs + 7) / 8) * 8 * nblk;
+  if (nt_w == 2 && kt_w == 4) {
+    hipLaunchKernelGGL((wgrad_frag_kernel<2, 4>), dim3((uint32_t)grid),
+                       dim3(256), 0, stream,
+                       reinterpret_cast<const short*>(AT),
+                       reinterpret_cast<const short*>(BT), dW, N, K, mchunks,
+                       nblk_n, nblk_k, nslabs, chunks_per_slab);
+  } else if (nt_w == 1 && kt_w == 8) {
+    hipLaunchKernelGGL((wgrad_frag_kernel<1, 8>), dim3((uint32_t)grid),
+                       dim3(256), 0, stream,
+                       reinterpret_cast<const short*>(AT),
+                       reinterpret_cast<const short*>(BT), dW, N, K, mchunks,
+                       nblk_n, nblk_k, nslabs, chunks_per_slab);
+  }
+}
+
+}  // namespace rsdl

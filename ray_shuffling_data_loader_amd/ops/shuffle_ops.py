@@ -253,3 +253,33 @@ def wgrad(dy: torch.Tensor, x: torch.Tensor, with_bias: bool = True):
     dw = dy.t().float() @ x.float()
     db = dy.float().sum(0) if with_bias else None
     return dw, db
+
+
+def t_frag_swizzle(t: torch.Tensor) -> torch.Tensor:
+    """[M, C] -> fragment-major layout of t^T: flat [C/32][Mp/16][2][32][8]
+    (Mp = M padded to a 16-multiple with zero rows). This is the input
+    layout of the fragment-major wgrad kernel (csrc/wgrad_frag.hip); the
+    hot producers (bwd_chain dz^T, fwd_chain a^T) emit it directly — this
+    torch implementation is the oracle/fallback."""
+    m, c = t.shape
+    mp = (m + 15) // 16 * 16
+    if mp != m:
+        t = torch.nn.functional.pad(t, (0, 0, 0, mp - m))
+    v = t.reshape(mp // 16, 2, 8, c // 32, 32)
+    return v.permute(3, 0, 1, 4, 2).contiguous().reshape(-1)
+
+
+_WGRAD_FRAG_CFG = {
+    # (N, K) -> (nt_w, kt_w); see csrc/wgrad_frag.hip launch configs
+    (512, 128): (2, 4),
+    (256, 512): (1, 8),
+    (128, 256): (1, 8),
+}
+
+
+def wgrad_frag(at_frag: torch.Tensor, bt_frag: torch.Tensor, n: int,
+               k: int, mchunks: int) -> torch.Tensor:
+    """dW = dz^T @ src from pre-swizzled fragment inputs (fp32 [N,K])."""
+    nt_w, kt_w = _WGRAD_FRAG_CFG[(n, k)]
+    hip = _load_hip()
+    return hip.wgrad_frag_bf16(at_frag, bt_frag, n, k, mchunks, nt_w, kt_w)

@@ -45,6 +45,7 @@ try:
         "csrc/fwd_chain.hip",
         "csrc/bwd_chain.hip",
         "csrc/wgrad_wide.hip",
+        "csrc/wgrad_frag.hip",
     ]
     if all(os.path.exists(s) for s in hip_sources):
         ext_modules.append(
