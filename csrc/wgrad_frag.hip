@@ -32,6 +32,7 @@
 #include <hip/hip_bf16.h>
 
 #include <cstdint>
+#include <cstdlib>
 
 namespace rsdl {
 
@@ -136,8 +137,13 @@ void launch_wgrad_frag(const void* AT, const void* BT, float* dW, int32_t N,
   const int32_t nblk_n = N / (nt_w * 128);
   const int32_t nblk_k = K / (kt_w * 32);
   const int32_t nblk = nblk_n * nblk_k;
-  // target ~2048 workgroups for 256 CUs at 2-3 WGs each
-  int64_t nslabs = (2048 + nblk - 1) / nblk;
+  // target workgroup count: more slabs = more parallelism but more
+  // atomic writers per output line (RSDL_WGRAD_WGS to tune).
+  static int64_t target = [] {
+    const char* e = std::getenv("RSDL_WGRAD_WGS");
+    return e ? atoll(e) : 2048;
+  }();
+  int64_t nslabs = (target + nblk - 1) / nblk;
   if (nslabs > mchunks) nslabs = mchunks;
   const int64_t chunks_per_slab = (mchunks + nslabs - 1) / nslabs;
   const int64_t grid = ((nslabs + 7) / 8) * 8 * nblk;
