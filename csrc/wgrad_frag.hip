@@ -77,6 +77,7 @@ __global__ void __launch_bounds__(256, 2) wgrad_frag_kernel(
 
   wf_f32x16 acc[NT_W][KT_W] = {};
   wf_bf16x8 a0[NT_W], b0[KT_W], a1[NT_W], b1[KT_W];
+  wf_bf16x8 a2[NT_W], b2[KT_W];
 
 #define WF_LOAD(abuf, bbuf, i)                                             \
   {                                                                        \
@@ -99,16 +100,23 @@ __global__ void __launch_bounds__(256, 2) wgrad_frag_kernel(
     }                                                                      \
   }
 
+  // 3-deep prefetch: at depth 1 only ~6 KB/wave was in flight and the
+  // kernel ran latency-bound at ~half the stream roofline (depth 4
+  // spills the <1,8> config's registers).
   WF_LOAD(a0, b0, 0);
+  if (1 < iters) WF_LOAD(a1, b1, 1);
   int64_t i = 0;
-  while (i + 2 <= iters) {
-    WF_LOAD(a1, b1, i + 1);
+  while (i + 3 <= iters) {
+    WF_LOAD(a2, b2, i + 2);
     WF_MFMA(a0, b0);
-    if (i + 2 < iters) WF_LOAD(a0, b0, i + 2);
+    if (i + 3 < iters) WF_LOAD(a0, b0, i + 3);
     WF_MFMA(a1, b1);
-    i += 2;
+    if (i + 4 < iters) WF_LOAD(a1, b1, i + 4);
+    WF_MFMA(a2, b2);
+    i += 3;
   }
   if (i < iters) WF_MFMA(a0, b0);
+  if (i + 1 < iters) WF_MFMA(a1, b1);
 #undef WF_LOAD
 #undef WF_MFMA
 
