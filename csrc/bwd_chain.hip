@@ -92,48 +92,44 @@ __device__ void bc_store_tile(const short* __restrict__ lds, short* out,
   }
 }
 
-// One dgrad layer: da = dz_src @ W (via WT [N,K] contiguous), then
-// dz_dst = da * (a_dst > 0) written IN PLACE over the a_dst tile.
-// K = dz_src width (contraction), N = output width. The layer's bias
-// partials db[n] = sum_m dz[m,n] are folded into the epilogue (the
-// masked fp32 values are already in registers; lanes l/l+32 share a
-// column, one shfl_xor combines them) — the old per-column scalar-LDS
-// reduction pass (~112 LDS reads/thread) is gone.
-template <int K, int N, int SRC_S, int DST_S>
+// One dgrad layer: da = dz_src @ W (via swizzled WT fragments), then
+// dz_dst = da * relu_mask. The mask arrives as one 32-bit word per
+// column (maskT[m_tile][n], produced by the forward epilogue) — the
+// backward never reads activation VALUES. dz leaves two ways:
+//   * WRITE_LDS: into the dst LDS tile (stride DST_S) when the next
+//     dgrad layer consumes it as its A source (dz2);
+//   * always: TRANSPOSED wgrad-fragment-major runs straight from
+//     registers to global (same shfl_xor half-row exchange as the
+//     forward emitter) — what the wgrad kernel reads, coalesced.
+// Bias partials db[n] fold into the epilogue as before.
+template <int K, int N, int SRC_S, int DST_S, bool WRITE_LDS>
 __device__ void bc_layer(const short* __restrict__ dz_src,
                          const short* __restrict__ WT,
-                         short* __restrict__ a_dst,
-                         float* __restrict__ db_out, int32_t wave,
-                         int32_t lane) {
+                         const uint32_t* __restrict__ mask_row,
+                         short* __restrict__ dst_lds,
+                         short* __restrict__ dzt_out,
+                         float* __restrict__ db_out, int64_t mchunks,
+                         int64_t mc0, int32_t wave, int32_t lane) {
   constexpr int NT = N / 128;
   constexpr int ITERS = K / 16;
+  static_assert(BC_MTILES == 1, "transposed emission assumes 32-row slabs");
   const int32_t n_base = wave * (N / 4);
   const int32_t frag_k0 = (lane >> 5) * 8;
   const int32_t ml = lane & 31;
+  const int32_t h = lane >> 5;
 
-  // Software-pipelined k-loop (same rationale as fc_layer: 1 wave/SIMD —
-  // the next fragments must be in flight during the current MFMAs).
-  const short* srcA[BC_MTILES];
-  #pragma unroll
-  for (int mt = 0; mt < BC_MTILES; mt++) {
-    srcA[mt] = &dz_src[(mt * 32 + ml) * SRC_S + frag_k0];
-  }
-  // Fragment-major swizzled weights (see fwd_chain.hip): block
-  // (ntile, kc) = 512 contiguous halfwords, lane slice at lane*8.
+  const short* srcA = &dz_src[ml * SRC_S + frag_k0];
   const short* srcB[NT];
   #pragma unroll
   for (int nt = 0; nt < NT; nt++) {
     srcB[nt] = &WT[((int64_t)(wave * NT + nt) * ITERS) * 512 + lane * 8];
   }
 
-  bc_f32x16 acc[BC_MTILES][NT] = {};
-  bc_bf16x8 a[2][BC_MTILES], b[2][NT];
+  bc_f32x16 acc[NT] = {};
+  bc_bf16x8 a[2], b[2][NT];
 
-  #pragma unroll
-  for (int mt = 0; mt < BC_MTILES; mt++) {
-    *reinterpret_cast<uint4*>(&a[0][mt]) =
-        *reinterpret_cast<const uint4*>(srcA[mt]);
-  }
+  *reinterpret_cast<uint4*>(&a[0]) =
+      *reinterpret_cast<const uint4*>(srcA);
   #pragma unroll
   for (int nt = 0; nt < NT; nt++) {
     *reinterpret_cast<uint4*>(&b[0][nt]) =
@@ -144,12 +140,8 @@ __device__ void bc_layer(const short* __restrict__ dz_src,
     const int cur = i & 1;
     const int nxt = cur ^ 1;
     if (i + 1 < ITERS) {
-      const int32_t k = (i + 1) * 16;
-      #pragma unroll
-      for (int mt = 0; mt < BC_MTILES; mt++) {
-        *reinterpret_cast<uint4*>(&a[nxt][mt]) =
-            *reinterpret_cast<const uint4*>(&srcA[mt][k]);
-      }
+      *reinterpret_cast<uint4*>(&a[nxt]) =
+          *reinterpret_cast<const uint4*>(&srcA[(i + 1) * 16]);
       #pragma unroll
       for (int nt = 0; nt < NT; nt++) {
         *reinterpret_cast<uint4*>(&b[nxt][nt]) =
@@ -157,77 +149,87 @@ __device__ void bc_layer(const short* __restrict__ dz_src,
       }
     }
     #pragma unroll
-    for (int mt = 0; mt < BC_MTILES; mt++) {
-      #pragma unroll
-      for (int nt = 0; nt < NT; nt++) {
-        acc[mt][nt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            a[cur][mt], b[cur][nt], acc[mt][nt], 0, 0, 0);
-      }
+    for (int nt = 0; nt < NT; nt++) {
+      acc[nt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          a[cur], b[cur][nt], acc[nt], 0, 0, 0);
     }
   }
-  // No barrier needed between the k-loop and the epilogue: each wave
-  // reads only dz_src (a different buffer) and masks/writes only its own
-  // n-range of a_dst; inter-layer ordering is handled by the
-  // __syncthreads() between bc_layer calls in the kernel body.
+
   #pragma unroll
   for (int nt = 0; nt < NT; nt++) {
     const int32_t n = n_base + nt * 32 + ml;
+    const uint32_t mw = mask_row[n];
+    float vals[16];
     float colsum = 0.f;
     #pragma unroll
-    for (int mt = 0; mt < BC_MTILES; mt++) {
-      #pragma unroll
-      for (int reg = 0; reg < 16; reg++) {
-        const int32_t mrow =
-            mt * 32 + (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
-        short* cell = &a_dst[mrow * DST_S + n];
-        const float live = bc_b2f(*cell) > 0.f ? 1.f : 0.f;
-        const float v = acc[mt][nt][reg] * live;
-        *cell = bc_f2b(v);
-        colsum += v;
-      }
+    for (int reg = 0; reg < 16; reg++) {
+      const int32_t mrow = (reg & 3) + 8 * (reg >> 2) + 4 * h;
+      const float v = ((mw >> mrow) & 1u) ? acc[nt][reg] : 0.f;
+      vals[reg] = v;
+      colsum += v;
+      if (WRITE_LDS) dst_lds[mrow * DST_S + n] = bc_f2b(v);
     }
     colsum += __shfl_xor(colsum, 32);
     if (lane < 32) db_out[n] = colsum;
+    // Transposed fragment emission (see fwd_chain fc_layer EMIT_T).
+    float ex0[4], ex1[4];
+    #pragma unroll
+    for (int j = 0; j < 4; j++) {
+      ex0[j] = __shfl_xor(h == 0 ? vals[4 + j] : vals[j], 32);
+      ex1[j] = __shfl_xor(h == 0 ? vals[12 + j] : vals[8 + j], 32);
+    }
+    short run0[8], run1[8];
+    #pragma unroll
+    for (int j = 0; j < 4; j++) {
+      run0[j] = bc_f2b(h == 0 ? vals[j] : ex0[j]);
+      run0[4 + j] = bc_f2b(h == 0 ? ex0[j] : vals[4 + j]);
+      run1[j] = bc_f2b(h == 0 ? vals[8 + j] : ex1[j]);
+      run1[4 + j] = bc_f2b(h == 0 ? ex1[j] : vals[12 + j]);
+    }
+    const int64_t nt_g = (int64_t)(n_base + nt * 32) >> 5;
+    short* blk0 = dzt_out + ((nt_g * mchunks + mc0) * 512) + h * 256 +
+                  ml * 8;
+    short* blk1 = dzt_out + ((nt_g * mchunks + mc0 + 1) * 512) + h * 256 +
+                  ml * 8;
+    __builtin_nontemporal_store(*reinterpret_cast<bc_u32x4*>(run0),
+                                reinterpret_cast<bc_u32x4*>(blk0));
+    __builtin_nontemporal_store(*reinterpret_cast<bc_u32x4*>(run1),
+                                reinterpret_cast<bc_u32x4*>(blk1));
   }
 }
 
 __global__ void __launch_bounds__(256) bwd_chain_kernel(
-    const short* __restrict__ dy,    // [M,1] bf16 (head grad)
-    const short* __restrict__ a1,    // [M,512] saved activations
-    const short* __restrict__ a2,    // [M,256]
-    const short* __restrict__ a3,    // [M,128]
-    const short* __restrict__ w4,    // [128]
-    const short* __restrict__ W3T,   // swizzled W3^T fragments
-    const short* __restrict__ W2T,   // swizzled W2^T fragments
-    short* __restrict__ dz1, short* __restrict__ dz2,
-    short* __restrict__ dz3,
+    const short* __restrict__ dy,     // [M,1] bf16 (head grad)
+    const short* __restrict__ a3,     // [M,128] saved activations
+    const uint32_t* __restrict__ mask1,  // [m_tiles][512] relu-mask words
+    const uint32_t* __restrict__ mask2,  // [m_tiles][256]
+    const short* __restrict__ w4,     // [128]
+    const short* __restrict__ W3T,    // swizzled W3^T fragments
+    const short* __restrict__ W2T,    // swizzled W2^T fragments
+    short* __restrict__ dz1t,         // wgrad fragment-major dz^T outputs
+    short* __restrict__ dz2t, short* __restrict__ dz3t,
     // [grid][512+256+128+1+256]: db1|db2|db3|db4|dW4 partials. dW4[k] =
     // sum_m dy[m]*a3[m,k] folded into the dz3 seed loop (a3 is already
     // in LDS there); the two per-k thread partials (even/odd m) are
     // summed host-side.
     float* __restrict__ db_part,
-    int64_t M) {
-  // LDS is the occupancy lever: t3 (dz3) ALIASES t1's storage — a1 is
-  // loaded only after dz3 has been stored to global, so t1+t2+epsilon
-  // = ~50 KB -> 3 workgroups/CU (the monolithic t1+t2+t3 layout was
-  // 59 KB -> 2).
-  __shared__ __align__(16) char smem[BC_MT * BC_S1 * 2 + BC_MT * BC_S2 * 2 +
-                                     BC_MT * 4 + 256 * 4];
-  short* t1 = reinterpret_cast<short*>(smem);
-  short* t3 = t1;  // aliased: live ranges disjoint (barrier-ordered)
-  short* t2 = reinterpret_cast<short*>(smem + BC_MT * BC_S1 * 2);
-  float* dyf = reinterpret_cast<float*>(smem + BC_MT * BC_S1 * 2 +
-                                        BC_MT * BC_S2 * 2);
-  float* s3sh = dyf + BC_MT;
+    int64_t M, int64_t mchunks) {
+  // The backward never reads a1/a2 (relu masks arrive as bit-words from
+  // the forward) and dz1 never touches LDS (emitted transposed straight
+  // from registers), so LDS is just t2 + t3 + partial scratch ~27 KB.
+  __shared__ __align__(16) short t2[BC_MT * BC_S2];
+  __shared__ __align__(16) short t3[BC_MT * BC_S3];
+  __shared__ float dyf[BC_MT];
+  __shared__ float s3sh[256];
 
   const int64_t m0 = (int64_t)blockIdx.x * BC_MT;
   const int32_t tid = threadIdx.x;
   const int32_t wave = tid >> 6;
   const int32_t lane = tid & 63;
+  const int64_t mc0 = (int64_t)blockIdx.x * 2;
   float* part =
       &db_part[(int64_t)blockIdx.x * (BC_N1 + BC_N2 + BC_N3 + 1 + 256)];
 
-  bc_load_tile<BC_N2, BC_S2>(a2, t2, m0, M, tid);
   bc_load_tile<BC_N3, BC_S3>(a3, t3, m0, M, tid);
   if (tid < BC_MT) {
     dyf[tid] = (m0 + tid < M) ? bc_b2f(dy[m0 + tid]) : 0.f;
@@ -262,42 +264,58 @@ __global__ void __launch_bounds__(256) bwd_chain_kernel(
   if (tid < BC_N3) {
     part[BC_N1 + BC_N2 + tid] = s3sh[tid] + s3sh[tid + BC_N3];
   }
+  // dz3^T emission: one 8-row run per work item read out of t3 (the
+  // wgrad kernel's A layout; 512 runs over 256 threads).
+  for (int32_t r = tid; r < (BC_N3 / 32) * 2 * 2 * 32; r += 256) {
+    const int32_t mlr = r & 31;
+    const int32_t hr = (r >> 5) & 1;
+    const int32_t mcl = (r >> 6) & 1;
+    const int32_t ntr = r >> 7;
+    const int32_t k = ntr * 32 + mlr;
+    const int32_t mbase = mcl * 16 + hr * 8;
+    short run[8];
+    #pragma unroll
+    for (int j = 0; j < 8; j++) {
+      run[j] = t3[(mbase + j) * BC_S3 + k];
+    }
+    short* blk = dz3t + (((int64_t)ntr * mchunks + mc0 + mcl) * 512) +
+                 hr * 256 + mlr * 8;
+    __builtin_nontemporal_store(*reinterpret_cast<bc_u32x4*>(run),
+                                reinterpret_cast<bc_u32x4*>(blk));
+  }
+  __syncthreads();
 
-  // da2 = dz3 @ W3 (via W3T), mask by a2 -> dz2 in place over t2;
-  // db2 partials written by the epilogue.
-  bc_layer<BC_N3, BC_N2, BC_S3, BC_S2>(t3, W3T, t2, &part[BC_N1], wave,
-                                       lane);
+  // da2 = dz3 @ W3 (via W3T), masked -> dz2: LDS tile (next layer's A
+  // source) + transposed emission + db2 partials.
+  bc_layer<BC_N3, BC_N2, BC_S3, BC_S2, true>(
+      t3, W3T, &mask2[(int64_t)blockIdx.x * BC_N2], t2, dz2t,
+      &part[BC_N1], mchunks, mc0, wave, lane);
   __syncthreads();
-  bc_store_tile<BC_N3, BC_S3>(t3, dz3, m0, M, tid);
-  bc_store_tile<BC_N2, BC_S2>(t2, dz2, m0, M, tid);
-  __syncthreads();  // t3 fully read before a1 overwrites its storage
-  bc_load_tile<BC_N1, BC_S1>(a1, t1, m0, M, tid);
-  __syncthreads();
-  // da1 = dz2 @ W2 (via W2T), mask by a1 -> dz1 in place over t1; db1.
-  bc_layer<BC_N2, BC_N1, BC_S2, BC_S1>(t2, W2T, t1, &part[0], wave,
-                                       lane);
-  __syncthreads();
-  bc_store_tile<BC_N1, BC_S1>(t1, dz1, m0, M, tid);
+  // da1 = dz2 @ W2 (via W2T), masked -> dz1: transposed emission only.
+  bc_layer<BC_N2, BC_N1, BC_S2, BC_S1, false>(
+      t2, W2T, &mask1[(int64_t)blockIdx.x * BC_N1], nullptr, dz1t,
+      &part[0], mchunks, mc0, wave, lane);
 }
 
 int64_t bwd_chain_grid(int64_t M) { return (M + BC_MT - 1) / BC_MT; }
 
-void launch_bwd_chain(const void* dy, const void* a1, const void* a2,
-                      const void* a3, const void* w4, const void* W3T,
-                      const void* W2T, void* dz1, void* dz2, void* dz3,
+void launch_bwd_chain(const void* dy, const void* a3, const void* mask1,
+                      const void* mask2, const void* w4, const void* W3T,
+                      const void* W2T, void* dz1t, void* dz2t, void* dz3t,
                       float* db_part, int64_t M, hipStream_t stream) {
   const int32_t grid = (int32_t)((M + BC_MT - 1) / BC_MT);
+  const int64_t mchunks = (int64_t)grid * 2;
   hipLaunchKernelGGL(bwd_chain_kernel, dim3(grid), dim3(256), 0, stream,
                      reinterpret_cast<const short*>(dy),
-                     reinterpret_cast<const short*>(a1),
-                     reinterpret_cast<const short*>(a2),
                      reinterpret_cast<const short*>(a3),
+                     reinterpret_cast<const uint32_t*>(mask1),
+                     reinterpret_cast<const uint32_t*>(mask2),
                      reinterpret_cast<const short*>(w4),
                      reinterpret_cast<const short*>(W3T),
                      reinterpret_cast<const short*>(W2T),
-                     reinterpret_cast<short*>(dz1),
-                     reinterpret_cast<short*>(dz2),
-                     reinterpret_cast<short*>(dz3), db_part, M);
+                     reinterpret_cast<short*>(dz1t),
+                     reinterpret_cast<short*>(dz2t),
+                     reinterpret_cast<short*>(dz3t), db_part, M, mchunks);
 }
 
 }  // namespace rsdl

@@ -86,13 +86,23 @@ __device__ __forceinline__ short fc_f2b(float f) {
 // the swizzled weights: [kc][h][ml][8], 512 halfwords per k-chunk) —
 // used for layer 1, whose input x is pre-swizzled host-side; SRC_S is
 // ignored. Otherwise src is the LDS tile of the previous layer.
+// EMIT_T: besides the LDS tile, the epilogue emits this m-tile's
+// activations TRANSPOSED in wgrad fragment-major layout
+// ([N/32][mchunks][2][32][8]; see csrc/wgrad_frag.hip) plus one 32-bit
+// relu-mask word per column (maskT[m_tile][n], bit i = row i alive) —
+// the backward chain then never reads the activations at all (masks
+// only) and the wgrad kernel gets coalesced B fragments for free.
+// Lanes l/l+32 hold complementary 4-row runs of a column; one shfl_xor
+// per 4 values assembles the 8-row fragment runs in registers.
 template <int K, int N, int SRC_S, int DST_S, bool RELU,
-          bool A_FRAGMAJOR = false>
+          bool A_FRAGMAJOR = false, bool EMIT_T = false>
 __device__ void fc_layer(const short* __restrict__ src_lds,
                          const short* __restrict__ W,
                          const float* __restrict__ bias,
                          short* __restrict__ dst_lds, int32_t wave,
-                         int32_t lane) {
+                         int32_t lane, short* __restrict__ at_out = nullptr,
+                         uint32_t* __restrict__ mask_row = nullptr,
+                         int64_t mchunks = 0, int64_t mc0 = 0) {
   constexpr int NT = N / 128;  // n-tiles of 32 per wave (4 waves)
   constexpr int ITERS = K / 16;
   const int32_t n_base = wave * (N / 4);
@@ -161,20 +171,62 @@ __device__ void fc_layer(const short* __restrict__ src_lds,
     }
   }
   // Epilogue: bias + (relu) + cast + store the D fragments into the dst
-  // tile at [mrow][n] (b16 column writes; DST_S padding spreads banks).
+  // tile at [mrow][n] (b16 column writes; DST_S padding spreads banks);
+  // with EMIT_T also assemble the transposed fragment runs + mask word.
+  static_assert(!EMIT_T || FC_MTILES == 1, "EMIT_T assumes 32-row slabs");
+  const int32_t h = lane >> 5;
   #pragma unroll
   for (int mt = 0; mt < FC_MTILES; mt++) {
     #pragma unroll
     for (int nt = 0; nt < NT; nt++) {
       const int32_t n = n_base + nt * 32 + ml;
       const float bv = bias[n];
+      float vals[16];
       #pragma unroll
       for (int reg = 0; reg < 16; reg++) {
         const int32_t mrow =
-            mt * 32 + (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+            mt * 32 + (reg & 3) + 8 * (reg >> 2) + 4 * h;
         float v = acc[mt][nt][reg] + bv;
         if (RELU) v = v > 0.f ? v : 0.f;
         dst_lds[mrow * DST_S + n] = fc_f2b(v);
+        vals[reg] = v;
+      }
+      if (EMIT_T) {
+        // Exchange the half-rows: lane h=0 assembles rows {0-7},{16-23},
+        // h=1 assembles {8-15},{24-31}.
+        float ex0[4], ex1[4];
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+          ex0[j] = __shfl_xor(h == 0 ? vals[4 + j] : vals[j], 32);
+          ex1[j] = __shfl_xor(h == 0 ? vals[12 + j] : vals[8 + j], 32);
+        }
+        short run0[8], run1[8];
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+          run0[j] = fc_f2b(h == 0 ? vals[j] : ex0[j]);
+          run0[4 + j] = fc_f2b(h == 0 ? ex0[j] : vals[4 + j]);
+          run1[j] = fc_f2b(h == 0 ? vals[8 + j] : ex1[j]);
+          run1[4 + j] = fc_f2b(h == 0 ? ex1[j] : vals[12 + j]);
+        }
+        // run0 -> (mc_local 0, h'=h); run1 -> (mc_local 1, h'=h)
+        const int64_t nt_g = (int64_t)(n_base + nt * 32) >> 5;
+        short* blk0 = at_out + ((nt_g * mchunks + mc0) * 512) + h * 256 +
+                      ml * 8;
+        short* blk1 = at_out + ((nt_g * mchunks + mc0 + 1) * 512) +
+                      h * 256 + ml * 8;
+        __builtin_nontemporal_store(*reinterpret_cast<fc_u32x4*>(run0),
+                                    reinterpret_cast<fc_u32x4*>(blk0));
+        __builtin_nontemporal_store(*reinterpret_cast<fc_u32x4*>(run1),
+                                    reinterpret_cast<fc_u32x4*>(blk1));
+        // Mask word for column n (bit i = bf16 value of row i > 0).
+        uint32_t w = 0;
+        #pragma unroll
+        for (int reg = 0; reg < 16; reg++) {
+          const int32_t mrow = (reg & 3) + 8 * (reg >> 2) + 4 * h;
+          w |= (uint32_t)(fc_f2b(vals[reg]) > 0) << mrow;
+        }
+        w |= __shfl_xor(w, 32);
+        if (h == 0) mask_row[n] = w;
       }
     }
   }
@@ -206,19 +258,24 @@ __global__ void __launch_bounds__(256) fwd_chain_kernel(
     const short* __restrict__ W2, const float* __restrict__ b2,
     const short* __restrict__ W3, const float* __restrict__ b3,
     const short* __restrict__ w4, const float* __restrict__ b4,
-    short* __restrict__ a1, short* __restrict__ a2,
+    short* __restrict__ a1t,        // wgrad fragment-major a1^T
+    uint32_t* __restrict__ mask1,   // [m_tiles][512] relu-mask words
+    short* __restrict__ a2t,        // wgrad fragment-major a2^T
+    uint32_t* __restrict__ mask2,   // [m_tiles][256]
     short* __restrict__ a3, short* __restrict__ out,
     // Optional fused MSE epilogue (target != nullptr): dyb[m] =
     // (2/M)*(out[m]-target[m]) in bf16 and loss_part[blockIdx] =
     // sum_m (out[m]-target[m])^2 — removes the eager loss/grad kernel
     // chain from the fused train step.
     const float* __restrict__ target, short* __restrict__ dyb,
-    float* __restrict__ loss_part, float inv_m, int64_t M) {
+    float* __restrict__ loss_part, float inv_m, int64_t M,
+    int64_t mchunks) {
   // LDS budget is the occupancy lever: t1 + t2 + lsum = ~50 KB -> 3
   // workgroups/CU. x needs no tile (pre-swizzled fragment-major in
-  // global, coalesced A loads), and t3 ALIASES t1's storage — a1 is
-  // stored to global right after layer 2 consumes it, before layer 3's
-  // epilogue writes t3.
+  // global, coalesced A loads), and t3 ALIASES t1's storage — layer 3
+  // runs after layer 2 consumed t1 (a1 itself leaves through the
+  // EMIT_T epilogue, transposed, plus mask words — no row-major a1/a2
+  // tensors exist at all).
   __shared__ __align__(16) char smem[FC_MT * FC_S1 * 2 + FC_MT * FC_S2 * 2 +
                                      FC_MT * 4];
   short* t1 = reinterpret_cast<short*>(smem);
@@ -231,16 +288,17 @@ __global__ void __launch_bounds__(256) fwd_chain_kernel(
   const int32_t tid = threadIdx.x;
   const int32_t wave = tid >> 6;
   const int32_t lane = tid & 63;
+  const int64_t mc0 = (int64_t)blockIdx.x * 2;
 
   // Layer 1: A fragments straight from the swizzled global x block.
   const short* xblk = &x0s[(int64_t)blockIdx.x * (FC_K0P / 16) * 512];
-  fc_layer<FC_K0P, FC_N1, 0, FC_S1, true, true>(xblk, W1, b1, t1, wave,
-                                                lane);
+  fc_layer<FC_K0P, FC_N1, 0, FC_S1, true, true, true>(
+      xblk, W1, b1, t1, wave, lane, a1t, &mask1[(int64_t)blockIdx.x * FC_N1],
+      mchunks, mc0);
   __syncthreads();
-  fc_layer<FC_N1, FC_N2, FC_S1, FC_S2, true>(t1, W2, b2, t2, wave, lane);
-  __syncthreads();
-  // a1 leaves LDS now so layer 3 can reuse t1's storage for t3.
-  fc_store_tile<FC_N1, FC_S1>(t1, a1, m0, M, tid);
+  fc_layer<FC_N1, FC_N2, FC_S1, FC_S2, true, false, true>(
+      t1, W2, b2, t2, wave, lane, a2t, &mask2[(int64_t)blockIdx.x * FC_N2],
+      mchunks, mc0);
   __syncthreads();
   fc_layer<FC_N2, FC_N3, FC_S2, FC_S3, true>(t2, W3, b3, t3, wave, lane);
   __syncthreads();
@@ -287,7 +345,6 @@ __global__ void __launch_bounds__(256) fwd_chain_kernel(
     }
   }
 
-  fc_store_tile<FC_N2, FC_S2>(t2, a2, m0, M, tid);
   fc_store_tile<FC_N3, FC_S3>(t3, a3, m0, M, tid);
 }
 
@@ -339,27 +396,71 @@ void launch_swizzle_x(const void* x, void* out, int64_t M,
                      reinterpret_cast<short*>(out), M, total);
 }
 
+// x [M,100] bf16 -> wgrad fragment-major x^T:
+// [128/32][mchunks][2][32][8] with zero pads (cols 100..127, rows >= M).
+// One thread per 16-B out block = 8 consecutive m's of one column k;
+// writes fully coalesced, reads gather 8 row-strided elements (L1-local:
+// neighboring threads read the same 8 rows).
+__global__ void __launch_bounds__(256) swizzle_xt_kernel(
+    const short* __restrict__ x, short* __restrict__ out, int64_t M,
+    int64_t total_blocks) {
+  const int64_t b = (int64_t)blockIdx.x * 256 + threadIdx.x;
+  if (b >= total_blocks) return;
+  // b = ((kt*mchunks + mc)*2 + h)*32 + ml
+  const int32_t ml = (int32_t)(b & 31);
+  int64_t r = b >> 5;
+  const int32_t h = (int32_t)(r & 1);
+  r >>= 1;
+  const int64_t mchunks = total_blocks >> 9;  // total = 4*mchunks*64
+  const int64_t mc = r % mchunks;
+  const int32_t kt = (int32_t)(r / mchunks);
+  const int32_t k = kt * 32 + ml;
+  const int64_t m0 = mc * 16 + h * 8;
+  short v[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (k < FC_K0) {
+    #pragma unroll
+    for (int j = 0; j < 8; j++) {
+      if (m0 + j < M) v[j] = x[(m0 + j) * FC_K0 + k];
+    }
+  }
+  __builtin_nontemporal_store(*reinterpret_cast<fc_u32x4*>(v),
+                              reinterpret_cast<fc_u32x4*>(&out[b * 8]));
+}
+
+void launch_swizzle_xt(const void* x, void* out, int64_t M,
+                       hipStream_t stream) {
+  const int64_t mtiles = (M + FC_MT - 1) / FC_MT;
+  const int64_t mchunks = mtiles * 2;
+  const int64_t total = 4 * mchunks * 2 * 32;  // kt * mc * h * ml
+  const int64_t grid = (total + 255) / 256;
+  hipLaunchKernelGGL(swizzle_xt_kernel, dim3((uint32_t)grid), dim3(256), 0,
+                     stream, reinterpret_cast<const short*>(x),
+                     reinterpret_cast<short*>(out), M, total);
+}
+
 int64_t fwd_chain_grid(int64_t M) { return (M + FC_MT - 1) / FC_MT; }
 
 void launch_fwd_chain(const void* x0s, const void* W1, const float* b1,
                       const void* W2, const float* b2, const void* W3,
                       const float* b3, const void* w4, const float* b4,
-                      void* a1, void* a2, void* a3, void* out,
+                      void* a1t, uint32_t* mask1, void* a2t,
+                      uint32_t* mask2, void* a3, void* out,
                       const float* target, void* dyb, float* loss_part,
                       int64_t M, hipStream_t stream) {
   const int32_t grid = (int32_t)((M + FC_MT - 1) / FC_MT);
+  const int64_t mchunks = (int64_t)grid * 2;
   hipLaunchKernelGGL(fwd_chain_kernel, dim3(grid), dim3(256), 0, stream,
                      reinterpret_cast<const short*>(x0s),
                      reinterpret_cast<const short*>(W1), b1,
                      reinterpret_cast<const short*>(W2), b2,
                      reinterpret_cast<const short*>(W3), b3,
                      reinterpret_cast<const short*>(w4), b4,
-                     reinterpret_cast<short*>(a1),
-                     reinterpret_cast<short*>(a2),
+                     reinterpret_cast<short*>(a1t), mask1,
+                     reinterpret_cast<short*>(a2t), mask2,
                      reinterpret_cast<short*>(a3),
                      reinterpret_cast<short*>(out), target,
                      reinterpret_cast<short*>(dyb), loss_part,
-                     target ? 1.f / (float)M : 0.f, M);
+                     target ? 1.f / (float)M : 0.f, M, mchunks);
 }
 
 }  // namespace rsdl

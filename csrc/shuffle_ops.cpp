@@ -71,19 +71,22 @@ void launch_relu_bwd_bias(const void* dy, const void* y, void* dx,
 int64_t fwd_chain_grid(int64_t M);
 void launch_swizzle_x(const void* x, void* out, int64_t M,
                       hipStream_t stream);
-void launch_fwd_chain(const void* x0, const void* W1, const float* b1,
+void launch_swizzle_xt(const void* x, void* out, int64_t M,
+                       hipStream_t stream);
+void launch_fwd_chain(const void* x0s, const void* W1, const float* b1,
                       const void* W2, const float* b2, const void* W3,
                       const float* b3, const void* w4, const float* b4,
-                      void* a1, void* a2, void* a3, void* out,
+                      void* a1t, uint32_t* mask1, void* a2t,
+                      uint32_t* mask2, void* a3, void* out,
                       const float* target, void* dyb, float* loss_part,
                       int64_t M, hipStream_t stream);
 int64_t bwd_chain_grid(int64_t M);
 void launch_wgrad_frag(const void* AT, const void* BT, float* dW, int32_t N,
                        int32_t K, int64_t mchunks, int32_t nt_w,
                        int32_t kt_w, hipStream_t stream);
-void launch_bwd_chain(const void* dy, const void* a1, const void* a2,
-                      const void* a3, const void* w4, const void* W3T,
-                      const void* W2T, void* dz1, void* dz2, void* dz3,
+void launch_bwd_chain(const void* dy, const void* a3, const void* mask1,
+                      const void* mask2, const void* w4, const void* W3T,
+                      const void* W2T, void* dz1t, void* dz2t, void* dz3t,
                       float* db_part, int64_t M, hipStream_t stream);
 
 namespace {
@@ -504,9 +507,16 @@ std::vector<at::Tensor> fwd_chain_bf16(
   const int64_t M = x.size(0);
   // The kernel's A-fragment loads need W1's k-dim padded 100 -> 112.
   auto W1p = w1_padded ? W1 : at::constant_pad_nd(W1, {0, 12}).contiguous();
-  // ...and x0's rows padded likewise, staged in LDS by the kernel itself.
-  auto a1 = at::empty({M, 512}, x.options());
-  auto a2 = at::empty({M, 256}, x.options());
+  // a1/a2 exist only TRANSPOSED in wgrad fragment-major layout (plus
+  // 32-bit relu-mask words per column) — the backward chain consumes
+  // masks, the wgrad kernel consumes the fragments; nothing reads
+  // row-major a1/a2.
+  const int64_t mtiles = std::max<int64_t>((M + 31) / 32, 1);
+  const int64_t mchunks = mtiles * 2;
+  auto a1t = at::empty({16 * mchunks * 512}, x.options());
+  auto a2t = at::empty({8 * mchunks * 512}, x.options());
+  auto mask1 = at::empty({mtiles, 512}, x.options().dtype(at::kInt));
+  auto mask2 = at::empty({mtiles, 256}, x.options().dtype(at::kInt));
   auto a3 = at::empty({M, 128}, x.options());
   auto out = at::empty({M, 1}, x.options());
   // Fused MSE epilogue: with a target, the kernel also emits the bf16
@@ -543,12 +553,15 @@ std::vector<at::Tensor> fwd_chain_bf16(
     launch_fwd_chain(xs.data_ptr(), W1s.data_ptr(), b1f.data_ptr<float>(),
                      W2s.data_ptr(), b2f.data_ptr<float>(), W3s.data_ptr(),
                      b3f.data_ptr<float>(), w4.data_ptr(),
-                     b4f.data_ptr<float>(), a1.data_ptr(), a2.data_ptr(),
+                     b4f.data_ptr<float>(), a1t.data_ptr(),
+                     reinterpret_cast<uint32_t*>(mask1.data_ptr<int32_t>()),
+                     a2t.data_ptr(),
+                     reinterpret_cast<uint32_t*>(mask2.data_ptr<int32_t>()),
                      a3.data_ptr(), out.data_ptr(), tgt_ptr, dyb_ptr,
                      lp_ptr, M, current_stream());
   }
-  if (with_loss) return {a1, a2, a3, out, dyb, loss_part};
-  return {a1, a2, a3, out};
+  if (with_loss) return {a1t, mask1, a2t, mask2, a3, out, dyb, loss_part};
+  return {a1t, mask1, a2t, mask2, a3, out};
 }
 
 
@@ -568,10 +581,11 @@ static at::Tensor swizzle_frag_T(const at::Tensor& W) {
 }
 
 std::vector<at::Tensor> bwd_chain_bf16(
-    const at::Tensor& dy, const at::Tensor& a1, const at::Tensor& a2,
-    const at::Tensor& a3, const at::Tensor& w4, const at::Tensor& W3,
+    const at::Tensor& dy, const at::Tensor& a3, const at::Tensor& mask1,
+    const at::Tensor& mask2, const at::Tensor& w4, const at::Tensor& W3,
     const at::Tensor& W2) {
   const int64_t M = dy.size(0);
+  const int64_t mtiles = std::max<int64_t>((M + 31) / 32, 1);
   TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == at::kBFloat16 &&
                   dy.numel() == M && dy.is_contiguous(),
               "bwd_chain: dy must be contiguous bf16 [M] or [M,1]");
@@ -581,9 +595,13 @@ std::vector<at::Tensor> bwd_chain_bf16(
                     a.is_contiguous(),
                 "bwd_chain: ", nm, " must be contiguous bf16 [M,", n, "]");
   };
-  chk_a(a1, 512, "a1");
-  chk_a(a2, 256, "a2");
   chk_a(a3, 128, "a3");
+  TORCH_CHECK(mask1.is_cuda() && mask1.scalar_type() == at::kInt &&
+                  mask1.is_contiguous() && mask1.numel() == mtiles * 512,
+              "bwd_chain: mask1 must be int32 [mtiles,512]");
+  TORCH_CHECK(mask2.is_cuda() && mask2.scalar_type() == at::kInt &&
+                  mask2.is_contiguous() && mask2.numel() == mtiles * 256,
+              "bwd_chain: mask2 must be int32 [mtiles,256]");
   TORCH_CHECK(w4.numel() == 128 && w4.scalar_type() == at::kBFloat16,
               "bwd_chain: w4 must be bf16 [128]");
   // Weights may arrive pre-transposed ([256,128] / [512,256], cached by
@@ -604,9 +622,11 @@ std::vector<at::Tensor> bwd_chain_bf16(
                                   : swizzle_frag(W3.contiguous());
   auto W2Ts = (W2.size(0) == 256) ? swizzle_frag_T(W2.contiguous())
                                   : swizzle_frag(W2.contiguous());
-  auto dz1 = at::empty({M, 512}, dy.options());
-  auto dz2 = at::empty({M, 256}, dy.options());
-  auto dz3 = at::empty({M, 128}, dy.options());
+  // dz outputs exist only transposed, in wgrad fragment-major layout.
+  const int64_t mchunks = mtiles * 2;
+  auto dz1t = at::empty({16 * mchunks * 512}, dy.options());
+  auto dz2t = at::empty({8 * mchunks * 512}, dy.options());
+  auto dz3t = at::empty({4 * mchunks * 512}, dy.options());
   const int64_t grid = bwd_chain_grid(M);
   constexpr int64_t kPartW = 512 + 256 + 128 + 1 + 256;
   // Every block writes its whole db_part row (bc_bias_partial covers all
@@ -616,10 +636,10 @@ std::vector<at::Tensor> bwd_chain_bf16(
                            dy.options().dtype(at::kFloat));
   if (M == 0) db_part.zero_();
   if (M > 0) {
-    launch_bwd_chain(dy.data_ptr(), a1.data_ptr(), a2.data_ptr(),
-                     a3.data_ptr(), w4c.data_ptr(), W3Ts.data_ptr(),
-                     W2Ts.data_ptr(), dz1.data_ptr(), dz2.data_ptr(),
-                     dz3.data_ptr(), db_part.data_ptr<float>(), M,
+    launch_bwd_chain(dy.data_ptr(), a3.data_ptr(), mask1.data_ptr(),
+                     mask2.data_ptr(), w4c.data_ptr(), W3Ts.data_ptr(),
+                     W2Ts.data_ptr(), dz1t.data_ptr(), dz2t.data_ptr(),
+                     dz3t.data_ptr(), db_part.data_ptr<float>(), M,
                      current_stream());
   }
   // Column reduction as a GEMV (ones^T @ db_part): hipBLASLt streams the
@@ -630,9 +650,9 @@ std::vector<at::Tensor> bwd_chain_bf16(
   auto db = at::mm(ones, db_part).reshape({kPartW});
   auto dw4 = (db.narrow(0, 897, 128) + db.narrow(0, 897 + 128, 128))
                  .reshape({1, 128});
-  return {dz1,
-          dz2,
-          dz3,
+  return {dz1t,
+          dz2t,
+          dz3t,
           db.narrow(0, 0, 512),
           db.narrow(0, 512, 256),
           db.narrow(0, 512 + 256, 128),
@@ -641,6 +661,21 @@ std::vector<at::Tensor> bwd_chain_bf16(
 }
 
 }  // namespace
+// x [M,100] -> wgrad fragment-major x^T ([128/32][mchunks][2][32][8],
+// zero-padded cols 100..127 and rows past M) for the dW1 wgrad.
+at::Tensor swizzle_xt_bf16(const at::Tensor& x) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
+                  x.dim() == 2 && x.size(1) == 100 && x.is_contiguous(),
+              "swizzle_xt: x must be contiguous bf16 [M,100]");
+  const int64_t M = x.size(0);
+  const int64_t mtiles = std::max<int64_t>((M + 31) / 32, 1);
+  auto out = at::empty({4 * mtiles * 2 * 512}, x.options());
+  if (M > 0) {
+    launch_swizzle_xt(x.data_ptr(), out.data_ptr(), M, current_stream());
+  }
+  return out;
+}
+
 // Fragment-major wgrad (csrc/wgrad_frag.hip): dW = dz^T @ src with both
 // inputs pre-swizzled to [C/32][Mp/16][2][32][8] fragment layout.
 at::Tensor wgrad_frag_bf16(const at::Tensor& AT, const at::Tensor& BT,
@@ -689,6 +724,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad_bf16", &rsdl::wgrad_bf16, py::arg("dy"), py::arg("x"),
         py::arg("with_bias") = true);
   m.def("relu_bwd_bias", &rsdl::relu_bwd_bias, py::arg("dy"), py::arg("y"));
+  m.def("swizzle_xt_bf16", &rsdl::swizzle_xt_bf16, py::arg("x"));
   m.def("wgrad_frag_bf16", &rsdl::wgrad_frag_bf16, py::arg("AT"),
         py::arg("BT"), py::arg("N"), py::arg("K"), py::arg("mchunks"),
         py::arg("nt_w"), py::arg("kt_w"));

@@ -1,23 +1,26 @@
-"""EXPERIMENTAL round-2 fused train step (docs/MEGAKERNEL_PLAN.md).
+"""Fused train step for the fixed TabularMLP(100-512-256-128-1) — the
+DEFAULT bench step (RSDL_FUSED_STEP=0 reverts to eager autocast).
 
-Manual forward+backward for the fixed TabularMLP(100-512-256-128-1)
-architecture using the fused chain kernels (csrc/fwd_chain.hip,
-csrc/bwd_chain.hip) plus the existing chunked-bmm split-K weight
-gradients. No autograd graph: the step IS the schedule —
-2 chain kernels + 4 wgrad GEMM groups + the caller's optimizer.
+No autograd graph; the step IS the schedule, 7 kernels total:
+  x swizzle -> fwd chain (3 MFMA layers + head + MSE loss/grad; emits
+  a1^T/a2^T wgrad fragments + relu-mask words) -> bwd chain (dgrad chain
+  from mask bits; emits dz^T fragments + db/dW4 partials) -> 3
+  fragment-major wgrad kernels -> GEMV partial reduce -> optimizer.
+Activations a1/a2 and gradients dz never exist in row-major form: the
+producers emit the transposed fragment layout the wgrad kernel reads
+(csrc/fwd_chain.hip, csrc/bwd_chain.hip, csrc/wgrad_frag.hip), and the
+backward reads 1-bit relu masks instead of activation values.
 
-NOT wired into any default path and NOT yet GPU-validated (written after
-round-1's GPU budget was spent); exercised by the RSDL_EXPERIMENTAL=1
-GPU test. Round 2: validate numerics, A/B in bench.py behind
-RSDL_FUSED_STEP=1, then consider hipGraph capture (the step shrinks to
-~10 launches).
+Validated by tests/test_gpu_kernels.py (chain numerics + whole-step
+parity vs eager autograd) and the lane-level CPU simulation in
+tests/test_chain_sim.py.
 """
 
 from typing import Tuple
 
 import torch
 
-from ray_shuffling_data_loader_amd.models.mlp import TabularMLP, _wgrad_chunks
+from ray_shuffling_data_loader_amd.models.mlp import TabularMLP
 
 
 def _layers(model: TabularMLP):
@@ -28,21 +31,6 @@ def _layers(model: TabularMLP):
     for m, (n, k) in zip(lin, shapes):
         assert m.weight.shape == (n, k), (m.weight.shape, (n, k))
     return lin
-
-
-def _wgrad_bmm(dz: torch.Tensor, src: torch.Tensor) -> torch.Tensor:
-    """dW = dz^T @ src via the split-K chunked bmm (models/mlp.py)."""
-    m = dz.shape[0]
-    c = _wgrad_chunks(m) if m >= 1 << 16 else 1
-    if c > 1:
-        return (
-            torch.bmm(
-                dz.view(c, m // c, dz.shape[1]).transpose(1, 2),
-                src.view(c, m // c, src.shape[1]),
-            )
-            .sum(0)
-        )
-    return dz.t() @ src
 
 
 def _weight_buffers(model: TabularMLP, lin):
@@ -80,22 +68,33 @@ def fused_step(
     hip = _load_hip()
     lin = _layers(model)
     M = x.shape[0]
+    from ray_shuffling_data_loader_amd.ops.shuffle_ops import wgrad_frag
+
     buf = _weight_buffers(model, lin)
     b1, b2, b3, b4 = (m.bias.detach() for m in lin)
-    # Loss + dy are fused into the forward kernel's head epilogue.
-    a1, a2, a3, out, dyb, loss_part = hip.fwd_chain_bf16(
+    # Forward chain: a1/a2 exist only TRANSPOSED (wgrad fragment-major)
+    # plus 32-bit relu-mask words; loss + dy fold into the head epilogue.
+    a1t, mask1, a2t, mask2, a3, out, dyb, loss_part = hip.fwd_chain_bf16(
         x, buf["W1p"], b1, buf["W2"], b2, buf["W3"], b3, buf["w4"], b4,
         target=target,
     )
     loss = loss_part.sum() / M
-    # dW4/db4 partials are folded into the backward kernel's seed loop.
-    dz1, dz2, dz3, db1, db2, db3, db4, dw4 = hip.bwd_chain_bf16(
-        dyb, a1, a2, a3, buf["w4"], buf["W3"], buf["W2"]
+    # Backward chain: consumes the masks (never the activations), emits
+    # dz^T fragments; dW4/db4 partials fold into its seed loop.
+    dz1t, dz2t, dz3t, db1, db2, db3, db4, dw4 = hip.bwd_chain_bf16(
+        dyb, a3, mask1, mask2, buf["w4"], buf["W3"], buf["W2"]
     )
+    # Weight grads: fragment-major MFMA wgrad kernel (csrc/wgrad_frag.hip)
+    # reading the transposed fragments both producers emitted.
+    mchunks = 2 * ((M + 31) // 32)
+    xt = hip.swizzle_xt_bf16(x)
+    dw1 = wgrad_frag(dz1t, xt, 512, 128, mchunks)[:, :100].contiguous()
+    dw2 = wgrad_frag(dz2t, a1t, 256, 512, mchunks)
+    dw3 = wgrad_frag(dz3t, a2t, 128, 256, mchunks)
     grads = [
-        (_wgrad_bmm(dz1, x), db1),
-        (_wgrad_bmm(dz2, a1), db2),
-        (_wgrad_bmm(dz3, a2), db3),
+        (dw1, db1),
+        (dw2, db2),
+        (dw3, db3),
         (dw4, db4),
     ]
     for m, (gw, gb) in zip(lin, grads):

@@ -283,3 +283,33 @@ def wgrad_frag(at_frag: torch.Tensor, bt_frag: torch.Tensor, n: int,
     nt_w, kt_w = _WGRAD_FRAG_CFG[(n, k)]
     hip = _load_hip()
     return hip.wgrad_frag_bf16(at_frag, bt_frag, n, k, mchunks, nt_w, kt_w)
+
+
+def t_frag_unswizzle(flat: torch.Tensor, m: int, c: int) -> torch.Tensor:
+    """Inverse of :func:`t_frag_swizzle`: fragment-major transposed flat
+    tensor -> row-major [m, c] (pad rows dropped)."""
+    mchunks = flat.numel() // (c * 16)
+    v = flat.reshape(c // 32, mchunks, 2, 32, 8)
+    return (
+        v.permute(1, 2, 4, 0, 3).reshape(mchunks * 16, c)[:m].contiguous()
+    )
+
+
+def relu_mask_words(a: torch.Tensor) -> torch.Tensor:
+    """[M, N] activations -> [ceil(M/32), N] int32 relu-mask words (bit i
+    of word (t, n) = a[t*32+i, n] > 0; pad rows 0). Torch oracle for the
+    mask layout the forward chain kernel emits."""
+    m, n = a.shape
+    mt = (m + 31) // 32
+    pad = mt * 32 - m
+    bits = a > 0
+    if pad:
+        bits = torch.cat(
+            [bits, torch.zeros(pad, n, dtype=torch.bool, device=a.device)]
+        )
+    bits = bits.view(mt, 32, n).long()
+    weights = (1 << torch.arange(32, device=a.device, dtype=torch.long))
+    words = (bits * weights.view(1, 32, 1)).sum(1)
+    return (words & 0xFFFFFFFF).to(torch.int32) if False else (
+        words - ((words >> 31) & 1) * (1 << 32)
+    ).to(torch.int32)

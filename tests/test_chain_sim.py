@@ -240,46 +240,74 @@ def test_fused_step_glue_cpu(monkeypatch):
     from ray_shuffling_data_loader_amd.models import fused_step as fs
     from ray_shuffling_data_loader_amd.models.mlp import TabularMLP
 
+    from ray_shuffling_data_loader_amd.ops.shuffle_ops import (
+        relu_mask_words,
+        t_frag_swizzle,
+        t_frag_unswizzle,
+    )
+
     class FakeHip:
+        """Exact torch mirror of the HIP binding surface: transposed
+        fragment-major a/dz tensors, relu-mask words, fused loss."""
+
         @staticmethod
         def fwd_chain_bf16(x, W1, b1, W2, b2, W3, b3, w4, b4, target=None):
-            # The binding accepts W1 pre-padded [512,112] (zeros in cols
-            # 100..111) or raw [512,100] — mirror that dispatch.
             if W1.shape[1] == 112:
                 W1 = W1[:, :100]
             a1 = torch.relu(x.float() @ W1.float().t() + b1.float())
             a2 = torch.relu(a1 @ W2.float().t() + b2.float())
             a3 = torch.relu(a2 @ W3.float().t() + b3.float())
             out = a3 @ w4.float().unsqueeze(1) + b4.float()
-            res = (a1.bfloat16(), a2.bfloat16(), a3.bfloat16(),
-                   out.bfloat16())
+            a1b, a2b = a1.bfloat16(), a2.bfloat16()
+            res = (t_frag_swizzle(a1b), relu_mask_words(a1b.float()),
+                   t_frag_swizzle(a2b), relu_mask_words(a2b.float()),
+                   a3.bfloat16(), out.bfloat16())
             if target is None:
                 return res
-            # Fused MSE epilogue parity (binding: dyb + loss partials).
             M = x.shape[0]
-            diff = res[3].float() - target.float().reshape(-1, 1)
+            diff = res[5].float() - target.float().reshape(-1, 1)
             dyb = ((2.0 / M) * diff).bfloat16()
             loss_part = diff.square().sum().reshape(1)
             return res + (dyb, loss_part)
 
         @staticmethod
-        def bwd_chain_bf16(dy, a1, a2, a3, w4, W3, W2):
-            # Accept model-layout or pre-transposed weights (binding
-            # dispatches on shape: W3T [256,128], W2T [512,256]).
-            if W3.shape[0] == 256:
-                W3 = W3.t()
-            if W2.shape[0] == 512:
-                W2 = W2.t()
+        def bwd_chain_bf16(dy, a3, mask1, mask2, w4, W3, W2):
+            M = dy.shape[0]
+
+            def mask_of(words, n):
+                mt = words.shape[0]
+                w = words.to(torch.int64) & 0xFFFFFFFF
+                shifts = torch.arange(32, device=words.device)
+                bits = (
+                    (w.view(mt, 1, n) >> shifts.view(1, 32, 1)) & 1
+                ).reshape(mt * 32, n)[:M]
+                return bits.bool()
+
             da3 = dy.float() @ w4.float().unsqueeze(0)
             dz3 = (da3 * (a3.float() > 0)).bfloat16()
             da2 = dz3.float() @ W3.float()
-            dz2 = (da2 * (a2.float() > 0)).bfloat16()
+            dz2 = (da2 * mask_of(mask2, 256)).bfloat16()
             da1 = dz2.float() @ W2.float()
-            dz1 = (da1 * (a1.float() > 0)).bfloat16()
+            dz1 = (da1 * mask_of(mask1, 512)).bfloat16()
             dw4 = dy.float().t() @ a3.float()
-            return (dz1, dz2, dz3,
+            return (t_frag_swizzle(dz1), t_frag_swizzle(dz2),
+                    t_frag_swizzle(dz3),
                     dz1.float().sum(0), dz2.float().sum(0),
                     dz3.float().sum(0), dy.float().sum(0), dw4)
+
+        @staticmethod
+        def swizzle_xt_bf16(x):
+            m = x.shape[0]
+            mp = (m + 31) // 32 * 32
+            xpad = torch.nn.functional.pad(x, (0, 28, 0, mp - m))
+            return t_frag_swizzle(xpad)
+
+        @staticmethod
+        def wgrad_frag_bf16(at_f, bt_f, n, k, mchunks, nt_w, kt_w):
+            m = mchunks * 16
+            dz = t_frag_unswizzle(at_f, m, n)
+            src = t_frag_unswizzle(bt_f, m, k)
+            return dz.t().float() @ src.float()
 
     monkeypatch.setattr(fs, "_load_hip", lambda: FakeHip, raising=False)
     import ray_shuffling_data_loader_amd.ops.shuffle_ops as so

@@ -543,10 +543,17 @@ def test_fwd_chain_matches_eager(dev):
           for n, k in [(512, 100), (256, 512), (128, 256), (1, 128)]]
     bs = [torch.randn(n, device=dev).bfloat16() / 8
           for n in (512, 256, 128, 1)]
-    a1, a2, a3, out = hip.fwd_chain_bf16(
+    from ray_shuffling_data_loader_amd.ops.shuffle_ops import (
+        relu_mask_words,
+        t_frag_unswizzle,
+    )
+
+    a1t, mask1, a2t, mask2, a3, out = hip.fwd_chain_bf16(
         x, Ws[0], bs[0], Ws[1], bs[1], Ws[2], bs[2],
         Ws[3].flatten(), bs[3],
     )
+    a1 = t_frag_unswizzle(a1t, M, 512)
+    a2 = t_frag_unswizzle(a2t, M, 256)
     # fp32 eager oracle on the bf16 inputs
     r = x.float()
     refs = []
@@ -555,6 +562,20 @@ def test_fwd_chain_matches_eager(dev):
         if i < 3:
             r = torch.relu(r)
         refs.append(r)
+    # relu-mask words match the bf16-rounded sign of the emitted values.
+    # Pad rows (M..Mp) carry arbitrary bits (their dz is zero regardless);
+    # mask them off before comparing.
+    mt = (M + 31) // 32
+    valid = torch.zeros(mt, dtype=torch.int64, device=dev)
+    full = M // 32
+    valid[:full] = -1 & 0xFFFFFFFF
+    if M % 32:
+        valid[full] = (1 << (M % 32)) - 1
+    for got_m, got_a, n in [(mask1, a1, 512), (mask2, a2, 256)]:
+        ref_words = relu_mask_words(got_a.float())
+        got = (got_m[:mt].to(torch.int64) & 0xFFFFFFFF) & valid.view(-1, 1)
+        ref = (ref_words.to(torch.int64) & 0xFFFFFFFF) & valid.view(-1, 1)
+        assert torch.equal(got, ref), n
     for got, ref, name in [(a1, refs[0], "a1"), (a2, refs[1], "a2"),
                            (a3, refs[2], "a3"), (out, refs[3], "out")]:
         ref_b = ref.bfloat16().float()
@@ -566,7 +587,7 @@ def test_fwd_chain_matches_eager(dev):
 
     # Fused MSE epilogue (target given): dyb + loss partials.
     tgt = torch.randn(M, 1, device=dev)
-    a1b, a2b, a3b, outb, dyb, loss_part = hip.fwd_chain_bf16(
+    _, _, _, _, a3b, outb, dyb, loss_part = hip.fwd_chain_bf16(
         x, Ws[0], bs[0], Ws[1], bs[1], Ws[2], bs[2],
         Ws[3].flatten(), bs[3], target=tgt,
     )
@@ -597,9 +618,19 @@ def test_bwd_chain_matches_eager(dev):
     w4 = (torch.randn(128, device=dev) / 11).bfloat16()
     W3 = (torch.randn(128, 256, device=dev) / 16).bfloat16()
     W2 = (torch.randn(256, 512, device=dev) / 22).bfloat16()
-    dz1, dz2, dz3, db1, db2, db3, db4, dw4 = hip.bwd_chain_bf16(
-        dy, a1, a2, a3, w4, W3, W2
+    from ray_shuffling_data_loader_amd.ops.shuffle_ops import (
+        relu_mask_words,
+        t_frag_unswizzle,
     )
+
+    mask1 = relu_mask_words(a1.float())
+    mask2 = relu_mask_words(a2.float())
+    dz1t, dz2t, dz3t, db1, db2, db3, db4, dw4 = hip.bwd_chain_bf16(
+        dy, a3, mask1, mask2, w4, W3, W2
+    )
+    dz1 = t_frag_unswizzle(dz1t, M, 512)
+    dz2 = t_frag_unswizzle(dz2t, M, 256)
+    dz3 = t_frag_unswizzle(dz3t, M, 128)
     # fp32 eager oracle
     da3 = dy.float() @ w4.float().unsqueeze(0)
     rz3 = da3 * (a3.float() > 0)
