@@ -145,12 +145,18 @@ void launch_wgrad_frag(const void* AT, const void* BT, float* dW, int32_t N,
   const int32_t nblk_n = N / (nt_w * 128);
   const int32_t nblk_k = K / (kt_w * 32);
   const int32_t nblk = nblk_n * nblk_k;
-  // target workgroup count: more slabs = more parallelism but more
-  // atomic writers per output line (RSDL_WGRAD_WGS to tune).
-  static int64_t target = [] {
+  // Workgroup-count target: more slabs = more stream parallelism but
+  // more atomic writers per output line. Measured optima at the
+  // flagship shapes (profiles/PERF.md): the small-output configs favor
+  // fewer slabs. RSDL_WGRAD_WGS overrides for tuning.
+  static int64_t env_target = [] {
     const char* e = std::getenv("RSDL_WGRAD_WGS");
-    return e ? atoll(e) : 2048;
+    return e ? atoll(e) : 0;
   }();
+  int64_t target = env_target;
+  if (target <= 0) {
+    target = (nt_w == 2) ? 256 : (K >= 512 ? 512 : 256);
+  }
   int64_t nslabs = (target + nblk - 1) / nblk;
   if (nslabs > mchunks) nslabs = mchunks;
   const int64_t chunks_per_slab = (mchunks + nslabs - 1) / nslabs;
