@@ -171,30 +171,32 @@ __device__ void bc_layer(const short* __restrict__ dz_src,
     }
     colsum += __shfl_xor(colsum, 32);
     if (lane < 32) db_out[n] = colsum;
-    // Transposed fragment emission (see fwd_chain fc_layer EMIT_T).
-    float ex0[4], ex1[4];
+    // Transposed fragment emission, packed-pair exchange (see
+    // fwd_chain fc_layer EMIT_T).
+    uint32_t p[8];
     #pragma unroll
-    for (int j = 0; j < 4; j++) {
-      ex0[j] = __shfl_xor(h == 0 ? vals[4 + j] : vals[j], 32);
-      ex1[j] = __shfl_xor(h == 0 ? vals[12 + j] : vals[8 + j], 32);
+    for (int q = 0; q < 8; q++) {
+      const short lo = bc_f2b(vals[2 * q]);
+      const short hi = bc_f2b(vals[2 * q + 1]);
+      p[q] = ((uint32_t)(uint16_t)hi << 16) | (uint16_t)lo;
     }
-    short run0[8], run1[8];
+    uint32_t rx[4];
     #pragma unroll
-    for (int j = 0; j < 4; j++) {
-      run0[j] = bc_f2b(h == 0 ? vals[j] : ex0[j]);
-      run0[4 + j] = bc_f2b(h == 0 ? ex0[j] : vals[4 + j]);
-      run1[j] = bc_f2b(h == 0 ? vals[8 + j] : ex1[j]);
-      run1[4 + j] = bc_f2b(h == 0 ? ex1[j] : vals[12 + j]);
+    for (int j = 0; j < 2; j++) {
+      rx[j] = __shfl_xor((int)(h == 0 ? p[2 + j] : p[j]), 32);
+      rx[2 + j] = __shfl_xor((int)(h == 0 ? p[6 + j] : p[4 + j]), 32);
     }
+    const bc_u32x4 run0 = {h == 0 ? p[0] : rx[0], h == 0 ? p[1] : rx[1],
+                           h == 0 ? rx[0] : p[2], h == 0 ? rx[1] : p[3]};
+    const bc_u32x4 run1 = {h == 0 ? p[4] : rx[2], h == 0 ? p[5] : rx[3],
+                           h == 0 ? rx[2] : p[6], h == 0 ? rx[3] : p[7]};
     const int64_t nt_g = (int64_t)(n_base + nt * 32) >> 5;
     short* blk0 = dzt_out + ((nt_g * mchunks + mc0) * 512) + h * 256 +
                   ml * 8;
     short* blk1 = dzt_out + ((nt_g * mchunks + mc0 + 1) * 512) + h * 256 +
                   ml * 8;
-    __builtin_nontemporal_store(*reinterpret_cast<bc_u32x4*>(run0),
-                                reinterpret_cast<bc_u32x4*>(blk0));
-    __builtin_nontemporal_store(*reinterpret_cast<bc_u32x4*>(run1),
-                                reinterpret_cast<bc_u32x4*>(blk1));
+    __builtin_nontemporal_store(run0, reinterpret_cast<bc_u32x4*>(blk0));
+    __builtin_nontemporal_store(run1, reinterpret_cast<bc_u32x4*>(blk1));
   }
 }
 

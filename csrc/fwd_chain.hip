@@ -192,39 +192,45 @@ __device__ void fc_layer(const short* __restrict__ src_lds,
         vals[reg] = v;
       }
       if (EMIT_T) {
-        // Exchange the half-rows: lane h=0 assembles rows {0-7},{16-23},
-        // h=1 assembles {8-15},{24-31}.
-        float ex0[4], ex1[4];
+        // Pack the bf16 values in row pairs, then exchange the
+        // half-rows as packed ints: 4 shfls, no post-exchange converts.
+        // Lane h=0 assembles rows {0-7},{16-23}; h=1 {8-15},{24-31}.
+        uint32_t p[8];
+        uint32_t w = 0;
         #pragma unroll
-        for (int j = 0; j < 4; j++) {
-          ex0[j] = __shfl_xor(h == 0 ? vals[4 + j] : vals[j], 32);
-          ex1[j] = __shfl_xor(h == 0 ? vals[12 + j] : vals[8 + j], 32);
+        for (int q = 0; q < 8; q++) {
+          const short lo = fc_f2b(vals[2 * q]);
+          const short hi = fc_f2b(vals[2 * q + 1]);
+          p[q] = ((uint32_t)(uint16_t)hi << 16) | (uint16_t)lo;
+          const int32_t mrow = ((2 * q) & 3) + 8 * (q >> 1) + 4 * h;
+          w |= (uint32_t)(lo > 0) << mrow;
+          w |= (uint32_t)(hi > 0) << (mrow + 1);
         }
-        short run0[8], run1[8];
+        uint32_t rx[4];
         #pragma unroll
-        for (int j = 0; j < 4; j++) {
-          run0[j] = fc_f2b(h == 0 ? vals[j] : ex0[j]);
-          run0[4 + j] = fc_f2b(h == 0 ? ex0[j] : vals[4 + j]);
-          run1[j] = fc_f2b(h == 0 ? vals[8 + j] : ex1[j]);
-          run1[4 + j] = fc_f2b(h == 0 ? ex1[j] : vals[12 + j]);
+        for (int j = 0; j < 2; j++) {
+          rx[j] = __shfl_xor((int)(h == 0 ? p[2 + j] : p[j]), 32);
+          rx[2 + j] = __shfl_xor((int)(h == 0 ? p[6 + j] : p[4 + j]), 32);
         }
+        const fc_u32x4 run0 = {h == 0 ? p[0] : rx[0],
+                               h == 0 ? p[1] : rx[1],
+                               h == 0 ? rx[0] : p[2],
+                               h == 0 ? rx[1] : p[3]};
+        const fc_u32x4 run1 = {h == 0 ? p[4] : rx[2],
+                               h == 0 ? p[5] : rx[3],
+                               h == 0 ? rx[2] : p[6],
+                               h == 0 ? rx[3] : p[7]};
         // run0 -> (mc_local 0, h'=h); run1 -> (mc_local 1, h'=h)
         const int64_t nt_g = (int64_t)(n_base + nt * 32) >> 5;
         short* blk0 = at_out + ((nt_g * mchunks + mc0) * 512) + h * 256 +
                       ml * 8;
         short* blk1 = at_out + ((nt_g * mchunks + mc0 + 1) * 512) +
                       h * 256 + ml * 8;
-        __builtin_nontemporal_store(*reinterpret_cast<fc_u32x4*>(run0),
+        __builtin_nontemporal_store(run0,
                                     reinterpret_cast<fc_u32x4*>(blk0));
-        __builtin_nontemporal_store(*reinterpret_cast<fc_u32x4*>(run1),
+        __builtin_nontemporal_store(run1,
                                     reinterpret_cast<fc_u32x4*>(blk1));
         // Mask word for column n (bit i = bf16 value of row i > 0).
-        uint32_t w = 0;
-        #pragma unroll
-        for (int reg = 0; reg < 16; reg++) {
-          const int32_t mrow = (reg & 3) + 8 * (reg >> 2) + 4 * h;
-          w |= (uint32_t)(fc_f2b(vals[reg]) > 0) << mrow;
-        }
         w |= __shfl_xor(w, 32);
         if (h == 0) mask_row[n] = w;
       }
