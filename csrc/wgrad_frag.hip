@@ -45,12 +45,19 @@ __global__ void __launch_bounds__(256, 2) wgrad_frag_kernel(
     float* __restrict__ dW,  // [N,K] fp32, pre-zeroed
     int32_t N, int32_t K, int64_t mchunks, int32_t nblk_n, int32_t nblk_k,
     int64_t nslabs, int64_t chunks_per_slab) {
-  const int32_t NBLK = nblk_n * nblk_k;
-  const int32_t xcd = blockIdx.x & 7;
-  const int32_t g = blockIdx.x >> 3;
-  const int32_t slab_local = g / NBLK;
-  const int32_t blk = g % NBLK;
-  const int64_t slab = (int64_t)slab_local * 8 + xcd;
+  const int32_t NBLK = (nblk_n < 0 ? -nblk_n : nblk_n) * nblk_k;
+  int64_t slab;
+  int32_t blk;
+  if (nblk_n < 0) {  // linear decode (A/B: RSDL_WGRAD_LINEAR)
+    nblk_n = -nblk_n;
+    blk = (int32_t)(blockIdx.x % (uint32_t)NBLK);
+    slab = blockIdx.x / (uint32_t)NBLK;
+  } else {
+    const int32_t xcd = blockIdx.x & 7;
+    const int32_t g = blockIdx.x >> 3;
+    blk = g % NBLK;
+    slab = (int64_t)(g / NBLK) * 8 + xcd;
+  }
   if (slab >= nslabs) return;
   const int32_t bn = blk / nblk_k;
   const int32_t bk = blk % nblk_k;
@@ -160,19 +167,22 @@ void launch_wgrad_frag(const void* AT, const void* BT, float* dW, int32_t N,
   int64_t nslabs = (target + nblk - 1) / nblk;
   if (nslabs > mchunks) nslabs = mchunks;
   const int64_t chunks_per_slab = (mchunks + nslabs - 1) / nslabs;
-  const int64_t grid = ((nslabs + 7) / 8) * 8 * nblk;
+  static bool linear = std::getenv("RSDL_WGRAD_LINEAR") != nullptr;
+  const int64_t grid =
+      linear ? nslabs * nblk : ((nslabs + 7) / 8) * 8 * nblk;
+  const int32_t nb_n = linear ? -nblk_n : nblk_n;
   if (nt_w == 2 && kt_w == 4) {
     hipLaunchKernelGGL((wgrad_frag_kernel<2, 4>), dim3((uint32_t)grid),
                        dim3(256), 0, stream,
                        reinterpret_cast<const short*>(AT),
                        reinterpret_cast<const short*>(BT), dW, N, K, mchunks,
-                       nblk_n, nblk_k, nslabs, chunks_per_slab);
+                       nb_n, nblk_k, nslabs, chunks_per_slab);
   } else if (nt_w == 1 && kt_w == 8) {
     hipLaunchKernelGGL((wgrad_frag_kernel<1, 8>), dim3((uint32_t)grid),
                        dim3(256), 0, stream,
                        reinterpret_cast<const short*>(AT),
                        reinterpret_cast<const short*>(BT), dW, N, K, mchunks,
-                       nblk_n, nblk_k, nslabs, chunks_per_slab);
+                       nb_n, nblk_k, nslabs, chunks_per_slab);
   }
 }
 
