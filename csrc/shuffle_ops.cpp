@@ -84,6 +84,10 @@ int64_t bwd_chain_grid(int64_t M);
 void launch_wgrad_frag(const void* AT, const void* BT, float* dW, int32_t N,
                        int32_t K, int64_t mchunks, int32_t nt_w,
                        int32_t kt_w, hipStream_t stream);
+int64_t wgrad_frag_nslabs(int64_t mchunks, int32_t N, int32_t K,
+                          int32_t nt_w, int32_t kt_w);
+void launch_slab_reduce(const float* part, float* out, int64_t nk,
+                        int64_t nslabs, hipStream_t stream);
 void launch_bwd_chain(const void* dy, const void* a3, const void* mask1,
                       const void* mask2, const void* w4, const void* W3T,
                       const void* W2T, void* dz1t, void* dz2t, void* dz3t,
@@ -693,11 +697,21 @@ at::Tensor wgrad_frag_bf16(const at::Tensor& AT, const at::Tensor& BT,
               "wgrad_frag: unsupported tile config");
   TORCH_CHECK(N % (nt_w * 128) == 0 && K % (kt_w * 32) == 0,
               "wgrad_frag: N/K not divisible by block shape");
-  auto dW = at::zeros({N, K}, AT.options().dtype(at::kFloat));
+  auto dW = at::empty({N, K}, AT.options().dtype(at::kFloat));
   if (mchunks > 0) {
-    launch_wgrad_frag(AT.data_ptr(), BT.data_ptr(), dW.data_ptr<float>(),
+    // Per-slab fp32 partials + second-stage reduce (no atomics).
+    const int64_t nslabs = wgrad_frag_nslabs(mchunks, (int32_t)N,
+                                             (int32_t)K, (int32_t)nt_w,
+                                             (int32_t)kt_w);
+    auto part =
+        at::empty({nslabs, N * K}, AT.options().dtype(at::kFloat));
+    launch_wgrad_frag(AT.data_ptr(), BT.data_ptr(), part.data_ptr<float>(),
                       (int32_t)N, (int32_t)K, mchunks, (int32_t)nt_w,
                       (int32_t)kt_w, current_stream());
+    launch_slab_reduce(part.data_ptr<float>(), dW.data_ptr<float>(),
+                       N * K, nslabs, current_stream());
+  } else {
+    dW.zero_();
   }
   return dW;
 }

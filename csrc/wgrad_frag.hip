@@ -128,7 +128,10 @@ __global__ void __launch_bounds__(256, 2) wgrad_frag_kernel(
 #undef WF_MFMA
 
   // Epilogue: D[row = n-in-tile][col = k-in-tile]; 32 lanes write 32
-  // consecutive k's -> coalesced 128-B atomic rows.
+  // consecutive k's. Plain per-slab partial stores (a second-stage
+  // reduce sums the slabs) — global fp32 atomics at ~16M adds per
+  // launch cost 35-65 us of contention.
+  float* part = dW + slab * (int64_t)N * K;
   #pragma unroll
   for (int nt = 0; nt < NT_W; nt++) {
     #pragma unroll
@@ -138,35 +141,75 @@ __global__ void __launch_bounds__(256, 2) wgrad_frag_kernel(
       for (int reg = 0; reg < 16; reg++) {
         const int32_t n = (nt0 + nt) * 32 + (reg & 3) + 8 * (reg >> 2) +
                           4 * (lane >> 5);
-        unsafeAtomicAdd(&dW[(int64_t)n * K + k], acc[nt][kt][reg]);
+        __builtin_nontemporal_store(acc[nt][kt][reg],
+                                    &part[(int64_t)n * K + k]);
       }
     }
   }
 }
 
+// out[i] = sum_s part[s][i] — the slab-partial reduction (fp32, float4).
+typedef __attribute__((__vector_size__(4 * sizeof(float)))) float wf_f32x4;
+__global__ void __launch_bounds__(256) slab_reduce_kernel(
+    const float* __restrict__ part, float* __restrict__ out, int64_t nk4,
+    int64_t nslabs) {
+  const int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x;
+  if (i >= nk4) return;
+  wf_f32x4 s = {0.f, 0.f, 0.f, 0.f};
+  for (int64_t sl = 0; sl < nslabs; sl++) {
+    s += __builtin_nontemporal_load(
+        reinterpret_cast<const wf_f32x4*>(part + (sl * nk4 + i) * 4));
+  }
+  *reinterpret_cast<wf_f32x4*>(out + i * 4) = s;
+}
+
 // layer configs: 1 -> NT_W=2,KT_W=4 (dW1 [512,128pad]); 2 -> 1,8
 // (dW2 [256,512]); 3 -> 1,8 (dW3 [128,256]).
+int64_t wgrad_frag_nslabs(int64_t mchunks, int32_t N, int32_t K,
+                          int32_t nt_w, int32_t kt_w);
+
+void launch_slab_reduce(const float* part, float* out, int64_t nk,
+                        int64_t nslabs, hipStream_t stream) {
+  const int64_t nk4 = nk / 4;
+  const int64_t grid = (nk4 + 255) / 256;
+  hipLaunchKernelGGL(slab_reduce_kernel, dim3((uint32_t)grid), dim3(256),
+                     0, stream, part, out, nk4, nslabs);
+}
+
+static int64_t frag_target(int32_t K, int32_t nt_w) {
+  static int64_t env_target = [] {
+    const char* e = std::getenv("RSDL_WGRAD_WGS");
+    return e ? atoll(e) : 0;
+  }();
+  if (env_target > 0) return env_target;
+  return (nt_w == 2) ? 256 : (K >= 512 ? 512 : 256);
+}
+
+int64_t wgrad_frag_nslabs(int64_t mchunks, int32_t N, int32_t K,
+                          int32_t nt_w, int32_t kt_w) {
+  const int32_t nblk = (N / (nt_w * 128)) * (K / (kt_w * 32));
+  int64_t nslabs = (frag_target(K, nt_w) + nblk - 1) / nblk;
+  if (nslabs > mchunks) nslabs = mchunks;
+  const int64_t cps = (mchunks + nslabs - 1) / nslabs;
+  return (mchunks + cps - 1) / cps;
+}
+
 void launch_wgrad_frag(const void* AT, const void* BT, float* dW, int32_t N,
                        int32_t K, int64_t mchunks, int32_t nt_w,
                        int32_t kt_w, hipStream_t stream) {
   const int32_t nblk_n = N / (nt_w * 128);
   const int32_t nblk_k = K / (kt_w * 32);
   const int32_t nblk = nblk_n * nblk_k;
-  // Workgroup-count target: more slabs = more stream parallelism but
-  // more atomic writers per output line. Measured optima at the
-  // flagship shapes (profiles/PERF.md): the small-output configs favor
-  // fewer slabs. RSDL_WGRAD_WGS overrides for tuning.
-  static int64_t env_target = [] {
-    const char* e = std::getenv("RSDL_WGRAD_WGS");
-    return e ? atoll(e) : 0;
-  }();
-  int64_t target = env_target;
-  if (target <= 0) {
-    target = (nt_w == 2) ? 256 : (K >= 512 ? 512 : 256);
-  }
+  // Workgroup-count target: more slabs = more stream parallelism but a
+  // bigger partial buffer to reduce. Measured optima at the flagship
+  // shapes (profiles/PERF.md). RSDL_WGRAD_WGS overrides for tuning.
+  int64_t target = frag_target(K, nt_w);
   int64_t nslabs = (target + nblk - 1) / nblk;
   if (nslabs > mchunks) nslabs = mchunks;
   const int64_t chunks_per_slab = (mchunks + nslabs - 1) / nslabs;
+  // re-derive so every slab is non-empty (partial rows must all be
+  // written; there is no zero-init)
+  nslabs = (mchunks + chunks_per_slab - 1) / chunks_per_slab;
   static bool linear = std::getenv("RSDL_WGRAD_LINEAR") != nullptr;
   const int64_t grid =
       linear ? nslabs * nblk : ((nslabs + 7) / 8) * 8 * nblk;
