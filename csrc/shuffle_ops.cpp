@@ -697,9 +697,10 @@ at::Tensor wgrad_frag_bf16(const at::Tensor& AT, const at::Tensor& BT,
               "wgrad_frag: unsupported tile config");
   TORCH_CHECK(N % (nt_w * 128) == 0 && K % (kt_w * 32) == 0,
               "wgrad_frag: N/K not divisible by block shape");
-  auto dW = at::empty({N, K}, AT.options().dtype(at::kFloat));
+  auto dW = at::zeros({N, K}, AT.options().dtype(at::kFloat));
   if (mchunks > 0) {
-    // Per-slab fp32 partials + second-stage reduce (no atomics).
+    // Per-slab fp32 partials + split-slab reduce (atomic combine over
+    // <= a handful of writers per line).
     const int64_t nslabs = wgrad_frag_nslabs(mchunks, (int32_t)N,
                                              (int32_t)K, (int32_t)nt_w,
                                              (int32_t)kt_w);
@@ -710,8 +711,6 @@ at::Tensor wgrad_frag_bf16(const at::Tensor& AT, const at::Tensor& BT,
                       (int32_t)kt_w, current_stream());
     launch_slab_reduce(part.data_ptr<float>(), dW.data_ptr<float>(),
                        N * K, nslabs, current_stream());
-  } else {
-    dW.zero_();
   }
   return dW;
 }

@@ -149,18 +149,27 @@ __global__ void __launch_bounds__(256, 2) wgrad_frag_kernel(
 }
 
 // out[i] = sum_s part[s][i] — the slab-partial reduction (fp32, float4).
+// 2-D grid: y splits the slab range so small outputs still fill the
+// chip; the few split partials combine with fp32 atomics (<= NSPLIT
+// writers per line — negligible contention). out must be zeroed.
 typedef __attribute__((__vector_size__(4 * sizeof(float)))) float wf_f32x4;
 __global__ void __launch_bounds__(256) slab_reduce_kernel(
     const float* __restrict__ part, float* __restrict__ out, int64_t nk4,
-    int64_t nslabs) {
+    int64_t nslabs, int64_t s1) {
   const int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x;
   if (i >= nk4) return;
+  const int64_t lo = (int64_t)blockIdx.y * s1;
+  const int64_t hi = min(lo + s1, nslabs);
   wf_f32x4 s = {0.f, 0.f, 0.f, 0.f};
-  for (int64_t sl = 0; sl < nslabs; sl++) {
+  for (int64_t sl = lo; sl < hi; sl++) {
     s += __builtin_nontemporal_load(
         reinterpret_cast<const wf_f32x4*>(part + (sl * nk4 + i) * 4));
   }
-  *reinterpret_cast<wf_f32x4*>(out + i * 4) = s;
+  float* o = out + i * 4;
+  unsafeAtomicAdd(&o[0], s[0]);
+  unsafeAtomicAdd(&o[1], s[1]);
+  unsafeAtomicAdd(&o[2], s[2]);
+  unsafeAtomicAdd(&o[3], s[3]);
 }
 
 // layer configs: 1 -> NT_W=2,KT_W=4 (dW1 [512,128pad]); 2 -> 1,8
@@ -171,9 +180,15 @@ int64_t wgrad_frag_nslabs(int64_t mchunks, int32_t N, int32_t K,
 void launch_slab_reduce(const float* part, float* out, int64_t nk,
                         int64_t nslabs, hipStream_t stream) {
   const int64_t nk4 = nk / 4;
-  const int64_t grid = (nk4 + 255) / 256;
-  hipLaunchKernelGGL(slab_reduce_kernel, dim3((uint32_t)grid), dim3(256),
-                     0, stream, part, out, nk4, nslabs);
+  const int64_t gx = (nk4 + 255) / 256;
+  // aim for >= 512 workgroups total
+  int64_t nsplit = (512 + gx - 1) / gx;
+  if (nsplit > nslabs) nsplit = nslabs;
+  const int64_t s1 = (nslabs + nsplit - 1) / nsplit;
+  nsplit = (nslabs + s1 - 1) / s1;
+  hipLaunchKernelGGL(slab_reduce_kernel,
+                     dim3((uint32_t)gx, (uint32_t)nsplit), dim3(256), 0,
+                     stream, part, out, nk4, nslabs, s1);
 }
 
 static int64_t frag_target(int32_t K, int32_t nt_w) {
