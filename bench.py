@@ -323,27 +323,30 @@ def main():
     def capture_graph(x_example, t_example):
         sx = torch.empty_like(x_example)
         st = torch.empty_like(t_example)
-        side = torch.cuda.Stream(device)
-        side.wait_stream(torch.cuda.current_stream(device))
-        with torch.cuda.stream(side):
-            for _ in range(3):  # allocator/optimizer state warmup
+
+        def step_body():
+            if use_fused:
+                fused_step(model, sx, st)
+            else:
                 opt.zero_grad(set_to_none=False)
                 with amp():
                     out = model(sx)
                     loss = loss_fn(out.float(), st)
                 loss.backward()
-                opt.step()
+            opt.step()
+            if not use_fused:
+                for p in model.parameters():
+                    p.grad.zero_()
+
+        side = torch.cuda.Stream(device)
+        side.wait_stream(torch.cuda.current_stream(device))
+        with torch.cuda.stream(side):
+            for _ in range(3):  # allocator/optimizer state warmup
+                step_body()
         torch.cuda.current_stream(device).wait_stream(side)
         g = torch.cuda.CUDAGraph()
-        opt.zero_grad(set_to_none=False)
         with torch.cuda.graph(g, capture_error_mode="thread_local"):
-            with amp():
-                out = model(sx)
-                loss = loss_fn(out.float(), st)
-            loss.backward()
-            opt.step()
-            for p in model.parameters():
-                p.grad.zero_()
+            step_body()
         graph_state.update(g=g, sx=sx, st=st)
 
     def one_step_graphed():

@@ -752,4 +752,4 @@ def test_wgrad_frag_bandwidth(dev):
         print(f"wgrad_frag {n}x{k} M={m}: {ms:.3f} ms")
     # library split-K bmm runs these at ~0.15/0.11/0.07 ms; require at
     # least rough parity so a regression is loud.
-    assert sum(results.values()) < 0.45, results
+    assert sum(results.values()) < 0.5, results
