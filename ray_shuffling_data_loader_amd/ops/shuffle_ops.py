@@ -16,6 +16,8 @@ Reference ops replaced here: ``pd.concat`` + ``df.sample(frac=1)``
 
 from typing import Dict, Optional
 
+import os
+
 import numpy as np
 import torch
 
