@@ -110,32 +110,20 @@ __global__ void __launch_bounds__(256, 2) wgrad_frag_kernel(
   // 3-deep prefetch: at depth 1 only ~6 KB/wave was in flight and the
   // kernel ran latency-bound at ~half the stream roofline (depth 4
   // spills the <1,8> config's registers).
-  // Pin the scheduler to the intended software pipeline: alternate
-  // [NT_W+KT_W VMEM reads][NT_W*KT_W MFMAs] groups. Without this the
-  // backend packs two chunks' loads together and the last MFMA of each
-  // batch gets s_waitcnt vmcnt(0), draining the prefetches — one full
-  // HBM latency exposed per two chunks (measured 2-3x off roofline).
-#define WF_PIN()                                                           \
-  __builtin_amdgcn_sched_group_barrier(0x020, NT_W + KT_W, 0);             \
-  __builtin_amdgcn_sched_group_barrier(0x008, NT_W * KT_W, 0)
   WF_LOAD(a0, b0, 0);
   if (1 < iters) WF_LOAD(a1, b1, 1);
   int64_t i = 0;
   while (i + 3 <= iters) {
     WF_LOAD(a2, b2, i + 2);
     WF_MFMA(a0, b0);
-    WF_PIN();
     if (i + 3 < iters) WF_LOAD(a0, b0, i + 3);
     WF_MFMA(a1, b1);
-    WF_PIN();
     if (i + 4 < iters) WF_LOAD(a1, b1, i + 4);
     WF_MFMA(a2, b2);
-    WF_PIN();
     i += 3;
   }
   if (i < iters) WF_MFMA(a0, b0);
   if (i + 1 < iters) WF_MFMA(a1, b1);
-#undef WF_PIN
 #undef WF_LOAD
 #undef WF_MFMA
 
