@@ -35,6 +35,13 @@ namespace rsdl {
 typedef __attribute__((__vector_size__(8 * sizeof(short)))) short bc_bf16x8;
 typedef __attribute__((__vector_size__(16 * sizeof(float)))) float bc_f32x16;
 typedef __attribute__((__vector_size__(4 * sizeof(unsigned int)))) unsigned int bc_u32x4;
+typedef __attribute__((__vector_size__(2 * sizeof(float)))) float bc_f32x2;
+
+__device__ __forceinline__ uint32_t bc_cvt_pk_bf16(bc_f32x2 v) {
+  uint32_t p;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(p) : "v"(v[0]), "v"(v[1]));
+  return p;
+}
 
 #define BC_MT 32
 #define BC_MTILES (BC_MT / 32)
@@ -159,27 +166,27 @@ __device__ void bc_layer(const short* __restrict__ dz_src,
   for (int nt = 0; nt < NT; nt++) {
     const int32_t n = n_base + nt * 32 + ml;
     const uint32_t mw = mask_row[n];
-    float vals[16];
-    float colsum = 0.f;
+    uint32_t p[8];
+    bc_f32x2 colsum2 = {0.f, 0.f};
     #pragma unroll
-    for (int reg = 0; reg < 16; reg++) {
-      const int32_t mrow = (reg & 3) + 8 * (reg >> 2) + 4 * h;
-      const float v = ((mw >> mrow) & 1u) ? acc[nt][reg] : 0.f;
-      vals[reg] = v;
-      colsum += v;
-      if (WRITE_LDS) dst_lds[mrow * DST_S + n] = bc_f2b(v);
+    for (int q = 0; q < 8; q++) {
+      const int32_t mrow = ((2 * q) & 3) + 8 * (q >> 1) + 4 * h;
+      const bc_f32x2 v2 = {
+          ((mw >> mrow) & 1u) ? acc[nt][2 * q] : 0.f,
+          ((mw >> (mrow + 1)) & 1u) ? acc[nt][2 * q + 1] : 0.f};
+      colsum2 += v2;
+      const uint32_t pk = bc_cvt_pk_bf16(v2);
+      p[q] = pk;
+      if (WRITE_LDS) {
+        dst_lds[mrow * DST_S + n] = (short)(pk & 0xFFFFu);
+        dst_lds[(mrow + 1) * DST_S + n] = (short)(pk >> 16);
+      }
     }
+    float colsum = colsum2[0] + colsum2[1];
     colsum += __shfl_xor(colsum, 32);
     if (lane < 32) db_out[n] = colsum;
     // Transposed fragment emission, packed-pair exchange (see
     // fwd_chain fc_layer EMIT_T).
-    uint32_t p[8];
-    #pragma unroll
-    for (int q = 0; q < 8; q++) {
-      const short lo = bc_f2b(vals[2 * q]);
-      const short hi = bc_f2b(vals[2 * q + 1]);
-      p[q] = ((uint32_t)(uint16_t)hi << 16) | (uint16_t)lo;
-    }
     uint32_t rx[4];
     #pragma unroll
     for (int j = 0; j < 2; j++) {
@@ -200,7 +207,7 @@ __device__ void bc_layer(const short* __restrict__ dz_src,
   }
 }
 
-__global__ void __launch_bounds__(256) bwd_chain_kernel(
+__global__ void __launch_bounds__(256, 3) bwd_chain_kernel(
     const short* __restrict__ dy,     // [M,1] bf16 (head grad)
     const short* __restrict__ a3,     // [M,128] saved activations
     const uint32_t* __restrict__ mask1,  // [m_tiles][512] relu-mask words

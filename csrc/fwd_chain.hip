@@ -52,6 +52,19 @@ typedef __attribute__((__vector_size__(8 * sizeof(short)))) short fc_bf16x8;
 typedef __attribute__((__vector_size__(16 * sizeof(float)))) float fc_f32x16;
 typedef __attribute__((__vector_size__(4 * sizeof(unsigned int)))) unsigned int fc_u32x4;
 typedef __attribute__((__vector_size__(2 * sizeof(unsigned int)))) unsigned int fc_u32x2;
+typedef __attribute__((__vector_size__(2 * sizeof(float)))) float fc_f32x2;
+
+// Packed pair helpers (one VALU op per TWO elements on gfx950).
+__device__ __forceinline__ fc_f32x2 fc_pk_max0(fc_f32x2 v) {
+  // no packed f32 max on gfx950; two v_max_f32
+  fc_f32x2 r = {fmaxf(v[0], 0.f), fmaxf(v[1], 0.f)};
+  return r;
+}
+__device__ __forceinline__ uint32_t fc_cvt_pk_bf16(fc_f32x2 v) {
+  uint32_t p;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(p) : "v"(v[0]), "v"(v[1]));
+  return p;
+}
 
 #define FC_MT 32            // rows per workgroup slab
 #define FC_MTILES (FC_MT / 32)  // 32-row MFMA m-tiles per slab
@@ -172,7 +185,10 @@ __device__ void fc_layer(const short* __restrict__ src_lds,
   }
   // Epilogue: bias + (relu) + cast + store the D fragments into the dst
   // tile at [mrow][n] (b16 column writes; DST_S padding spreads banks);
-  // with EMIT_T also assemble the transposed fragment runs + mask word.
+  // with EMIT_T also emit the transposed fragment runs + mask word.
+  // All arithmetic runs on packed row pairs (v_pk_add/max_f32 +
+  // v_cvt_pk_bf16_f32): regs 2q/2q+1 are consecutive rows, and the
+  // packed bf16 word IS the emission payload.
   static_assert(!EMIT_T || FC_MTILES == 1, "EMIT_T assumes 32-row slabs");
   const int32_t h = lane >> 5;
   #pragma unroll
@@ -180,32 +196,34 @@ __device__ void fc_layer(const short* __restrict__ src_lds,
     #pragma unroll
     for (int nt = 0; nt < NT; nt++) {
       const int32_t n = n_base + nt * 32 + ml;
-      const float bv = bias[n];
-      float vals[16];
+      const fc_f32x2 bv2 = {bias[n], bias[n]};
+      uint32_t p[8];
       #pragma unroll
-      for (int reg = 0; reg < 16; reg++) {
+      for (int q = 0; q < 8; q++) {
         const int32_t mrow =
-            mt * 32 + (reg & 3) + 8 * (reg >> 2) + 4 * h;
-        float v = acc[mt][nt][reg] + bv;
-        if (RELU) v = v > 0.f ? v : 0.f;
-        dst_lds[mrow * DST_S + n] = fc_f2b(v);
-        vals[reg] = v;
+            mt * 32 + ((2 * q) & 3) + 8 * (q >> 1) + 4 * h;
+        fc_f32x2 v2 = {acc[mt][nt][2 * q], acc[mt][nt][2 * q + 1]};
+        v2 += bv2;
+        if (RELU) v2 = fc_pk_max0(v2);
+        const uint32_t pk = fc_cvt_pk_bf16(v2);
+        p[q] = pk;
+        dst_lds[mrow * DST_S + n] = (short)(pk & 0xFFFFu);
+        dst_lds[(mrow + 1) * DST_S + n] = (short)(pk >> 16);
       }
       if (EMIT_T) {
-        // Pack the bf16 values in row pairs, then exchange the
-        // half-rows as packed ints: 4 shfls, no post-exchange converts.
-        // Lane h=0 assembles rows {0-7},{16-23}; h=1 {8-15},{24-31}.
-        uint32_t p[8];
+        // Mask word for column n (bit i = bf16 value of row i > 0): a
+        // positive nonzero bf16 is 0 < bits < 0x8000.
         uint32_t w = 0;
         #pragma unroll
         for (int q = 0; q < 8; q++) {
-          const short lo = fc_f2b(vals[2 * q]);
-          const short hi = fc_f2b(vals[2 * q + 1]);
-          p[q] = ((uint32_t)(uint16_t)hi << 16) | (uint16_t)lo;
           const int32_t mrow = ((2 * q) & 3) + 8 * (q >> 1) + 4 * h;
-          w |= (uint32_t)(lo > 0) << mrow;
-          w |= (uint32_t)(hi > 0) << (mrow + 1);
+          const uint32_t lo = p[q] & 0xFFFFu;
+          const uint32_t hi = p[q] >> 16;
+          w |= (uint32_t)(lo - 1u < 0x7FFFu) << mrow;
+          w |= (uint32_t)(hi - 1u < 0x7FFFu) << (mrow + 1);
         }
+        // Exchange the half-rows as packed ints (4 shfls); lane h=0
+        // assembles rows {0-7},{16-23}, h=1 {8-15},{24-31}.
         uint32_t rx[4];
         #pragma unroll
         for (int j = 0; j < 2; j++) {
@@ -220,7 +238,6 @@ __device__ void fc_layer(const short* __restrict__ src_lds,
                                h == 0 ? p[5] : rx[3],
                                h == 0 ? rx[2] : p[6],
                                h == 0 ? rx[3] : p[7]};
-        // run0 -> (mc_local 0, h'=h); run1 -> (mc_local 1, h'=h)
         const int64_t nt_g = (int64_t)(n_base + nt * 32) >> 5;
         short* blk0 = at_out + ((nt_g * mchunks + mc0) * 512) + h * 256 +
                       ml * 8;
@@ -230,7 +247,6 @@ __device__ void fc_layer(const short* __restrict__ src_lds,
                                     reinterpret_cast<fc_u32x4*>(blk0));
         __builtin_nontemporal_store(run1,
                                     reinterpret_cast<fc_u32x4*>(blk1));
-        // Mask word for column n (bit i = bf16 value of row i > 0).
         w |= __shfl_xor(w, 32);
         if (h == 0) mask_row[n] = w;
       }
