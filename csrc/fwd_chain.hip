@@ -418,6 +418,80 @@ void launch_swizzle_x(const void* x, void* out, int64_t M,
                      reinterpret_cast<short*>(out), M, total);
 }
 
+// Combined swizzle: one pass over x emits BOTH the forward fragment
+// layout (xs: [mt][7][2][32][8] over k-contraction) and the wgrad
+// fragment layout (xt: [4][mchunks][2][32][8] over m-contraction).
+// One thread per 16-B fwd block (8 k for one row m); the wgrad-side
+// elements it holds are scattered, so the wgrad half works via LDS: the
+// workgroup stages a [32 m x 128 k] tile and re-emits it transposed.
+__global__ void __launch_bounds__(256) swizzle_x_both_kernel(
+    const short* __restrict__ x, short* __restrict__ xs,
+    short* __restrict__ xt, int64_t M, int64_t mtiles) {
+  __shared__ short tile[32 * 136];  // [m][k] stride 136 (16-B rows)
+  const int64_t mt = blockIdx.x;
+  if (mt >= mtiles) return;
+  const int32_t tid = threadIdx.x;
+  const int64_t m0 = mt * 32;
+  // stage: 32 rows x 100 cols, 8-B vectors (25 uint2 slots per row)
+  for (int32_t u = tid; u < 32 * 25; u += 256) {
+    const int32_t m = u / 25;
+    const int32_t c = (u % 25) * 4;
+    fc_u32x2 v = {0, 0};
+    if (m0 + m < M) {
+      v = __builtin_nontemporal_load(
+          reinterpret_cast<const fc_u32x2*>(&x[(m0 + m) * FC_K0 + c]));
+    }
+    *reinterpret_cast<fc_u32x2*>(&tile[m * 136 + c]) = v;
+  }
+  // zero-pad cols 100..127 (fwd uses 100..111; wgrad k-tiles reach 127)
+  for (int32_t u = tid; u < 32 * 7; u += 256) {
+    const int32_t m = u / 7;
+    const int32_t c = 100 + (u % 7) * 4;
+    fc_u32x2 z = {0, 0};
+    *reinterpret_cast<fc_u32x2*>(&tile[m * 136 + c]) = z;
+  }
+  __syncthreads();
+  // fwd layout: block (kc, h, ml) -> 8 k of row ml
+  for (int32_t u = tid; u < 7 * 2 * 32; u += 256) {
+    const int32_t ml = u & 31;
+    const int32_t h = (u >> 5) & 1;
+    const int32_t kc = u >> 6;
+    const int32_t k0 = kc * 16 + h * 8;
+    fc_u32x4 v = *reinterpret_cast<const fc_u32x4*>(&tile[ml * 136 + k0]);
+    __builtin_nontemporal_store(
+        v, reinterpret_cast<fc_u32x4*>(
+               &xs[(mt * 14 + (int64_t)u / 32) * 256 + ml * 8]));
+  }
+  // wgrad layout: block (kt, mc_local, h, ml) -> 8 m of column kt*32+ml
+  const int64_t mchunks = mtiles * 2;
+  for (int32_t u = tid; u < 4 * 2 * 2 * 32; u += 256) {
+    const int32_t ml = u & 31;
+    const int32_t h = (u >> 5) & 1;
+    const int32_t mcl = (u >> 6) & 1;
+    const int32_t kt = u >> 7;
+    const int32_t k = kt * 32 + ml;
+    const int32_t mb = mcl * 16 + h * 8;
+    short v[8];
+    #pragma unroll
+    for (int j = 0; j < 8; j++) v[j] = tile[(mb + j) * 136 + k];
+    __builtin_nontemporal_store(
+        *reinterpret_cast<fc_u32x4*>(v),
+        reinterpret_cast<fc_u32x4*>(
+            &xt[(((int64_t)kt * mchunks + mt * 2 + mcl) * 2 + h) * 256 +
+                ml * 8]));
+  }
+}
+
+void launch_swizzle_x_both(const void* x, void* xs, void* xt, int64_t M,
+                           hipStream_t stream) {
+  const int64_t mtiles = (M + FC_MT - 1) / FC_MT;
+  hipLaunchKernelGGL(swizzle_x_both_kernel, dim3((uint32_t)mtiles),
+                     dim3(256), 0, stream,
+                     reinterpret_cast<const short*>(x),
+                     reinterpret_cast<short*>(xs),
+                     reinterpret_cast<short*>(xt), M, mtiles);
+}
+
 // x [M,100] bf16 -> wgrad fragment-major x^T:
 // [128/32][mchunks][2][32][8] with zero pads (cols 100..127, rows >= M).
 // One thread per 16-B out block = 8 consecutive m's of one column k;

@@ -73,6 +73,8 @@ void launch_swizzle_x(const void* x, void* out, int64_t M,
                       hipStream_t stream);
 void launch_swizzle_xt(const void* x, void* out, int64_t M,
                        hipStream_t stream);
+void launch_swizzle_x_both(const void* x, void* xs, void* xt, int64_t M,
+                           hipStream_t stream);
 void launch_fwd_chain(const void* x0s, const void* W1, const float* b1,
                       const void* W2, const float* b2, const void* W3,
                       const float* b3, const void* w4, const float* b4,
@@ -481,7 +483,8 @@ std::vector<at::Tensor> fwd_chain_bf16(
     const at::Tensor& x, const at::Tensor& W1, const at::Tensor& b1,
     const at::Tensor& W2, const at::Tensor& b2, const at::Tensor& W3,
     const at::Tensor& b3, const at::Tensor& w4, const at::Tensor& b4,
-    const c10::optional<at::Tensor>& target) {
+    const c10::optional<at::Tensor>& target,
+    const c10::optional<at::Tensor>& xt_out) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
                   x.dim() == 2 && x.size(1) == 100 && x.is_contiguous(),
               "fwd_chain: x must be contiguous bf16 [M,100]");
@@ -553,7 +556,13 @@ std::vector<at::Tensor> fwd_chain_bf16(
     // real rows).
     const int64_t Mp = (M + 31) / 32 * 32;
     auto xs = at::empty({Mp * 112}, x.options());
-    launch_swizzle_x(x.data_ptr(), xs.data_ptr(), M, current_stream());
+    if (xt_out.has_value()) {
+      // one pass over x emits both the forward and the wgrad layouts
+      launch_swizzle_x_both(x.data_ptr(), xs.data_ptr(),
+                            xt_out->data_ptr(), M, current_stream());
+    } else {
+      launch_swizzle_x(x.data_ptr(), xs.data_ptr(), M, current_stream());
+    }
     launch_fwd_chain(xs.data_ptr(), W1s.data_ptr(), b1f.data_ptr<float>(),
                      W2s.data_ptr(), b2f.data_ptr<float>(), W3s.data_ptr(),
                      b3f.data_ptr<float>(), w4.data_ptr(),
@@ -744,7 +753,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fwd_chain_bf16", &rsdl::fwd_chain_bf16, py::arg("x"),
         py::arg("W1"), py::arg("b1"), py::arg("W2"), py::arg("b2"),
         py::arg("W3"), py::arg("b3"), py::arg("w4"), py::arg("b4"),
-        py::arg("target") = c10::nullopt);
+        py::arg("target") = c10::nullopt,
+        py::arg("xt_out") = c10::nullopt);
   m.def("bwd_chain_bf16", &rsdl::bwd_chain_bf16);
   m.attr("DT_F32") = (int)rsdl::DT_F32;
   m.attr("DT_F64") = (int)rsdl::DT_F64;

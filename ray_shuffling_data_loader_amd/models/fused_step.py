@@ -74,9 +74,14 @@ def fused_step(
     b1, b2, b3, b4 = (m.bias.detach() for m in lin)
     # Forward chain: a1/a2 exist only TRANSPOSED (wgrad fragment-major)
     # plus 32-bit relu-mask words; loss + dy fold into the head epilogue.
+    # The x swizzle emits BOTH x layouts (fwd fragments + wgrad x^T) in
+    # one pass over x.
+    mchunks = 2 * ((M + 31) // 32)
+    xt = torch.empty(4 * mchunks * 512, dtype=torch.bfloat16,
+                     device=x.device)
     a1t, mask1, a2t, mask2, a3, out, dyb, loss_part = hip.fwd_chain_bf16(
         x, buf["W1p"], b1, buf["W2"], b2, buf["W3"], b3, buf["w4"], b4,
-        target=target,
+        target=target, xt_out=xt,
     )
     loss = loss_part.sum() / M
     # Backward chain: consumes the masks (never the activations), emits
@@ -85,9 +90,7 @@ def fused_step(
         dyb, a3, mask1, mask2, buf["w4"], buf["W3"], buf["W2"]
     )
     # Weight grads: fragment-major MFMA wgrad kernel (csrc/wgrad_frag.hip)
-    # reading the transposed fragments both producers emitted.
-    mchunks = 2 * ((M + 31) // 32)
-    xt = hip.swizzle_xt_bf16(x)
+    # reading the transposed fragments the producers emitted.
     dw1 = wgrad_frag(dz1t, xt, 512, 128, mchunks)[:, :100].contiguous()
     dw2 = wgrad_frag(dz2t, a1t, 256, 512, mchunks)
     dw3 = wgrad_frag(dz3t, a2t, 128, 256, mchunks)

@@ -251,7 +251,13 @@ def test_fused_step_glue_cpu(monkeypatch):
         fragment-major a/dz tensors, relu-mask words, fused loss."""
 
         @staticmethod
-        def fwd_chain_bf16(x, W1, b1, W2, b2, W3, b3, w4, b4, target=None):
+        def fwd_chain_bf16(x, W1, b1, W2, b2, W3, b3, w4, b4, target=None,
+                           xt_out=None):
+            if xt_out is not None:
+                m = x.shape[0]
+                mp = (m + 31) // 32 * 32
+                xt_out.copy_(t_frag_swizzle(
+                    torch.nn.functional.pad(x, (0, 28, 0, mp - m))))
             if W1.shape[1] == 112:
                 W1 = W1[:, :100]
             a1 = torch.relu(x.float() @ W1.float().t() + b1.float())
@@ -359,7 +365,13 @@ def test_fused_step_flat_grad_views_cpu(monkeypatch):
 
     class FakeHip:
         @staticmethod
-        def fwd_chain_bf16(x, W1, b1, W2, b2, W3, b3, w4, b4, target=None):
+        def fwd_chain_bf16(x, W1, b1, W2, b2, W3, b3, w4, b4, target=None,
+                           xt_out=None):
+            if xt_out is not None:
+                m = x.shape[0]
+                mp = (m + 31) // 32 * 32
+                xt_out.copy_(t_frag_swizzle(
+                    torch.nn.functional.pad(x, (0, 28, 0, mp - m))))
             if W1.shape[1] == 112:
                 W1 = W1[:, :100]
             a1 = torch.relu(x.float() @ W1.float().t() + b1.float())

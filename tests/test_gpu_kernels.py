@@ -585,12 +585,16 @@ def test_fwd_chain_matches_eager(dev):
         assert err.max() <= 0.12 * scale + 0.05, (
             name, err.max().item(), scale.item())
 
-    # Fused MSE epilogue (target given): dyb + loss partials.
+    # Fused MSE epilogue (target given): dyb + loss partials; also the
+    # combined swizzle's wgrad-layout x^T output vs the standalone kernel.
     tgt = torch.randn(M, 1, device=dev)
+    mchunks = 2 * ((M + 31) // 32)
+    xt = torch.empty(4 * mchunks * 512, dtype=torch.bfloat16, device=dev)
     _, _, _, _, a3b, outb, dyb, loss_part = hip.fwd_chain_bf16(
         x, Ws[0], bs[0], Ws[1], bs[1], Ws[2], bs[2],
-        Ws[3].flatten(), bs[3], target=tgt,
+        Ws[3].flatten(), bs[3], target=tgt, xt_out=xt,
     )
+    assert torch.equal(xt, hip.swizzle_xt_bf16(x))
     assert torch.equal(outb, out)
     diff = outb.float() - tgt
     ref_dy = (2.0 / M) * diff
