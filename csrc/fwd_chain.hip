@@ -471,14 +471,15 @@ __global__ void __launch_bounds__(256) swizzle_x_both_kernel(
     const int32_t kt = u >> 7;
     const int32_t k = kt * 32 + ml;
     const int32_t mb = mcl * 16 + h * 8;
-    short v[8];
+    fc_u32x4 pack;
+    short* vp = reinterpret_cast<short*>(&pack);
     #pragma unroll
-    for (int j = 0; j < 8; j++) v[j] = tile[(mb + j) * 136 + k];
+    for (int j = 0; j < 8; j++) vp[j] = tile[(mb + j) * 136 + k];
     __builtin_nontemporal_store(
-        *reinterpret_cast<fc_u32x4*>(v),
-        reinterpret_cast<fc_u32x4*>(
-            &xt[(((int64_t)kt * mchunks + mt * 2 + mcl) * 2 + h) * 256 +
-                ml * 8]));
+        pack, reinterpret_cast<fc_u32x4*>(
+                  &xt[(((int64_t)kt * mchunks + mt * 2 + mcl) * 2 + h) *
+                          256 +
+                      ml * 8]));
   }
 }
 
