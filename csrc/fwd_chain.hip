@@ -508,7 +508,8 @@ __global__ void __launch_bounds__(256) swizzle_xt_kernel(
   int64_t r = b >> 5;
   const int32_t h = (int32_t)(r & 1);
   r >>= 1;
-  const int64_t mchunks = total_blocks >> 9;  // total = 4*mchunks*64
+  // total = 4 (kt) * mchunks * 2 (h) * 32 (ml) = 256 * mchunks
+  const int64_t mchunks = total_blocks >> 8;
   const int64_t mc = r % mchunks;
   const int32_t kt = (int32_t)(r / mchunks);
   const int32_t k = kt * 32 + ml;
