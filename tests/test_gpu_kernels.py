@@ -594,7 +594,22 @@ def test_fwd_chain_matches_eager(dev):
         x, Ws[0], bs[0], Ws[1], bs[1], Ws[2], bs[2],
         Ws[3].flatten(), bs[3], target=tgt, xt_out=xt,
     )
-    assert torch.equal(xt, hip.swizzle_xt_bf16(x))
+    # Both x^T producers must match the torch layout oracle EXACTLY —
+    # comparing the two kernels against each other only proves they share
+    # bugs (a >>9/>>8 mchunks slip in the standalone survived the fused
+    # parity tolerance until this check existed).
+    from ray_shuffling_data_loader_amd.ops.shuffle_ops import (
+        t_frag_swizzle,
+    )
+
+    mp = (M + 31) // 32 * 32
+    xt_oracle = t_frag_swizzle(
+        torch.nn.functional.pad(x, (0, 28, 0, mp - M))
+    )
+    assert torch.equal(xt, xt_oracle), "combined swizzle x^T layout"
+    assert torch.equal(
+        hip.swizzle_xt_bf16(x), xt_oracle
+    ), "standalone swizzle x^T layout"
     assert torch.equal(outb, out)
     diff = outb.float() - tgt
     ref_dy = (2.0 / M) * diff
