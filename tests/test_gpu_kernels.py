@@ -708,10 +708,13 @@ def test_fused_step_matches_eager(dev):
     ):
         assert p.grad is not None and q.grad is not None, n
         err = (p.grad - q.grad).abs()
-        tol = 0.05 * q.grad.abs().mean().clamp(min=1e-5) + 5e-4
-        assert err.max() <= tol.item() * 20 or torch.allclose(
-            p.grad, q.grad, rtol=0.1, atol=1e-3
-        ), (n, err.max().item(), q.grad.abs().mean().item())
+        # The previous 20x-slop tolerance absorbed a REAL dW1 corruption
+        # (the swizzle_xt mchunks bug). Max-error vs the autocast eager
+        # grads: bf16 rounding + reduction-order differences measure
+        # well under 8% of the mean magnitude; layout bugs produce O(1).
+        tol = 0.08 * q.grad.abs().mean().clamp(min=1e-5) + 1e-3
+        assert err.max() <= tol.item(), (
+            n, err.max().item(), q.grad.abs().mean().item())
 
 
 def test_wgrad_frag_matches_reference(dev):
