@@ -49,10 +49,16 @@ def _weight_buffers(model: TabularMLP, lin):
             "w4": torch.empty(128, **bf),
         }
         model._fused_buf = buf
-    buf["W1p"][:, :100].copy_(lin[0].weight.detach())
-    buf["W2"].copy_(lin[1].weight.detach())
-    buf["W3"].copy_(lin[2].weight.detach())
-    buf["w4"].copy_(lin[3].weight.detach().view(-1))
+    # One multi-tensor cast-copy instead of four kernels.
+    torch._foreach_copy_(
+        [buf["W1p"][:, :100], buf["W2"], buf["W3"], buf["w4"]],
+        [
+            lin[0].weight.detach(),
+            lin[1].weight.detach(),
+            lin[2].weight.detach(),
+            lin[3].weight.detach().view(-1),
+        ],
+    )
     return buf
 
 
