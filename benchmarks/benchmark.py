@@ -129,6 +129,7 @@ def run_trial(args, filenames, trial_idx):
             args.num_trainers,
             stats_collector=stats,
             source_cache=args.source_cache,
+            reader_threads=args.reader_threads,
             seed=args.seed + trial_idx if args.seed is not None else None,
         )
     print(f"Trial {trial_idx} done in {duration:.3f}s")
@@ -178,6 +179,8 @@ def parse_args():
                    help="tabular = reference DATA_SPEC; float100 = the "
                    "MI355X flagship 100-float32-column shape")
     p.add_argument("--num-float-cols", type=int, default=100)
+    p.add_argument("--reader-threads", type=int, default=8,
+                   help="ingest decode threads per engine")
     p.add_argument("--source-cache", type=str, default="none",
                    choices=["auto", "device", "host", "none"],
                    help="'none' re-reads Parquet every epoch like the "

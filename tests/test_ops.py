@@ -132,3 +132,46 @@ def test_pack_with_scatter_perm():
     packed = pack_columns(cols, schema, perm=perm)
     out = unpack_permute(packed, schema)
     assert torch.equal(out["x"][perm], cols["x"])
+
+
+def test_t_frag_swizzle_roundtrip_shapes():
+    """Round-trip through the wgrad fragment layout at awkward shapes."""
+    import torch
+
+    from ray_shuffling_data_loader_amd.ops.shuffle_ops import (
+        t_frag_swizzle,
+        t_frag_unswizzle,
+    )
+
+    for m, c in [(16, 32), (48, 64), (100, 128), (333, 256), (1024, 512)]:
+        t = torch.randn(m, c).bfloat16()
+        flat = t_frag_swizzle(t)
+        mp = (m + 15) // 16 * 16
+        assert flat.numel() == (c // 32) * (mp // 16) * 512
+        back = t_frag_unswizzle(flat, m, c)
+        assert torch.equal(back, t), (m, c)
+        # pad rows must be zero (wgrad contributions)
+        if mp != m:
+            full = t_frag_unswizzle(flat, mp, c)
+            assert torch.all(full[m:] == 0), (m, c)
+
+
+def test_relu_mask_words_bits():
+    import torch
+
+    from ray_shuffling_data_loader_amd.ops.shuffle_ops import (
+        relu_mask_words,
+    )
+
+    for m, n in [(32, 32), (70, 64), (129, 96)]:
+        a = torch.randn(m, n)
+        w = relu_mask_words(a)
+        mt = (m + 31) // 32
+        assert w.shape == (mt, n) and w.dtype == torch.int32
+        bits = (w.to(torch.int64) & 0xFFFFFFFF).view(mt, 1, n)
+        sh = torch.arange(32).view(1, 32, 1)
+        got = ((bits >> sh) & 1).reshape(mt * 32, n)[:m].bool()
+        assert torch.equal(got, a > 0), (m, n)
+        # pad bits zero
+        rest = ((bits >> sh) & 1).reshape(mt * 32, n)[m:]
+        assert torch.all(rest == 0)
