@@ -236,8 +236,10 @@ __global__ void __launch_bounds__(256, 3) bwd_chain_kernel(
   const int32_t wave = tid >> 6;
   const int32_t lane = tid & 63;
   const int64_t mc0 = (int64_t)blockIdx.x * 2;
+  // stride padded to a 4-multiple for the host-side float4 slab reduce
   float* part =
-      &db_part[(int64_t)blockIdx.x * (BC_N1 + BC_N2 + BC_N3 + 1 + 256)];
+      &db_part[(int64_t)blockIdx.x *
+               ((BC_N1 + BC_N2 + BC_N3 + 1 + 256 + 3) / 4 * 4)];
 
   bc_load_tile<BC_N3, BC_S3>(a3, t3, m0, M, tid);
   if (tid < BC_MT) {

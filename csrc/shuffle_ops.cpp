@@ -642,10 +642,10 @@ std::vector<at::Tensor> bwd_chain_bf16(
   auto dz3t = at::empty({4 * mchunks * 512}, dy.options());
   const int64_t grid = bwd_chain_grid(M);
   constexpr int64_t kPartW = 512 + 256 + 128 + 1 + 256;
-  // Every block writes its whole db_part row (bc_bias_partial covers all
-  // columns incl. db4 and the dW4 partials), so empty() is safe — no
-  // memset kernel per call.
-  auto db_part = at::empty({std::max<int64_t>(grid, 1), kPartW},
+  // Width padded to a 4-multiple for the float4 slab reduce; the 3 pad
+  // columns are never read back.
+  constexpr int64_t kPartWPad = (kPartW + 3) / 4 * 4;
+  auto db_part = at::empty({std::max<int64_t>(grid, 1), kPartWPad},
                            dy.options().dtype(at::kFloat));
   if (M == 0) db_part.zero_();
   if (M > 0) {
@@ -655,12 +655,13 @@ std::vector<at::Tensor> bwd_chain_bf16(
                      dz3t.data_ptr(), db_part.data_ptr<float>(), M,
                      current_stream());
   }
-  // Column reduction as a GEMV (ones^T @ db_part): hipBLASLt streams the
-  // 28 MB slab near roofline where the generic [rows,1153] column-reduce
-  // kernel ran at ~0.3 TB/s.
-  auto ones =
-      at::ones({1, db_part.size(0)}, dy.options().dtype(at::kFloat));
-  auto db = at::mm(ones, db_part).reshape({kPartW});
+  // Column reduction with the split-slab reduce kernel (the GEMV ran at
+  // ~0.9 TB/s; this streams the 28 MB slab near roofline). Pad columns
+  // may sum garbage; they are never read.
+  auto db_full = at::zeros({kPartWPad}, dy.options().dtype(at::kFloat));
+  launch_slab_reduce(db_part.data_ptr<float>(), db_full.data_ptr<float>(),
+                     kPartWPad, db_part.size(0), current_stream());
+  auto db = db_full.narrow(0, 0, kPartW);
   auto dw4 = (db.narrow(0, 897, 128) + db.narrow(0, 897 + 128, 128))
                  .reshape({1, 128});
   return {dz1t,
