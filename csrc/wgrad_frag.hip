@@ -160,11 +160,30 @@ __global__ void __launch_bounds__(256) slab_reduce_kernel(
   if (i >= nk4) return;
   const int64_t lo = (int64_t)blockIdx.y * s1;
   const int64_t hi = min(lo + s1, nslabs);
-  wf_f32x4 s = {0.f, 0.f, 0.f, 0.f};
-  for (int64_t sl = lo; sl < hi; sl++) {
-    s += __builtin_nontemporal_load(
+  // 4 independent accumulators with grouped loads: a single add-chain
+  // loop serializes on one outstanding load per iteration (the backend
+  // waits vmcnt(0) before each accumulate).
+  wf_f32x4 s0 = {0.f, 0.f, 0.f, 0.f}, s1v = s0, s2 = s0, s3 = s0;
+  int64_t sl = lo;
+  for (; sl + 4 <= hi; sl += 4) {
+    const wf_f32x4 l0 = __builtin_nontemporal_load(
+        reinterpret_cast<const wf_f32x4*>(part + ((sl + 0) * nk4 + i) * 4));
+    const wf_f32x4 l1 = __builtin_nontemporal_load(
+        reinterpret_cast<const wf_f32x4*>(part + ((sl + 1) * nk4 + i) * 4));
+    const wf_f32x4 l2 = __builtin_nontemporal_load(
+        reinterpret_cast<const wf_f32x4*>(part + ((sl + 2) * nk4 + i) * 4));
+    const wf_f32x4 l3 = __builtin_nontemporal_load(
+        reinterpret_cast<const wf_f32x4*>(part + ((sl + 3) * nk4 + i) * 4));
+    s0 += l0;
+    s1v += l1;
+    s2 += l2;
+    s3 += l3;
+  }
+  for (; sl < hi; sl++) {
+    s0 += __builtin_nontemporal_load(
         reinterpret_cast<const wf_f32x4*>(part + (sl * nk4 + i) * 4));
   }
+  const wf_f32x4 s = (s0 + s1v) + (s2 + s3);
   float* o = out + i * 4;
   unsafeAtomicAdd(&o[0], s[0]);
   unsafeAtomicAdd(&o[1], s[1]);
