@@ -775,3 +775,35 @@ def test_wgrad_frag_bandwidth(dev):
     # library split-K bmm runs these at ~0.15/0.11/0.07 ms; require at
     # least rough parity so a regression is loud.
     assert sum(results.values()) < 0.5, results
+
+
+def test_read_files_packed_gpu_matches_cpu(dev, tmp_path):
+    """Row-group-parallel ingest on GPU (pinned arena + async H2D + pack
+    kernel) must produce byte-identical payload to the CPU path."""
+    from ray_shuffling_data_loader_amd.data_generation import (
+        float_data_spec,
+        generate_data,
+    )
+    from ray_shuffling_data_loader_amd.io import (
+        infer_schema,
+        read_files_packed,
+    )
+
+    filenames, _ = generate_data(
+        30_000, 3, 4, 0.0, str(tmp_path), spec=float_data_spec(9),
+        include_key=True,
+    )
+    filenames = list(filenames)
+    schema = infer_schema(filenames[0])
+    cpu = read_files_packed(
+        filenames, schema, torch.device("cpu"), reader_threads=4
+    )
+    gpu = read_files_packed(filenames, schema, dev, reader_threads=4)
+    torch.cuda.synchronize()
+    # payload bytes only (pad bytes are undefined on GPU)
+    for spec in schema.columns:
+        off = schema.offsets[spec.name]
+        nb = spec.row_bytes
+        assert torch.equal(
+            gpu[:, off : off + nb].cpu(), cpu[:, off : off + nb]
+        ), spec.name
