@@ -97,10 +97,14 @@ __global__ void __launch_bounds__(256, 2) wgrad_frag_kernel(
           *reinterpret_cast<const uint4*>(&bptr[kt][(i)*512]);             \
     }                                                                      \
   }
+// Consume fragments in REVERSE load order: the final MFMA of a batch
+// then depends on the FIRST-issued load, so the backend's waitcnt for
+// it leaves the newer prefetch loads outstanding (forward order ended
+// each batch with s_waitcnt vmcnt(0), draining the pipeline).
 #define WF_MFMA(abuf, bbuf)                                                \
   {                                                                        \
-    _Pragma("unroll") for (int nt = 0; nt < NT_W; nt++) {                  \
-      _Pragma("unroll") for (int kt = 0; kt < KT_W; kt++) {                \
+    _Pragma("unroll") for (int nt = NT_W - 1; nt >= 0; nt--) {             \
+      _Pragma("unroll") for (int kt = KT_W - 1; kt >= 0; kt--) {           \
         acc[nt][kt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(             \
             abuf[nt], bbuf[kt], acc[nt][kt], 0, 0, 0);                     \
       }                                                                    \
