@@ -168,9 +168,12 @@ __device__ void bc_layer(const short* __restrict__ dz_src,
     const uint32_t mw = mask_row[n];
     uint32_t p[8];
     bc_f32x2 colsum2 = {0.f, 0.f};
+    short* colbase =
+        WRITE_LDS ? dst_lds + 4 * h * DST_S + n : nullptr;
     #pragma unroll
     for (int q = 0; q < 8; q++) {
       const int32_t mrow = ((2 * q) & 3) + 8 * (q >> 1) + 4 * h;
+      const int32_t roff = (((2 * q) & 3) + 8 * (q >> 1)) * DST_S;
       const bc_f32x2 v2 = {
           ((mw >> mrow) & 1u) ? acc[nt][2 * q] : 0.f,
           ((mw >> (mrow + 1)) & 1u) ? acc[nt][2 * q + 1] : 0.f};
@@ -178,8 +181,8 @@ __device__ void bc_layer(const short* __restrict__ dz_src,
       const uint32_t pk = bc_cvt_pk_bf16(v2);
       p[q] = pk;
       if (WRITE_LDS) {
-        dst_lds[mrow * DST_S + n] = (short)(pk & 0xFFFFu);
-        dst_lds[(mrow + 1) * DST_S + n] = (short)(pk >> 16);
+        colbase[roff] = (short)(pk & 0xFFFFu);
+        colbase[roff + DST_S] = (short)(pk >> 16);
       }
     }
     float colsum = colsum2[0] + colsum2[1];

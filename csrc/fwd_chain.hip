@@ -197,18 +197,21 @@ __device__ void fc_layer(const short* __restrict__ src_lds,
     for (int nt = 0; nt < NT; nt++) {
       const int32_t n = n_base + nt * 32 + ml;
       const fc_f32x2 bv2 = {bias[n], bias[n]};
+      // One base address per column; row offsets are compile-time, so
+      // the b16 stores use ds_write immediate offsets instead of ~1.5
+      // VALU address ops each.
+      short* colbase = dst_lds + (mt * 32 + 4 * h) * DST_S + n;
       uint32_t p[8];
       #pragma unroll
       for (int q = 0; q < 8; q++) {
-        const int32_t mrow =
-            mt * 32 + ((2 * q) & 3) + 8 * (q >> 1) + 4 * h;
+        const int32_t roff = (((2 * q) & 3) + 8 * (q >> 1)) * DST_S;
         fc_f32x2 v2 = {acc[mt][nt][2 * q], acc[mt][nt][2 * q + 1]};
         v2 += bv2;
         if (RELU) v2 = fc_pk_max0(v2);
         const uint32_t pk = fc_cvt_pk_bf16(v2);
         p[q] = pk;
-        dst_lds[mrow * DST_S + n] = (short)(pk & 0xFFFFu);
-        dst_lds[(mrow + 1) * DST_S + n] = (short)(pk >> 16);
+        colbase[roff] = (short)(pk & 0xFFFFu);
+        colbase[roff + DST_S] = (short)(pk >> 16);
       }
       if (EMIT_T) {
         // Mask word for column n (bit i = bf16 value of row i > 0): a
