@@ -66,7 +66,10 @@ def shuffle(
     return engine.run()
 
 
+# Small LRU of engines for repeated shuffle_epoch calls: each cached
+# engine may pin an HBM-resident source block, so the cache is bounded.
 _EPOCH_ENGINES = {}
+_EPOCH_ENGINES_MAX = 4
 
 
 def shuffle_epoch(
@@ -100,6 +103,12 @@ def shuffle_epoch(
                 stats_collector=stats_collector,
                 **engine_kwargs,
             )
+            while len(_EPOCH_ENGINES) >= _EPOCH_ENGINES_MAX:
+                _EPOCH_ENGINES.pop(next(iter(_EPOCH_ENGINES)))
+            _EPOCH_ENGINES[key] = engine
+        else:
+            # LRU refresh
+            _EPOCH_ENGINES.pop(key, None)
             _EPOCH_ENGINES[key] = engine
         engine.num_epochs = max(engine.num_epochs, epoch + 1)
         engine.consumer = batch_consumer
