@@ -1,17 +1,16 @@
-// EXPERIMENTAL (round-2 WIP, see docs/MEGAKERNEL_PLAN.md): fused backward
-// chain for the flagship TabularMLP on MI355X (gfx950). Companion to
-// csrc/fwd_chain.hip.
+// Fused backward chain for the flagship TabularMLP on MI355X (gfx950) —
+// the DEFAULT bench train-step backward. Companion to csrc/fwd_chain.hip.
 //
 // Given dy = dLoss/dout [M,1] and the saved activations a1/a2/a3:
 //   da3 = dy * w4          (outer product)      dz3 = da3 * (a3 > 0)
 //   da2 = dz3 @ W3         ([M,128] -> [M,256]) dz2 = da2 * (a2 > 0)
 //   da1 = dz2 @ W2         ([M,256] -> [M,512]) dz1 = da1 * (a1 > 0)
 //   db_l = sum_m dz_l (l = 1..3), db4 = sum_m dy
-// One 64-row slab per workgroup; the a_l tiles are loaded once and
-// overwritten IN PLACE by their dz_l (the mask consumes the value it
-// replaces), so LDS peaks at ~118 KB. dz1/dz2/dz3 are written to global
-// for the (separate, reduction-shaped) wgrad kernels; bias partials go to
-// per-workgroup slabs finalized by one at::sum.
+// One 32-row slab per workgroup. The relu masks arrive as 1-bit words
+// from the forward (a1/a2 VALUES are never read); dz1/dz2/dz3 leave
+// ONLY as wgrad fragment-major transposes, emitted straight from the
+// epilogue registers; bias/dW4 partials go to per-workgroup slabs
+// reduced host-side by the slab-reduce kernel.
 //
 // dgrad orientation: da[m,c] = sum_n dz[m,n] W[n,c] — the contraction is
 // over the layer's OUTPUT index n, so the MFMA B fragment (contiguous
@@ -22,8 +21,7 @@
 // probe-verified in tools/mfma_probe.hip). In bc_layer terms WT is
 // [N, K]: N = dgrad output width (in_features), K = contraction width.
 //
-// Exercised only by the RSDL_EXPERIMENTAL=1 GPU test; not on any default
-// path.
+// Validated by tests/test_gpu_kernels.py and tests/test_chain_sim.py.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>

@@ -1,5 +1,6 @@
-// EXPERIMENTAL (round-2 WIP, see docs/MEGAKERNEL_PLAN.md): fused forward
-// chain for the flagship TabularMLP on MI355X (gfx950).
+// Fused forward chain for the flagship TabularMLP on MI355X (gfx950) —
+// the DEFAULT bench train-step forward (models/fused_step.py;
+// RSDL_FUSED_STEP=0 reverts to eager).
 //
 //   a1 = relu(x0 @ W1^T + b1)   [M,100] -> [M,512]
 //   a2 = relu(a1 @ W2^T + b2)   -> [M,256]
@@ -38,8 +39,9 @@
 // weight traffic coalesced and line-minimal. The host binding performs
 // the swizzle per call (weights are 0.44 MB total; a few us).
 //
-// Not wired into any default path: built and bound, exercised only by the
-// RSDL_EXPERIMENTAL=1 GPU test.
+// Validated by tests/test_gpu_kernels.py (chain numerics, layout
+// oracles, whole-step parity) and tests/test_chain_sim.py (lane-level
+// index simulation).
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>

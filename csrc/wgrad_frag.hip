@@ -19,9 +19,11 @@
 // Structure: NO LDS, no barriers — per m-chunk (16 rows) each wave loads
 // its (NT_W + KT_W) fragments directly from global (coalesced; waves of
 // one workgroup share B blocks -> L1 hits) and issues NT_W*KT_W MFMAs,
-// double-buffered. The workgroup owns an output block of
-// [128*NT_W n x 32*KT_W k] and one m-slab; fp32 partials land in dW with
-// unsafeAtomicAdd (coalesced 32x4B rows).
+// 3-deep prefetched, consumed in reverse load order so the final wait
+// leaves newer prefetches outstanding. The workgroup owns an output
+// block of [128*NT_W n x 32*KT_W k] and one m-slab; fp32 partials land
+// in per-slab buffers reduced by slab_reduce_kernel (global atomics at
+// ~16M adds/launch measured 35-65 us of contention).
 //
 // XCD-aware decode: consecutive blockIdx round-robin across the 8 XCDs,
 // so the grid is decoded as (xcd, seq) with all output-blocks of one
