@@ -240,6 +240,7 @@ def main():
             for p in params:
                 p.grad = flat_grad[off : off + p.numel()].view_as(p)
                 off += p.numel()
+            model._rsdl_flat_grads = True
     elif world > 1:
         model = torch.nn.parallel.DistributedDataParallel(model)
     try:

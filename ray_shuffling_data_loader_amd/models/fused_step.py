@@ -106,17 +106,17 @@ def fused_step(
         (dw3, db3),
         (dw4, db4),
     ]
+    # The DP bench pre-creates .grad as views of one flat buffer (so the
+    # gradient all-reduce is a single collective) and marks the model;
+    # only then do we COPY into them. Otherwise assign the fresh tensors
+    # (copying would add 8 small kernels per step at N=1).
+    flat_mode = getattr(model, "_rsdl_flat_grads", False)
     for m, (gw, gb) in zip(lin, grads):
-        # Copy into pre-existing .grad buffers when present (the DP bench
-        # pre-creates them as views of one flat buffer so the gradient
-        # all-reduce is a single collective); otherwise assign.
-        if m.weight.grad is not None and m.weight.grad.shape == gw.shape:
-            m.weight.grad.copy_(gw)
-        else:
-            m.weight.grad = gw.to(m.weight.dtype)
         gb = gb.reshape(m.bias.shape)
-        if m.bias.grad is not None:
+        if flat_mode:
+            m.weight.grad.copy_(gw)
             m.bias.grad.copy_(gb)
         else:
+            m.weight.grad = gw.to(m.weight.dtype)
             m.bias.grad = gb.to(m.bias.dtype)
     return loss

@@ -444,6 +444,7 @@ def test_fused_step_flat_grad_views_cpu(monkeypatch):
     }
 
     # Pass 2: flat-view grads (the bench's N>1 wiring) -> copies in place.
+    model._rsdl_flat_grads = True
     params = list(model.parameters())
     flat = torch.zeros(sum(p.numel() for p in params))
     off = 0
