@@ -1,11 +1,12 @@
 """Fused train step for the fixed TabularMLP(100-512-256-128-1) — the
 DEFAULT bench step (RSDL_FUSED_STEP=0 reverts to eager autocast).
 
-No autograd graph; the step IS the schedule, 7 kernels total:
-  x swizzle -> fwd chain (3 MFMA layers + head + MSE loss/grad; emits
-  a1^T/a2^T wgrad fragments + relu-mask words) -> bwd chain (dgrad chain
-  from mask bits; emits dz^T fragments + db/dW4 partials) -> 3
-  fragment-major wgrad kernels -> GEMV partial reduce -> optimizer.
+No autograd graph; the step IS the schedule (~12 launches):
+  x swizzle (emits both x layouts) -> fwd chain (3 MFMA layers + head +
+  MSE loss/grad; emits a1^T/a2^T wgrad fragments + relu-mask words) ->
+  bwd chain (dgrad chain from mask bits; emits dz^T fragments + db/dW4
+  partials) -> 3 fragment-major wgrad kernels -> slab-partial reduces ->
+  optimizer.
 Activations a1/a2 and gradients dz never exist in row-major form: the
 producers emit the transposed fragment layout the wgrad kernel reads
 (csrc/fwd_chain.hip, csrc/bwd_chain.hip, csrc/wgrad_frag.hip), and the
