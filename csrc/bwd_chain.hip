@@ -30,6 +30,31 @@
 
 namespace rsdl {
 
+// A/B'd: nontemporal streaming accesses measure ~4% faster than
+// cacheable (0.885 vs 0.919 ms fused step) — the activation streams
+// evicting L2 hurts more than write-combining helps, even with the
+// weight set fragment-swizzled. Keep nontemporal.
+#ifndef RSDL_PLAIN_STREAMS
+#define RSDL_PLAIN_STREAMS 0
+#endif
+template <typename T>
+__device__ __forceinline__ void _rsdl_store(T v, T* p) {
+#if RSDL_PLAIN_STREAMS
+  *p = v;
+#else
+  __builtin_nontemporal_store(v, p);
+#endif
+}
+template <typename T>
+__device__ __forceinline__ T _rsdl_load(const T* p) {
+#if RSDL_PLAIN_STREAMS
+  return *p;
+#else
+  return __builtin_nontemporal_load(p);
+#endif
+}
+
+
 typedef __attribute__((__vector_size__(8 * sizeof(short)))) short bc_bf16x8;
 typedef __attribute__((__vector_size__(16 * sizeof(float)))) float bc_f32x16;
 typedef __attribute__((__vector_size__(4 * sizeof(unsigned int)))) unsigned int bc_u32x4;
@@ -75,7 +100,7 @@ __device__ void bc_load_tile(const short* __restrict__ g, short* lds,
     if (m0 + m < M) {
       // Non-temporal: the activation stream (449 MB/step) must not evict
       // the L2-resident transposed weights every workgroup re-reads.
-      v = __builtin_nontemporal_load(
+      v = _rsdl_load(
           reinterpret_cast<const bc_u32x4*>(&g[(m0 + m) * N + c]));
     }
     *reinterpret_cast<bc_u32x4*>(&lds[m * S + c]) = v;
@@ -90,7 +115,7 @@ __device__ void bc_store_tile(const short* __restrict__ lds, short* out,
     const int32_t m = u / VPR;
     const int32_t c = (u % VPR) * 8;
     if (m0 + m < M) {
-      __builtin_nontemporal_store(
+      _rsdl_store(
           *reinterpret_cast<const bc_u32x4*>(&lds[m * S + c]),
           reinterpret_cast<bc_u32x4*>(&out[(m0 + m) * N + c]));
     }
@@ -203,8 +228,8 @@ __device__ void bc_layer(const short* __restrict__ dz_src,
                   ml * 8;
     short* blk1 = dzt_out + ((nt_g * mchunks + mc0 + 1) * 512) + h * 256 +
                   ml * 8;
-    __builtin_nontemporal_store(run0, reinterpret_cast<bc_u32x4*>(blk0));
-    __builtin_nontemporal_store(run1, reinterpret_cast<bc_u32x4*>(blk1));
+    _rsdl_store(run0, reinterpret_cast<bc_u32x4*>(blk0));
+    _rsdl_store(run1, reinterpret_cast<bc_u32x4*>(blk1));
   }
 }
 
@@ -292,7 +317,7 @@ __global__ void __launch_bounds__(256, 3) bwd_chain_kernel(
     }
     short* blk = dz3t + (((int64_t)ntr * mchunks + mc0 + mcl) * 512) +
                  hr * 256 + mlr * 8;
-    __builtin_nontemporal_store(*reinterpret_cast<bc_u32x4*>(run),
+    _rsdl_store(*reinterpret_cast<bc_u32x4*>(run),
                                 reinterpret_cast<bc_u32x4*>(blk));
   }
   __syncthreads();
