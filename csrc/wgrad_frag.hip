@@ -41,8 +41,8 @@ namespace rsdl {
 typedef __attribute__((__vector_size__(8 * sizeof(short)))) short wf_bf16x8;
 typedef __attribute__((__vector_size__(16 * sizeof(float)))) float wf_f32x16;
 
-template <int NT_W, int KT_W>
-__global__ void __launch_bounds__(256, 2) wgrad_frag_kernel(
+template <int NT_W, int KT_W, int MIN_WAVES = 2>
+__global__ void __launch_bounds__(256, MIN_WAVES) wgrad_frag_kernel(
     const short* __restrict__ AT, const short* __restrict__ BT,
     float* __restrict__ dW,  // [N,K] fp32, pre-zeroed
     int32_t N, int32_t K, int64_t mchunks, int32_t nblk_n, int32_t nblk_k,
