@@ -221,7 +221,16 @@ class ShuffleEngine:
         if self.source_cache in ("auto", "device"):
             self._cached_source = src
         elif self.source_cache == "host":
-            self._cached_source = src.cpu()
+            if src.device.type == "cuda":
+                # pinned host cache: the per-epoch re-upload then runs as
+                # an async DMA instead of a pageable copy
+                host = torch.empty(
+                    src.shape, dtype=src.dtype, pin_memory=True
+                )
+                host.copy_(src)
+                self._cached_source = host
+            else:
+                self._cached_source = src
         return src
 
     # ----- epoch pipeline -----------------------------------------------------
