@@ -492,11 +492,13 @@ def main():
         pass
 
     if os.environ.get("RSDL_LOG_MEM") == "1" and is_cuda and rank == 0:
+        # stderr: stdout carries exactly ONE JSON line (driver contract)
         print(
             f"[mem] allocated={torch.cuda.memory_allocated()/2**30:.2f}GiB "
             f"max_allocated={torch.cuda.max_memory_allocated()/2**30:.2f}GiB "
             f"reserved={torch.cuda.memory_reserved()/2**30:.2f}GiB",
             flush=True,
+            file=sys.stderr,
         )
 
     rows_per_sec = n_gpus * args.batch_size * args.steps / elapsed
