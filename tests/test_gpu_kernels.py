@@ -831,3 +831,34 @@ def test_fused_step_actually_learns(dev):
     start = sum(losses[:5]) / 5
     end = sum(losses[-5:]) / 5
     assert end < 0.4 * start, (start, end, losses[::10])
+
+
+def test_fused_step_edge_batch_sizes(dev):
+    """Fused step at awkward batch sizes (pad-heavy tiles, M < one slab,
+    M == 1): grads must track the eager autocast reference at every M."""
+    from ray_shuffling_data_loader_amd.models.fused_step import fused_step
+    from ray_shuffling_data_loader_amd.models.mlp import TabularMLP
+
+    torch.manual_seed(33)
+    for m_rows in [1, 17, 31, 32, 33, 255, 4097]:
+        model = TabularMLP(100).to(dev)
+        ref = TabularMLP(100).to(dev)
+        ref.load_state_dict(model.state_dict())
+        x = torch.randn(m_rows, 100, device=dev).bfloat16()
+        t = torch.randn(m_rows, 1, device=dev)
+        loss = fused_step(model, x, t)
+        with torch.autocast("cuda", torch.bfloat16):
+            out = ref(x)
+            ref_loss = torch.nn.functional.mse_loss(out.float(), t)
+        ref_loss.backward()
+        assert torch.isfinite(loss), m_rows
+        assert (
+            abs(loss.item() - ref_loss.item())
+            <= 0.05 * abs(ref_loss.item()) + 1e-3
+        ), (m_rows, loss.item(), ref_loss.item())
+        for (n, p), (_, q) in zip(
+            model.named_parameters(), ref.named_parameters()
+        ):
+            err = (p.grad - q.grad).abs().max()
+            tol = 0.1 * q.grad.abs().mean().clamp(min=1e-5) + 2e-3
+            assert err <= tol, (m_rows, n, err.item())
