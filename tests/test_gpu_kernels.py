@@ -860,5 +860,8 @@ def test_fused_step_edge_batch_sizes(dev):
             model.named_parameters(), ref.named_parameters()
         ):
             err = (p.grad - q.grad).abs().max()
-            tol = 0.1 * q.grad.abs().mean().clamp(min=1e-5) + 2e-3
+            # max-scaled: at tiny M most grad entries are relu-masked to
+            # ~zero, so a mean-scaled bound under-tolerates the bf16
+            # rounding-order differences on the few live entries.
+            tol = 0.08 * q.grad.abs().max().clamp(min=1e-5) + 2e-3
             assert err <= tol, (m_rows, n, err.item())
