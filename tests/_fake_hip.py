@@ -13,6 +13,16 @@ from ray_shuffling_data_loader_amd.ops.shuffle_ops import (
 )
 
 
+def _swz32(t):
+    """t_frag_swizzle with rows padded to a 32-multiple — the kernels'
+    slab granularity (the torch helper alone pads only to 16)."""
+    m = t.shape[0]
+    mp = (m + 31) // 32 * 32
+    if mp != m:
+        t = torch.nn.functional.pad(t, (0, 0, 0, mp - m))
+    return t_frag_swizzle(t)
+
+
 class FakeHip:
     @staticmethod
     def fwd_chain_bf16(x, W1, b1, W2, b2, W3, b3, w4, b4, target=None,
@@ -29,8 +39,8 @@ class FakeHip:
         a3 = torch.relu(a2 @ W3.float().t() + b3.float())
         out = a3 @ w4.float().unsqueeze(1) + b4.float()
         a1b, a2b = a1.bfloat16(), a2.bfloat16()
-        res = (t_frag_swizzle(a1b), relu_mask_words(a1b.float()),
-               t_frag_swizzle(a2b), relu_mask_words(a2b.float()),
+        res = (_swz32(a1b), relu_mask_words(a1b.float()),
+               _swz32(a2b), relu_mask_words(a2b.float()),
                a3.bfloat16(), out.bfloat16())
         if target is None:
             return res
@@ -60,8 +70,8 @@ class FakeHip:
         da1 = dz2.float() @ W2.float()
         dz1 = (da1 * mask_of(mask1, 512)).bfloat16()
         dw4 = dy.float().t() @ a3.float()
-        return (t_frag_swizzle(dz1), t_frag_swizzle(dz2),
-                t_frag_swizzle(dz3),
+        return (_swz32(dz1), _swz32(dz2),
+                _swz32(dz3),
                 dz1.float().sum(0), dz2.float().sum(0),
                 dz3.float().sum(0), dy.float().sum(0), dw4)
 
