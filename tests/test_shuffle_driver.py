@@ -79,3 +79,55 @@ def test_consume_helper():
     consume(0, c, 0, [RowBlock({"key": torch.arange(5)})])
     assert c.done[(0, 0)]
     assert torch.equal(c.rows[(0, 0)][0], torch.arange(5))
+
+
+def test_shuffle_epoch_reuses_cached_engine(tmp_path):
+    """Repeated shuffle_epoch calls with the same (files, reducers,
+    trainers) must reuse ONE engine (one ingest total) instead of
+    re-reading the source per call, and the cache stays bounded."""
+    import importlib
+
+    sh = importlib.import_module("ray_shuffling_data_loader_amd.shuffle")
+
+    sh._EPOCH_ENGINES.clear()
+    filenames, _ = generate_data(1000, 2, 1, 0.0, str(tmp_path))
+    c = CollectingConsumer(1, 3)
+    for e in range(3):
+        shuffle_epoch(e, list(filenames), c, num_reducers=2,
+                      num_trainers=1)
+    assert len(sh._EPOCH_ENGINES) == 1
+    eng = next(iter(sh._EPOCH_ENGINES.values()))
+    assert eng.num_epochs >= 3
+    for e in range(3):
+        keys = torch.cat(c.rows[(e, 0)])
+        assert sorted(keys.tolist()) == list(range(1000)), e
+
+    # Distinct configs evict LRU-style at the bound (4).
+    for r in (3, 4, 5, 6):
+        c2 = CollectingConsumer(1, 1)
+        shuffle_epoch(0, list(filenames), c2, num_reducers=r,
+                      num_trainers=1)
+    assert len(sh._EPOCH_ENGINES) == sh._EPOCH_ENGINES_MAX
+    # the original (reducers=2) entry was the oldest -> evicted
+    assert all(k[1] != 2 for k in sh._EPOCH_ENGINES)
+    sh._EPOCH_ENGINES.clear()
+
+
+def test_shuffle_epoch_explicit_engine(tmp_path):
+    """Passing engine= bypasses the cache entirely."""
+    import importlib
+
+    sh = importlib.import_module("ray_shuffling_data_loader_amd.shuffle")
+    from ray_shuffling_data_loader_amd.engine import ShuffleEngine
+
+    sh._EPOCH_ENGINES.clear()
+    filenames, _ = generate_data(500, 1, 1, 0.0, str(tmp_path))
+    c = CollectingConsumer(1, 1)
+    eng = ShuffleEngine(list(filenames), c, num_epochs=1,
+                        num_reducers=2, num_trainers=1)
+    shuffle_epoch(0, list(filenames), c, num_reducers=2,
+                  num_trainers=1, engine=eng)
+    assert len(sh._EPOCH_ENGINES) == 0
+    keys = torch.cat(c.rows[(0, 0)])
+    assert sorted(keys.tolist()) == list(range(500))
+    sh._EPOCH_ENGINES.clear()
