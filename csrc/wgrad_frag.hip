@@ -41,7 +41,7 @@ namespace rsdl {
 typedef __attribute__((__vector_size__(8 * sizeof(short)))) short wf_bf16x8;
 typedef __attribute__((__vector_size__(16 * sizeof(float)))) float wf_f32x16;
 
-template <int NT_W, int KT_W, int MIN_WAVES = 2>
+template <int NT_W, int KT_W, int MIN_WAVES = 2, int SCHED = 0>
 __global__ void __launch_bounds__(256, MIN_WAVES) wgrad_frag_kernel(
     const short* __restrict__ AT, const short* __restrict__ BT,
     float* __restrict__ dW,  // [N,K] fp32, pre-zeroed
@@ -113,25 +113,109 @@ __global__ void __launch_bounds__(256, MIN_WAVES) wgrad_frag_kernel(
     }                                                                      \
   }
 
+// SCHED=1: pin the load/MFMA group order with sched_barrier(0).
+// Without it the backend SINKS one batch's loads to just before their
+// consuming MFMAs (seen in the gfx950 asm: a global_load_dwordx4 group
+// at offset:2048 immediately followed by s_waitcnt vmcnt(4)/(2)/(0)
+// into MFMAs on those registers), collapsing the 3-deep pipeline once
+// per 3 chunks and exposing a full HBM latency. With the fences every
+// batch's consume sits >= 2 load-groups (12 loads) behind its issue.
+// Static property verified by disassembly (profiles/r02/
+// wgrad_sched_asm.md); default OFF until measured on hardware.
+#define WF_FENCE()                                                         \
+  if (SCHED) __builtin_amdgcn_sched_barrier(0)
   // 3-deep prefetch: at depth 1 only ~6 KB/wave was in flight and the
   // kernel ran latency-bound at ~half the stream roofline (depth 4
   // spills the <1,8> config's registers).
-  WF_LOAD(a0, b0, 0);
-  if (1 < iters) WF_LOAD(a1, b1, 1);
   int64_t i = 0;
-  while (i + 3 <= iters) {
-    WF_LOAD(a2, b2, i + 2);
+  // Fenced variants: depth 3 fits <2,4>'s registers; <1,8> (9-quad load
+  // groups) spills 36 VGPRs at depth 3 once the fences extend the
+  // buffer live ranges, so it runs the 2-buffer depth-2 loop instead.
+  // Both hoist an UNCONDITIONAL 2-chunk preheader under an iters>=2
+  // guard: with the shared `if (1 < iters)` preheader below, the loop
+  // header has a predecessor path with a different outstanding-load
+  // count and the waitcnt pass merges to vmcnt(0).
+  constexpr bool SCHED3 = SCHED && (NT_W + KT_W <= 6);
+  constexpr bool SCHED2 = SCHED && (NT_W + KT_W > 6);
+  if (!SCHED) {
+    WF_LOAD(a0, b0, 0);
+    if (1 < iters) WF_LOAD(a1, b1, 1);
+  } else if (iters < 2) {
+    WF_LOAD(a0, b0, 0);
     WF_MFMA(a0, b0);
-    if (i + 3 < iters) WF_LOAD(a0, b0, i + 3);
-    WF_MFMA(a1, b1);
-    if (i + 4 < iters) WF_LOAD(a1, b1, i + 4);
-    WF_MFMA(a2, b2);
-    i += 3;
   }
-  if (i < iters) WF_MFMA(a0, b0);
-  if (i + 1 < iters) WF_MFMA(a1, b1);
+  if (SCHED && iters < 2) {
+    // handled above
+  } else if (SCHED3) {
+    WF_LOAD(a0, b0, 0);
+    WF_LOAD(a1, b1, 1);
+    // Straight-line steady loop: every prefetch is UNCONDITIONAL (the
+    // loop bound guarantees i+4 < iters), so the outstanding-load count
+    // is path-independent and the backend can emit precise partial
+    // vmcnt waits instead of the vmcnt(0) it is forced to at the
+    // control-flow joins of the conditional-load loop below. Fences pin
+    // the load/MFMA group order so no load sinks to its consumer.
+    while (i + 5 <= iters) {
+      WF_LOAD(a2, b2, i + 2);
+      WF_FENCE();
+      WF_MFMA(a0, b0);
+      WF_FENCE();
+      WF_LOAD(a0, b0, i + 3);
+      WF_FENCE();
+      WF_MFMA(a1, b1);
+      WF_FENCE();
+      WF_LOAD(a1, b1, i + 4);
+      WF_FENCE();
+      WF_MFMA(a2, b2);
+      WF_FENCE();
+      i += 3;
+    }
+    // Drain: rem = iters - i in [1,4]; a0/a1 hold chunks i, i+1.
+    if (i + 2 < iters) WF_LOAD(a2, b2, i + 2);
+    if (i < iters) WF_MFMA(a0, b0);
+    if (i + 3 < iters) WF_LOAD(a0, b0, i + 3);
+    if (i + 1 < iters) WF_MFMA(a1, b1);
+    if (i + 2 < iters) WF_MFMA(a2, b2);
+    if (i + 3 < iters) WF_MFMA(a0, b0);
+  } else if (SCHED2) {
+    // Same idea at prefetch depth 2 (two buffers): each load group sits
+    // one consume-group (NT_W*KT_W MFMAs) ahead of its use, waits stay
+    // at vmcnt(NT_W+KT_W) in a single-block loop.
+    WF_LOAD(a0, b0, 0);
+    WF_LOAD(a1, b1, 1);
+    while (i + 4 <= iters) {
+      WF_FENCE();
+      WF_MFMA(a0, b0);
+      WF_FENCE();
+      WF_LOAD(a0, b0, i + 2);
+      WF_FENCE();
+      WF_MFMA(a1, b1);
+      WF_FENCE();
+      WF_LOAD(a1, b1, i + 3);
+      WF_FENCE();
+      i += 2;
+    }
+    // Drain: rem = iters - i in [1,3]; a0/a1 hold chunks i, i+1.
+    if (i + 2 < iters) WF_LOAD(a2, b2, i + 2);
+    if (i < iters) WF_MFMA(a0, b0);
+    if (i + 1 < iters) WF_MFMA(a1, b1);
+    if (i + 2 < iters) WF_MFMA(a2, b2);
+  } else {
+    while (i + 3 <= iters) {
+      WF_LOAD(a2, b2, i + 2);
+      WF_MFMA(a0, b0);
+      if (i + 3 < iters) WF_LOAD(a0, b0, i + 3);
+      WF_MFMA(a1, b1);
+      if (i + 4 < iters) WF_LOAD(a1, b1, i + 4);
+      WF_MFMA(a2, b2);
+      i += 3;
+    }
+    if (i < iters) WF_MFMA(a0, b0);
+    if (i + 1 < iters) WF_MFMA(a1, b1);
+  }
 #undef WF_LOAD
 #undef WF_MFMA
+#undef WF_FENCE
 
   // Epilogue: D[row = n-in-tile][col = k-in-tile]; 32 lanes write 32
   // consecutive k's. Plain per-slab partial stores (a second-stage
@@ -252,17 +336,26 @@ void launch_wgrad_frag(const void* AT, const void* BT, float* dW, int32_t N,
   // written; there is no zero-init)
   nslabs = (mchunks + chunks_per_slab - 1) / chunks_per_slab;
   static bool linear = std::getenv("RSDL_WGRAD_LINEAR") != nullptr;
+  // Pinned-schedule variant (see WF_FENCE above); flip on for A/B runs.
+  static bool sched = [] {
+    const char* e = std::getenv("RSDL_WGRAD_SCHED");
+    return e && e[0] == '1';
+  }();
   const int64_t grid =
       linear ? nslabs * nblk : ((nslabs + 7) / 8) * 8 * nblk;
   const int32_t nb_n = linear ? -nblk_n : nblk_n;
   if (nt_w == 2 && kt_w == 4) {
-    hipLaunchKernelGGL((wgrad_frag_kernel<2, 4>), dim3((uint32_t)grid),
+    auto kern = sched ? wgrad_frag_kernel<2, 4, 2, 1>
+                      : wgrad_frag_kernel<2, 4>;
+    hipLaunchKernelGGL(kern, dim3((uint32_t)grid),
                        dim3(256), 0, stream,
                        reinterpret_cast<const short*>(AT),
                        reinterpret_cast<const short*>(BT), dW, N, K, mchunks,
                        nb_n, nblk_k, nslabs, chunks_per_slab);
   } else if (nt_w == 1 && kt_w == 8) {
-    hipLaunchKernelGGL((wgrad_frag_kernel<1, 8>), dim3((uint32_t)grid),
+    auto kern = sched ? wgrad_frag_kernel<1, 8, 2, 1>
+                      : wgrad_frag_kernel<1, 8>;
+    hipLaunchKernelGGL(kern, dim3((uint32_t)grid),
                        dim3(256), 0, stream,
                        reinterpret_cast<const short*>(AT),
                        reinterpret_cast<const short*>(BT), dW, N, K, mchunks,
