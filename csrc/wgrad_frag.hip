@@ -242,6 +242,7 @@ __global__ void __launch_bounds__(256, MIN_WAVES) wgrad_frag_kernel(
 // chip; the few split partials combine with fp32 atomics (<= NSPLIT
 // writers per line — negligible contention). out must be zeroed.
 typedef __attribute__((__vector_size__(4 * sizeof(float)))) float wf_f32x4;
+template <int SCHED = 0>
 __global__ void __launch_bounds__(256) slab_reduce_kernel(
     const float* __restrict__ part, float* __restrict__ out, int64_t nk4,
     int64_t nslabs, int64_t s1) {
@@ -257,19 +258,68 @@ __global__ void __launch_bounds__(256) slab_reduce_kernel(
   // bypassed that and measured ~1.6-2.9 TB/s.
   wf_f32x4 s0 = {0.f, 0.f, 0.f, 0.f}, s1v = s0, s2 = s0, s3 = s0;
   int64_t sl = lo;
-  for (; sl + 4 <= hi; sl += 4) {
-    const wf_f32x4 l0 =
-        *reinterpret_cast<const wf_f32x4*>(part + ((sl + 0) * nk4 + i) * 4);
-    const wf_f32x4 l1 =
-        *reinterpret_cast<const wf_f32x4*>(part + ((sl + 1) * nk4 + i) * 4);
-    const wf_f32x4 l2 =
-        *reinterpret_cast<const wf_f32x4*>(part + ((sl + 2) * nk4 + i) * 4);
-    const wf_f32x4 l3 =
-        *reinterpret_cast<const wf_f32x4*>(part + ((sl + 3) * nk4 + i) * 4);
-    s0 += l0;
-    s1v += l1;
-    s2 += l2;
-    s3 += l3;
+  if (SCHED) {
+    // Double-buffered 4-load groups, 2x-unrolled so the buffer roles
+    // swap instead of copying (a la[j]=lb[j] rotation materializes as
+    // 12 v_movs + a vmcnt(0)): the next group is UNCONDITIONALLY in
+    // flight while the current one accumulates, waits sit at vmcnt(4)
+    // in a single-block loop.
+    wf_f32x4 la[4], lb[4];
+#define SR_LOAD(buf, base)                                                 \
+  {                                                                        \
+    _Pragma("unroll") for (int j = 0; j < 4; j++) {                        \
+      buf[j] = *reinterpret_cast<const wf_f32x4*>(                         \
+          part + (((base) + j) * nk4 + i) * 4);                            \
+    }                                                                      \
+  }
+#define SR_ACC(buf)                                                        \
+  {                                                                        \
+    s0 += buf[0];                                                          \
+    s1v += buf[1];                                                         \
+    s2 += buf[2];                                                          \
+    s3 += buf[3];                                                          \
+  }
+    if (sl + 4 <= hi) {
+      SR_LOAD(la, sl);
+      while (sl + 12 <= hi) {
+        SR_LOAD(lb, sl + 4);
+        __builtin_amdgcn_sched_barrier(0);
+        SR_ACC(la);
+        __builtin_amdgcn_sched_barrier(0);
+        SR_LOAD(la, sl + 8);
+        __builtin_amdgcn_sched_barrier(0);
+        SR_ACC(lb);
+        __builtin_amdgcn_sched_barrier(0);
+        sl += 8;
+      }
+      // rem groups: la holds sl; possibly one more unloaded group.
+      if (sl + 8 <= hi) {
+        SR_LOAD(lb, sl + 4);
+        SR_ACC(la);
+        SR_ACC(lb);
+        sl += 8;
+      } else {
+        SR_ACC(la);
+        sl += 4;
+      }
+    }
+#undef SR_LOAD
+#undef SR_ACC
+  } else {
+    for (; sl + 4 <= hi; sl += 4) {
+      const wf_f32x4 l0 = *reinterpret_cast<const wf_f32x4*>(
+          part + ((sl + 0) * nk4 + i) * 4);
+      const wf_f32x4 l1 = *reinterpret_cast<const wf_f32x4*>(
+          part + ((sl + 1) * nk4 + i) * 4);
+      const wf_f32x4 l2 = *reinterpret_cast<const wf_f32x4*>(
+          part + ((sl + 2) * nk4 + i) * 4);
+      const wf_f32x4 l3 = *reinterpret_cast<const wf_f32x4*>(
+          part + ((sl + 3) * nk4 + i) * 4);
+      s0 += l0;
+      s1v += l1;
+      s2 += l2;
+      s3 += l3;
+    }
   }
   for (; sl < hi; sl++) {
     s0 += *reinterpret_cast<const wf_f32x4*>(part + (sl * nk4 + i) * 4);
@@ -296,7 +346,12 @@ void launch_slab_reduce(const float* part, float* out, int64_t nk,
   if (nsplit > nslabs) nsplit = nslabs;
   const int64_t s1 = (nslabs + nsplit - 1) / nsplit;
   nsplit = (nslabs + s1 - 1) / s1;
-  hipLaunchKernelGGL(slab_reduce_kernel,
+  static bool sched = [] {
+    const char* e = std::getenv("RSDL_WGRAD_SCHED");
+    return e && e[0] == '1';
+  }();
+  auto kern = sched ? slab_reduce_kernel<1> : slab_reduce_kernel<0>;
+  hipLaunchKernelGGL(kern,
                      dim3((uint32_t)gx, (uint32_t)nsplit), dim3(256), 0,
                      stream, part, out, nk4, nslabs, s1);
 }
