@@ -77,3 +77,36 @@ def test_bench_world2_gloo(tmp_path):
     assert d["n_gpus"] == 2
     assert d["config"]["parallelism"] == "dp2"
     assert d["value"] > 0
+
+
+def test_bench_world4_gloo(tmp_path):
+    """Same path at world 4: 4-way all-to-all splits and the MAX-over-4
+    timing reduce (closest CPU rehearsal of the driver's 8-GPU tier)."""
+    env = dict(os.environ)
+    env["RSDL_TUNABLEOP"] = "0"
+    proc = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--standalone", "--local-addr", "127.0.0.1",
+            "--nproc-per-node", "4",
+            os.path.join(REPO, "bench.py"),
+            "--gpus", "4", "--steps", "2", "--warmup", "1",
+            "--rows-per-gpu", "40000", "--batch-size", "10000",
+            "--num-cols", "8", "--files-per-gpu", "1",
+            "--reducers-per-gpu", "1", "--device", "cpu",
+            "--dtype", "fp32", "--data-dir", str(tmp_path),
+        ],
+        capture_output=True,
+        text=True,
+        timeout=600,
+        env=env,
+        cwd=REPO,
+    )
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    lines = [ln for ln in proc.stdout.splitlines() if ln.strip()]
+    json_lines = [ln for ln in lines if ln.lstrip().startswith("{")]
+    assert len(json_lines) == 1, f"exactly one JSON line, got: {lines}"
+    d = json.loads(json_lines[0])
+    assert d["n_gpus"] == 4
+    assert d["config"]["parallelism"] == "dp4"
+    assert d["value"] > 0
