@@ -132,7 +132,8 @@ __device__ void bc_store_tile(const short* __restrict__ lds, short* out,
 //     registers to global (same shfl_xor half-row exchange as the
 //     forward emitter) — what the wgrad kernel reads, coalesced.
 // Bias partials db[n] fold into the epilogue as before.
-template <int K, int N, int SRC_S, int DST_S, bool WRITE_LDS>
+template <int K, int N, int SRC_S, int DST_S, bool WRITE_LDS,
+          bool PI16 = false>
 __device__ void bc_layer(const short* __restrict__ dz_src,
                          const short* __restrict__ WT,
                          const uint32_t* __restrict__ mask_row,
@@ -211,18 +212,33 @@ __device__ void bc_layer(const short* __restrict__ dz_src,
     float colsum = colsum2[0] + colsum2[1];
     colsum += __shfl_xor(colsum, 32);
     if (lane < 32) db_out[n] = colsum;
-    // Transposed fragment emission, packed-pair exchange (see
-    // fwd_chain fc_layer EMIT_T).
-    uint32_t rx[4];
-    #pragma unroll
-    for (int j = 0; j < 2; j++) {
-      rx[j] = __shfl_xor((int)(h == 0 ? p[2 + j] : p[j]), 32);
-      rx[2 + j] = __shfl_xor((int)(h == 0 ? p[6 + j] : p[4 + j]), 32);
+    // Transposed fragment emission (see fwd_chain fc_layer EMIT_T):
+    // PI16 writes each half-wave's own packed pairs as the runs (the
+    // consumers agree on the pi16 intra-chunk M-permutation); default
+    // is the packed-pair half-row exchange.
+    bc_u32x4 run0, run1;
+    if (PI16) {
+      #pragma unroll
+      for (int j = 0; j < 4; j++) {
+        run0[j] = p[j];
+        run1[j] = p[4 + j];
+      }
+    } else {
+      uint32_t rx[4];
+      #pragma unroll
+      for (int j = 0; j < 2; j++) {
+        rx[j] = __shfl_xor((int)(h == 0 ? p[2 + j] : p[j]), 32);
+        rx[2 + j] = __shfl_xor((int)(h == 0 ? p[6 + j] : p[4 + j]), 32);
+      }
+      run0[0] = h == 0 ? p[0] : rx[0];
+      run0[1] = h == 0 ? p[1] : rx[1];
+      run0[2] = h == 0 ? rx[0] : p[2];
+      run0[3] = h == 0 ? rx[1] : p[3];
+      run1[0] = h == 0 ? p[4] : rx[2];
+      run1[1] = h == 0 ? p[5] : rx[3];
+      run1[2] = h == 0 ? rx[2] : p[6];
+      run1[3] = h == 0 ? rx[3] : p[7];
     }
-    const bc_u32x4 run0 = {h == 0 ? p[0] : rx[0], h == 0 ? p[1] : rx[1],
-                           h == 0 ? rx[0] : p[2], h == 0 ? rx[1] : p[3]};
-    const bc_u32x4 run1 = {h == 0 ? p[4] : rx[2], h == 0 ? p[5] : rx[3],
-                           h == 0 ? rx[2] : p[6], h == 0 ? rx[3] : p[7]};
     const int64_t nt_g = (int64_t)(n_base + nt * 32) >> 5;
     short* blk0 = dzt_out + ((nt_g * mchunks + mc0) * 512) + h * 256 +
                   ml * 8;
@@ -233,6 +249,7 @@ __device__ void bc_layer(const short* __restrict__ dz_src,
   }
 }
 
+template <bool PI16 = false>
 __global__ void __launch_bounds__(256, 3) bwd_chain_kernel(
     const short* __restrict__ dy,     // [M,1] bf16 (head grad)
     const short* __restrict__ a3,     // [M,128] saved activations
@@ -313,7 +330,10 @@ __global__ void __launch_bounds__(256, 3) bwd_chain_kernel(
     short run[8];
     #pragma unroll
     for (int j = 0; j < 8; j++) {
-      run[j] = t3[(mbase + j) * BC_S3 + k];
+      const int32_t mr = PI16
+          ? (mcl * 16 + (j & 3) + ((j & 4) << 1) + 4 * hr)
+          : (mbase + j);
+      run[j] = t3[mr * BC_S3 + k];
     }
     short* blk = dz3t + (((int64_t)ntr * mchunks + mc0 + mcl) * 512) +
                  hr * 256 + mlr * 8;
@@ -324,12 +344,12 @@ __global__ void __launch_bounds__(256, 3) bwd_chain_kernel(
 
   // da2 = dz3 @ W3 (via W3T), masked -> dz2: LDS tile (next layer's A
   // source) + transposed emission + db2 partials.
-  bc_layer<BC_N3, BC_N2, BC_S3, BC_S2, true>(
+  bc_layer<BC_N3, BC_N2, BC_S3, BC_S2, true, PI16>(
       t3, W3T, &mask2[(int64_t)blockIdx.x * BC_N2], t2, dz2t,
       &part[BC_N1], mchunks, mc0, wave, lane);
   __syncthreads();
   // da1 = dz2 @ W2 (via W2T), masked -> dz1: transposed emission only.
-  bc_layer<BC_N2, BC_N1, BC_S2, BC_S1, false>(
+  bc_layer<BC_N2, BC_N1, BC_S2, BC_S1, false, PI16>(
       t2, W2T, &mask1[(int64_t)blockIdx.x * BC_N1], nullptr, dz1t,
       &part[0], mchunks, mc0, wave, lane);
 }
@@ -339,10 +359,12 @@ int64_t bwd_chain_grid(int64_t M) { return (M + BC_MT - 1) / BC_MT; }
 void launch_bwd_chain(const void* dy, const void* a3, const void* mask1,
                       const void* mask2, const void* w4, const void* W3T,
                       const void* W2T, void* dz1t, void* dz2t, void* dz3t,
-                      float* db_part, int64_t M, hipStream_t stream) {
+                      float* db_part, int64_t M, int pi16,
+                      hipStream_t stream) {
   const int32_t grid = (int32_t)((M + BC_MT - 1) / BC_MT);
   const int64_t mchunks = (int64_t)grid * 2;
-  hipLaunchKernelGGL(bwd_chain_kernel, dim3(grid), dim3(256), 0, stream,
+  auto kern = pi16 ? bwd_chain_kernel<true> : bwd_chain_kernel<false>;
+  hipLaunchKernelGGL(kern, dim3(grid), dim3(256), 0, stream,
                      reinterpret_cast<const short*>(dy),
                      reinterpret_cast<const short*>(a3),
                      reinterpret_cast<const uint32_t*>(mask1),

@@ -135,7 +135,7 @@ __device__ __forceinline__ short fc_f2b(float f) {
 // Lanes l/l+32 hold complementary 4-row runs of a column; one shfl_xor
 // per 4 values assembles the 8-row fragment runs in registers.
 template <int K, int N, int SRC_S, int DST_S, bool RELU,
-          bool A_FRAGMAJOR = false, bool EMIT_T = false>
+          bool A_FRAGMAJOR = false, bool EMIT_T = false, bool PI16 = false>
 __device__ void fc_layer(const short* __restrict__ src_lds,
                          const short* __restrict__ W,
                          const float* __restrict__ bias,
@@ -252,22 +252,38 @@ __device__ void fc_layer(const short* __restrict__ src_lds,
           w |= (uint32_t)(lo - 1u < 0x7FFFu) << mrow;
           w |= (uint32_t)(hi - 1u < 0x7FFFu) << (mrow + 1);
         }
-        // Exchange the half-rows as packed ints (4 shfls); lane h=0
-        // assembles rows {0-7},{16-23}, h=1 {8-15},{24-31}.
-        uint32_t rx[4];
-        #pragma unroll
-        for (int j = 0; j < 2; j++) {
-          rx[j] = __shfl_xor((int)(h == 0 ? p[2 + j] : p[j]), 32);
-          rx[2 + j] = __shfl_xor((int)(h == 0 ? p[6 + j] : p[4 + j]), 32);
+        // PI16: every producer/consumer of the transposed layout agrees
+        // on the pi16 intra-chunk M-permutation (swap bits 2<->3 of the
+        // 16-row position), under which each half-wave's own packed
+        // pairs ARE the 8-element runs — the 4-shfl half-row exchange
+        // and its slot selects disappear. M is the contraction dim of
+        // every consumer (wgrad_frag), so dW is invariant.
+        // Default (!PI16): exchange the half-rows as packed ints (4
+        // shfls); lane h=0 assembles rows {0-7},{16-23}, h=1
+        // {8-15},{24-31}.
+        fc_u32x4 run0, run1;
+        if (PI16) {
+          #pragma unroll
+          for (int j = 0; j < 4; j++) {
+            run0[j] = p[j];
+            run1[j] = p[4 + j];
+          }
+        } else {
+          uint32_t rx[4];
+          #pragma unroll
+          for (int j = 0; j < 2; j++) {
+            rx[j] = __shfl_xor((int)(h == 0 ? p[2 + j] : p[j]), 32);
+            rx[2 + j] = __shfl_xor((int)(h == 0 ? p[6 + j] : p[4 + j]), 32);
+          }
+          run0[0] = h == 0 ? p[0] : rx[0];
+          run0[1] = h == 0 ? p[1] : rx[1];
+          run0[2] = h == 0 ? rx[0] : p[2];
+          run0[3] = h == 0 ? rx[1] : p[3];
+          run1[0] = h == 0 ? p[4] : rx[2];
+          run1[1] = h == 0 ? p[5] : rx[3];
+          run1[2] = h == 0 ? rx[2] : p[6];
+          run1[3] = h == 0 ? rx[3] : p[7];
         }
-        const fc_u32x4 run0 = {h == 0 ? p[0] : rx[0],
-                               h == 0 ? p[1] : rx[1],
-                               h == 0 ? rx[0] : p[2],
-                               h == 0 ? rx[1] : p[3]};
-        const fc_u32x4 run1 = {h == 0 ? p[4] : rx[2],
-                               h == 0 ? p[5] : rx[3],
-                               h == 0 ? rx[2] : p[6],
-                               h == 0 ? rx[3] : p[7]};
         const int64_t nt_g = (int64_t)(n_base + nt * 32) >> 5;
         short* blk0 = at_out + ((nt_g * mchunks + mc0) * 512) + h * 256 +
                       ml * 8;
@@ -304,6 +320,7 @@ __device__ void fc_store_tile(const short* __restrict__ lds, short* out,
   }
 }
 
+template <bool PI16 = false>
 __global__ void __launch_bounds__(256) fwd_chain_kernel(
     const short* __restrict__ x0s,  // fragment-major swizzled x (see note)
     const short* __restrict__ W1, const float* __restrict__ b1,
@@ -344,11 +361,11 @@ __global__ void __launch_bounds__(256) fwd_chain_kernel(
 
   // Layer 1: A fragments straight from the swizzled global x block.
   const short* xblk = &x0s[(int64_t)blockIdx.x * (FC_K0P / 16) * 512];
-  fc_layer<FC_K0P, FC_N1, 0, FC_S1, true, true, true>(
+  fc_layer<FC_K0P, FC_N1, 0, FC_S1, true, true, true, PI16>(
       xblk, W1, b1, t1, wave, lane, a1t, &mask1[(int64_t)blockIdx.x * FC_N1],
       mchunks, mc0);
   __syncthreads();
-  fc_layer<FC_N1, FC_N2, FC_S1, FC_S2, true, false, true>(
+  fc_layer<FC_N1, FC_N2, FC_S1, FC_S2, true, false, true, PI16>(
       t1, W2, b2, t2, wave, lane, a2t, &mask2[(int64_t)blockIdx.x * FC_N2],
       mchunks, mc0);
   __syncthreads();
@@ -454,6 +471,7 @@ void launch_swizzle_x(const void* x, void* out, int64_t M,
 // One thread per 16-B fwd block (8 k for one row m); the wgrad-side
 // elements it holds are scattered, so the wgrad half works via LDS: the
 // workgroup stages a [32 m x 128 k] tile and re-emits it transposed.
+template <bool PI16 = false>
 __global__ void __launch_bounds__(256) swizzle_x_both_kernel(
     const short* __restrict__ x, short* __restrict__ xs,
     short* __restrict__ xt, int64_t M, int64_t mtiles) {
@@ -504,7 +522,14 @@ __global__ void __launch_bounds__(256) swizzle_x_both_kernel(
     fc_u32x4 pack;
     short* vp = reinterpret_cast<short*>(&pack);
     #pragma unroll
-    for (int j = 0; j < 8; j++) vp[j] = tile[(mb + j) * 136 + k];
+    for (int j = 0; j < 8; j++) {
+      // PI16 intra-chunk M-permutation (pos -> row = swap bits 2<->3):
+      // slice h's positions hold rows {0-3,8-11} (h=0) / {4-7,12-15}.
+      const int32_t mr = PI16
+          ? (mcl * 16 + (j & 3) + ((j & 4) << 1) + 4 * h)
+          : (mb + j);
+      vp[j] = tile[mr * 136 + k];
+    }
     _rsdl_store(
         pack, reinterpret_cast<fc_u32x4*>(
                   &xt[(((int64_t)kt * mchunks + mt * 2 + mcl) * 2 + h) *
@@ -514,9 +539,11 @@ __global__ void __launch_bounds__(256) swizzle_x_both_kernel(
 }
 
 void launch_swizzle_x_both(const void* x, void* xs, void* xt, int64_t M,
-                           hipStream_t stream) {
+                           int pi16, hipStream_t stream) {
   const int64_t mtiles = (M + FC_MT - 1) / FC_MT;
-  hipLaunchKernelGGL(swizzle_x_both_kernel, dim3((uint32_t)mtiles),
+  auto kern =
+      pi16 ? swizzle_x_both_kernel<true> : swizzle_x_both_kernel<false>;
+  hipLaunchKernelGGL(kern, dim3((uint32_t)mtiles),
                      dim3(256), 0, stream,
                      reinterpret_cast<const short*>(x),
                      reinterpret_cast<short*>(xs),
@@ -528,6 +555,7 @@ void launch_swizzle_x_both(const void* x, void* xs, void* xt, int64_t M,
 // One thread per 16-B out block = 8 consecutive m's of one column k;
 // writes fully coalesced, reads gather 8 row-strided elements (L1-local:
 // neighboring threads read the same 8 rows).
+template <bool PI16 = false>
 __global__ void __launch_bounds__(256) swizzle_xt_kernel(
     const short* __restrict__ x, short* __restrict__ out, int64_t M,
     int64_t total_blocks) {
@@ -548,20 +576,24 @@ __global__ void __launch_bounds__(256) swizzle_xt_kernel(
   if (k < FC_K0) {
     #pragma unroll
     for (int j = 0; j < 8; j++) {
-      if (m0 + j < M) v[j] = x[(m0 + j) * FC_K0 + k];
+      const int64_t m = PI16
+          ? (mc * 16 + (j & 3) + ((j & 4) << 1) + 4 * h)
+          : (m0 + j);
+      if (m < M) v[j] = x[m * FC_K0 + k];
     }
   }
   _rsdl_store(*reinterpret_cast<fc_u32x4*>(v),
                               reinterpret_cast<fc_u32x4*>(&out[b * 8]));
 }
 
-void launch_swizzle_xt(const void* x, void* out, int64_t M,
+void launch_swizzle_xt(const void* x, void* out, int64_t M, int pi16,
                        hipStream_t stream) {
   const int64_t mtiles = (M + FC_MT - 1) / FC_MT;
   const int64_t mchunks = mtiles * 2;
   const int64_t total = 4 * mchunks * 2 * 32;  // kt * mc * h * ml
   const int64_t grid = (total + 255) / 256;
-  hipLaunchKernelGGL(swizzle_xt_kernel, dim3((uint32_t)grid), dim3(256), 0,
+  auto kern = pi16 ? swizzle_xt_kernel<true> : swizzle_xt_kernel<false>;
+  hipLaunchKernelGGL(kern, dim3((uint32_t)grid), dim3(256), 0,
                      stream, reinterpret_cast<const short*>(x),
                      reinterpret_cast<short*>(out), M, total);
 }
@@ -574,10 +606,11 @@ void launch_fwd_chain(const void* x0s, const void* W1, const float* b1,
                       void* a1t, uint32_t* mask1, void* a2t,
                       uint32_t* mask2, void* a3, void* out,
                       const float* target, void* dyb, float* loss_part,
-                      int64_t M, hipStream_t stream) {
+                      int64_t M, int pi16, hipStream_t stream) {
   const int32_t grid = (int32_t)((M + FC_MT - 1) / FC_MT);
   const int64_t mchunks = (int64_t)grid * 2;
-  hipLaunchKernelGGL(fwd_chain_kernel, dim3(grid), dim3(256), 0, stream,
+  auto kern = pi16 ? fwd_chain_kernel<true> : fwd_chain_kernel<false>;
+  hipLaunchKernelGGL(kern, dim3(grid), dim3(256), 0, stream,
                      reinterpret_cast<const short*>(x0s),
                      reinterpret_cast<const short*>(W1), b1,
                      reinterpret_cast<const short*>(W2), b2,

@@ -71,17 +71,17 @@ void launch_relu_bwd_bias(const void* dy, const void* y, void* dx,
 int64_t fwd_chain_grid(int64_t M);
 void launch_swizzle_x(const void* x, void* out, int64_t M,
                       hipStream_t stream);
-void launch_swizzle_xt(const void* x, void* out, int64_t M,
+void launch_swizzle_xt(const void* x, void* out, int64_t M, int pi16,
                        hipStream_t stream);
 void launch_swizzle_x_both(const void* x, void* xs, void* xt, int64_t M,
-                           hipStream_t stream);
+                           int pi16, hipStream_t stream);
 void launch_fwd_chain(const void* x0s, const void* W1, const float* b1,
                       const void* W2, const float* b2, const void* W3,
                       const float* b3, const void* w4, const float* b4,
                       void* a1t, uint32_t* mask1, void* a2t,
                       uint32_t* mask2, void* a3, void* out,
                       const float* target, void* dyb, float* loss_part,
-                      int64_t M, hipStream_t stream);
+                      int64_t M, int pi16, hipStream_t stream);
 int64_t bwd_chain_grid(int64_t M);
 void launch_wgrad_frag(const void* AT, const void* BT, float* dW, int32_t N,
                        int32_t K, int64_t mchunks, int32_t nt_w,
@@ -93,7 +93,8 @@ void launch_slab_reduce(const float* part, float* out, int64_t nk,
 void launch_bwd_chain(const void* dy, const void* a3, const void* mask1,
                       const void* mask2, const void* w4, const void* W3T,
                       const void* W2T, void* dz1t, void* dz2t, void* dz3t,
-                      float* db_part, int64_t M, hipStream_t stream);
+                      float* db_part, int64_t M, int pi16,
+                      hipStream_t stream);
 
 namespace {
 
@@ -484,7 +485,7 @@ std::vector<at::Tensor> fwd_chain_bf16(
     const at::Tensor& W2, const at::Tensor& b2, const at::Tensor& W3,
     const at::Tensor& b3, const at::Tensor& w4, const at::Tensor& b4,
     const c10::optional<at::Tensor>& target,
-    const c10::optional<at::Tensor>& xt_out) {
+    const c10::optional<at::Tensor>& xt_out, bool pi16) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
                   x.dim() == 2 && x.size(1) == 100 && x.is_contiguous(),
               "fwd_chain: x must be contiguous bf16 [M,100]");
@@ -559,7 +560,8 @@ std::vector<at::Tensor> fwd_chain_bf16(
     if (xt_out.has_value()) {
       // one pass over x emits both the forward and the wgrad layouts
       launch_swizzle_x_both(x.data_ptr(), xs.data_ptr(),
-                            xt_out->data_ptr(), M, current_stream());
+                            xt_out->data_ptr(), M, pi16 ? 1 : 0,
+                            current_stream());
     } else {
       launch_swizzle_x(x.data_ptr(), xs.data_ptr(), M, current_stream());
     }
@@ -571,7 +573,7 @@ std::vector<at::Tensor> fwd_chain_bf16(
                      a2t.data_ptr(),
                      reinterpret_cast<uint32_t*>(mask2.data_ptr<int32_t>()),
                      a3.data_ptr(), out.data_ptr(), tgt_ptr, dyb_ptr,
-                     lp_ptr, M, current_stream());
+                     lp_ptr, M, pi16 ? 1 : 0, current_stream());
   }
   if (with_loss) return {a1t, mask1, a2t, mask2, a3, out, dyb, loss_part};
   return {a1t, mask1, a2t, mask2, a3, out};
@@ -596,7 +598,7 @@ static at::Tensor swizzle_frag_T(const at::Tensor& W) {
 std::vector<at::Tensor> bwd_chain_bf16(
     const at::Tensor& dy, const at::Tensor& a3, const at::Tensor& mask1,
     const at::Tensor& mask2, const at::Tensor& w4, const at::Tensor& W3,
-    const at::Tensor& W2) {
+    const at::Tensor& W2, bool pi16) {
   const int64_t M = dy.size(0);
   const int64_t mtiles = std::max<int64_t>((M + 31) / 32, 1);
   TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == at::kBFloat16 &&
@@ -653,7 +655,7 @@ std::vector<at::Tensor> bwd_chain_bf16(
                      mask2.data_ptr(), w4c.data_ptr(), W3Ts.data_ptr(),
                      W2Ts.data_ptr(), dz1t.data_ptr(), dz2t.data_ptr(),
                      dz3t.data_ptr(), db_part.data_ptr<float>(), M,
-                     current_stream());
+                     pi16 ? 1 : 0, current_stream());
   }
   // Column reduction with the split-slab reduce kernel (the GEMV ran at
   // ~0.9 TB/s; this streams the 28 MB slab near roofline). Pad columns
@@ -677,7 +679,7 @@ std::vector<at::Tensor> bwd_chain_bf16(
 }  // namespace
 // x [M,100] -> wgrad fragment-major x^T ([128/32][mchunks][2][32][8],
 // zero-padded cols 100..127 and rows past M) for the dW1 wgrad.
-at::Tensor swizzle_xt_bf16(const at::Tensor& x) {
+at::Tensor swizzle_xt_bf16(const at::Tensor& x, bool pi16) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
                   x.dim() == 2 && x.size(1) == 100 && x.is_contiguous(),
               "swizzle_xt: x must be contiguous bf16 [M,100]");
@@ -685,7 +687,8 @@ at::Tensor swizzle_xt_bf16(const at::Tensor& x) {
   const int64_t mtiles = std::max<int64_t>((M + 31) / 32, 1);
   auto out = at::empty({4 * mtiles * 2 * 512}, x.options());
   if (M > 0) {
-    launch_swizzle_xt(x.data_ptr(), out.data_ptr(), M, current_stream());
+    launch_swizzle_xt(x.data_ptr(), out.data_ptr(), M, pi16 ? 1 : 0,
+                      current_stream());
   }
   return out;
 }
@@ -747,7 +750,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad_bf16", &rsdl::wgrad_bf16, py::arg("dy"), py::arg("x"),
         py::arg("with_bias") = true);
   m.def("relu_bwd_bias", &rsdl::relu_bwd_bias, py::arg("dy"), py::arg("y"));
-  m.def("swizzle_xt_bf16", &rsdl::swizzle_xt_bf16, py::arg("x"));
+  m.def("swizzle_xt_bf16", &rsdl::swizzle_xt_bf16, py::arg("x"),
+        py::arg("pi16") = false);
   m.def("wgrad_frag_bf16", &rsdl::wgrad_frag_bf16, py::arg("AT"),
         py::arg("BT"), py::arg("N"), py::arg("K"), py::arg("mchunks"),
         py::arg("nt_w"), py::arg("kt_w"));
@@ -755,8 +759,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("W1"), py::arg("b1"), py::arg("W2"), py::arg("b2"),
         py::arg("W3"), py::arg("b3"), py::arg("w4"), py::arg("b4"),
         py::arg("target") = c10::nullopt,
-        py::arg("xt_out") = c10::nullopt);
-  m.def("bwd_chain_bf16", &rsdl::bwd_chain_bf16);
+        py::arg("xt_out") = c10::nullopt, py::arg("pi16") = false);
+  m.def("bwd_chain_bf16", &rsdl::bwd_chain_bf16, py::arg("dy"),
+        py::arg("a3"), py::arg("mask1"), py::arg("mask2"), py::arg("w4"),
+        py::arg("W3"), py::arg("W2"), py::arg("pi16") = false);
   m.attr("DT_F32") = (int)rsdl::DT_F32;
   m.attr("DT_F64") = (int)rsdl::DT_F64;
   m.attr("DT_I32") = (int)rsdl::DT_I32;

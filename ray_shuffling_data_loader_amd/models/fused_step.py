@@ -17,11 +17,20 @@ parity vs eager autograd) and the lane-level CPU simulation in
 tests/test_chain_sim.py.
 """
 
+import os
 from typing import Tuple
 
 import torch
 
 from ray_shuffling_data_loader_amd.models.mlp import TabularMLP
+
+# pi16 emission layout (RSDL_PI16=1): the chain kernels write each MFMA
+# half-wave's packed pairs directly as the transposed-fragment runs
+# under a shared intra-chunk M-permutation, dropping the cross-lane
+# exchange from both epilogues. dW is invariant (M is the contraction
+# dim of every consumer); schedule evidence in profiles/r02. Tests may
+# override via fused_step._PI16.
+_PI16 = os.environ.get("RSDL_PI16", "0") == "1"
 
 
 def _layers(model: TabularMLP):
@@ -88,13 +97,13 @@ def fused_step(
                      device=x.device)
     a1t, mask1, a2t, mask2, a3, out, dyb, loss_part = hip.fwd_chain_bf16(
         x, buf["W1p"], b1, buf["W2"], b2, buf["W3"], b3, buf["w4"], b4,
-        target=target, xt_out=xt,
+        target=target, xt_out=xt, pi16=_PI16,
     )
     loss = loss_part.sum() / M
     # Backward chain: consumes the masks (never the activations), emits
     # dz^T fragments; dW4/db4 partials fold into its seed loop.
     dz1t, dz2t, dz3t, db1, db2, db3, db4, dw4 = hip.bwd_chain_bf16(
-        dyb, a3, mask1, mask2, buf["w4"], buf["W3"], buf["W2"]
+        dyb, a3, mask1, mask2, buf["w4"], buf["W3"], buf["W2"], pi16=_PI16
     )
     # Weight grads: fragment-major MFMA wgrad kernel (csrc/wgrad_frag.hip)
     # reading the transposed fragments the producers emitted.

@@ -257,17 +257,29 @@ def wgrad(dy: torch.Tensor, x: torch.Tensor, with_bias: bool = True):
     return dw, db
 
 
-def t_frag_swizzle(t: torch.Tensor) -> torch.Tensor:
+# pi16 intra-chunk M-permutation: position p <-> row with bits 2 and 3
+# of the 4-bit within-chunk index swapped (an involution). Under it each
+# MFMA half-wave's own packed pairs are already contiguous 8-element
+# runs, so the chain kernels' emission needs no cross-lane exchange; dW
+# is invariant because M is the contraction dim of every consumer.
+_PI16_IDX = [(p & 3) | ((p & 4) << 1) | ((p & 8) >> 1) for p in range(16)]
+
+
+def t_frag_swizzle(t: torch.Tensor, pi16: bool = False) -> torch.Tensor:
     """[M, C] -> fragment-major layout of t^T: flat [C/32][Mp/16][2][32][8]
     (Mp = M padded to a 16-multiple with zero rows). This is the input
     layout of the fragment-major wgrad kernel (csrc/wgrad_frag.hip); the
     hot producers (bwd_chain dz^T, fwd_chain a^T) emit it directly — this
-    torch implementation is the oracle/fallback."""
+    torch implementation is the oracle/fallback. ``pi16`` applies the
+    pi16 intra-chunk M-permutation (must match the producers' flag)."""
     m, c = t.shape
     mp = (m + 15) // 16 * 16
     if mp != m:
         t = torch.nn.functional.pad(t, (0, 0, 0, mp - m))
-    v = t.reshape(mp // 16, 2, 8, c // 32, 32)
+    v = t.reshape(mp // 16, 16, c)
+    if pi16:
+        v = v[:, _PI16_IDX]
+    v = v.reshape(mp // 16, 2, 8, c // 32, 32)
     return v.permute(3, 0, 1, 4, 2).contiguous().reshape(-1)
 
 
@@ -287,14 +299,17 @@ def wgrad_frag(at_frag: torch.Tensor, bt_frag: torch.Tensor, n: int,
     return hip.wgrad_frag_bf16(at_frag, bt_frag, n, k, mchunks, nt_w, kt_w)
 
 
-def t_frag_unswizzle(flat: torch.Tensor, m: int, c: int) -> torch.Tensor:
+def t_frag_unswizzle(
+    flat: torch.Tensor, m: int, c: int, pi16: bool = False
+) -> torch.Tensor:
     """Inverse of :func:`t_frag_swizzle`: fragment-major transposed flat
     tensor -> row-major [m, c] (pad rows dropped)."""
     mchunks = flat.numel() // (c * 16)
     v = flat.reshape(c // 32, mchunks, 2, 32, 8)
-    return (
-        v.permute(1, 2, 4, 0, 3).reshape(mchunks * 16, c)[:m].contiguous()
-    )
+    v = v.permute(1, 2, 4, 0, 3).reshape(mchunks, 16, c)
+    if pi16:
+        v = v[:, _PI16_IDX]  # involution: same index recovers rows
+    return v.reshape(mchunks * 16, c)[:m].contiguous()
 
 
 def relu_mask_words(a: torch.Tensor) -> torch.Tensor:

@@ -13,25 +13,25 @@ from ray_shuffling_data_loader_amd.ops.shuffle_ops import (
 )
 
 
-def _swz32(t):
+def _swz32(t, pi16=False):
     """t_frag_swizzle with rows padded to a 32-multiple — the kernels'
     slab granularity (the torch helper alone pads only to 16)."""
     m = t.shape[0]
     mp = (m + 31) // 32 * 32
     if mp != m:
         t = torch.nn.functional.pad(t, (0, 0, 0, mp - m))
-    return t_frag_swizzle(t)
+    return t_frag_swizzle(t, pi16)
 
 
 class FakeHip:
     @staticmethod
     def fwd_chain_bf16(x, W1, b1, W2, b2, W3, b3, w4, b4, target=None,
-                       xt_out=None):
+                       xt_out=None, pi16=False):
         if xt_out is not None:
             m = x.shape[0]
             mp = (m + 31) // 32 * 32
             xt_out.copy_(t_frag_swizzle(
-                torch.nn.functional.pad(x, (0, 28, 0, mp - m))))
+                torch.nn.functional.pad(x, (0, 28, 0, mp - m)), pi16))
         if W1.shape[1] == 112:
             W1 = W1[:, :100]
         a1 = torch.relu(x.float() @ W1.float().t() + b1.float())
@@ -39,8 +39,8 @@ class FakeHip:
         a3 = torch.relu(a2 @ W3.float().t() + b3.float())
         out = a3 @ w4.float().unsqueeze(1) + b4.float()
         a1b, a2b = a1.bfloat16(), a2.bfloat16()
-        res = (_swz32(a1b), relu_mask_words(a1b.float()),
-               _swz32(a2b), relu_mask_words(a2b.float()),
+        res = (_swz32(a1b, pi16), relu_mask_words(a1b.float()),
+               _swz32(a2b, pi16), relu_mask_words(a2b.float()),
                a3.bfloat16(), out.bfloat16())
         if target is None:
             return res
@@ -50,7 +50,7 @@ class FakeHip:
         return res + (dyb, diff.square().sum().reshape(1))
 
     @staticmethod
-    def bwd_chain_bf16(dy, a3, mask1, mask2, w4, W3, W2):
+    def bwd_chain_bf16(dy, a3, mask1, mask2, w4, W3, W2, pi16=False):
         m_rows = dy.shape[0]
 
         def mask_of(words, n):
@@ -70,16 +70,17 @@ class FakeHip:
         da1 = dz2.float() @ W2.float()
         dz1 = (da1 * mask_of(mask1, 512)).bfloat16()
         dw4 = dy.float().t() @ a3.float()
-        return (_swz32(dz1), _swz32(dz2),
-                _swz32(dz3),
+        return (_swz32(dz1, pi16), _swz32(dz2, pi16),
+                _swz32(dz3, pi16),
                 dz1.float().sum(0), dz2.float().sum(0),
                 dz3.float().sum(0), dy.float().sum(0), dw4)
 
     @staticmethod
-    def swizzle_xt_bf16(x):
+    def swizzle_xt_bf16(x, pi16=False):
         m = x.shape[0]
         mp = (m + 31) // 32 * 32
-        return t_frag_swizzle(torch.nn.functional.pad(x, (0, 28, 0, mp - m)))
+        return t_frag_swizzle(
+            torch.nn.functional.pad(x, (0, 28, 0, mp - m)), pi16)
 
     @staticmethod
     def wgrad_frag_bf16(at_f, bt_f, n, k, mchunks, nt_w, kt_w):

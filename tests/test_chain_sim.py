@@ -331,3 +331,38 @@ def test_fused_step_flat_grad_views_cpu(monkeypatch):
         assert torch.equal(seg, p.grad), n  # still the view
         assert torch.allclose(p.grad, assigned[n], atol=1e-5), n
         off += p.numel()
+
+
+def test_fused_step_pi16_consistency():
+    """fused_step under RSDL_PI16 must produce the same loss and grads
+    as the default layout (FakeHip mirror; the permutation cancels in
+    every wgrad because M is the contraction dim)."""
+    import copy
+
+    import torch
+
+    from _fake_hip import FakeHip
+    from ray_shuffling_data_loader_amd.models import fused_step as fs
+    from ray_shuffling_data_loader_amd.models.mlp import TabularMLP
+
+    import ray_shuffling_data_loader_amd.ops.shuffle_ops as so
+
+    orig_load, orig_pi = so._load_hip, fs._PI16
+    so._load_hip = lambda: FakeHip
+    try:
+        torch.manual_seed(11)
+        m0 = TabularMLP(100)
+        m1 = copy.deepcopy(m0)
+        x = torch.randn(300, 100).bfloat16()
+        t = torch.randn(300, 1)
+        fs._PI16 = False
+        l0 = fs.fused_step(m0, x, t)
+        fs._PI16 = True
+        l1 = fs.fused_step(m1, x, t)
+        assert torch.allclose(l0, l1, rtol=1e-5)
+        for (n, p), (_, q) in zip(
+            m0.named_parameters(), m1.named_parameters()
+        ):
+            assert torch.allclose(p.grad, q.grad, atol=1e-4), n
+    finally:
+        so._load_hip, fs._PI16 = orig_load, orig_pi

@@ -875,3 +875,114 @@ def test_fused_step_edge_batch_sizes(dev):
             # component oracles are the precision checks.
             tol = 0.3 * q.grad.abs().max().clamp(min=1e-5) + 2e-2
             assert err <= tol, (m_rows, n, err.item())
+
+
+@pytest.mark.gpu
+def test_pi16_layout_and_chain_consistency(dev):
+    """pi16 emission layout (exchange-free epilogues): the swizzle must
+    bit-match the torch pi16 oracle, and the chain kernels' pi16 outputs
+    must be the SAME VALUES as the default layout, just permuted — so
+    after unswizzling with the matching flag everything is bit-equal
+    (the arithmetic is identical; only the storage order changes)."""
+    from ray_shuffling_data_loader_amd.ops import shuffle_ops
+    from ray_shuffling_data_loader_amd.ops.shuffle_ops import (
+        t_frag_swizzle,
+        t_frag_unswizzle,
+    )
+
+    hip = shuffle_ops._load_hip()
+    torch.manual_seed(13)
+    M = 4096 + 17
+    x = torch.randn(M, 100, device=dev).bfloat16()
+
+    mp = (M + 31) // 32 * 32
+    xpad = torch.nn.functional.pad(x, (0, 28, 0, mp - M))
+    assert torch.equal(
+        hip.swizzle_xt_bf16(x, pi16=True), t_frag_swizzle(xpad, True)
+    ), "standalone swizzle x^T pi16 layout"
+
+    Ws = [
+        (torch.randn(512, 100, device=dev) / 11).bfloat16(),
+        (torch.randn(256, 512, device=dev) / 16).bfloat16(),
+        (torch.randn(128, 256, device=dev) / 22).bfloat16(),
+        (torch.randn(1, 128, device=dev) / 11).bfloat16(),
+    ]
+    bs = [
+        torch.randn(n, device=dev).bfloat16() for n in (512, 256, 128, 1)
+    ]
+    tgt = torch.randn(M, 1, device=dev)
+    mchunks = 2 * ((M + 31) // 32)
+
+    outs = {}
+    for pi in (False, True):
+        xt = torch.empty(
+            4 * mchunks * 512, dtype=torch.bfloat16, device=dev
+        )
+        outs[pi] = hip.fwd_chain_bf16(
+            x, Ws[0], bs[0], Ws[1], bs[1], Ws[2], bs[2],
+            Ws[3].flatten(), bs[3], target=tgt, xt_out=xt, pi16=pi,
+        ) + (xt,)
+    a1t0, m10, a2t0, m20, a30, out0, dyb0, lp0, xt0 = outs[False]
+    a1t1, m11, a2t1, m21, a31, out1, dyb1, lp1, xt1 = outs[True]
+    # masks / row-major outputs: identical values, identical layout
+    for a, b, name in [(m10, m11, "mask1"), (m20, m21, "mask2"),
+                       (a30, a31, "a3"), (out0, out1, "out"),
+                       (dyb0, dyb1, "dyb"), (lp0, lp1, "loss_part")]:
+        assert torch.equal(a, b), name
+    # transposed emissions: bit-equal after layout-aware unswizzle
+    mpad = mchunks * 16
+    for t0, t1, c, name in [(a1t0, a1t1, 512, "a1t"),
+                            (a2t0, a2t1, 256, "a2t"),
+                            (xt0, xt1, 128, "xt")]:
+        assert torch.equal(
+            t_frag_unswizzle(t0, mpad, c),
+            t_frag_unswizzle(t1, mpad, c, pi16=True),
+        ), name
+
+    bw = {}
+    for pi in (False, True):
+        bw[pi] = hip.bwd_chain_bf16(
+            dyb0, a30, m10, m20, Ws[3].flatten(), Ws[2], Ws[1], pi16=pi
+        )
+    for i, name in [(3, "db1"), (4, "db2"), (5, "db3"), (6, "db4"),
+                    (7, "dw4")]:
+        assert torch.equal(bw[False][i], bw[True][i]), name
+    for i, c, name in [(0, 512, "dz1t"), (1, 256, "dz2t"),
+                       (2, 128, "dz3t")]:
+        assert torch.equal(
+            t_frag_unswizzle(bw[False][i], mpad, c),
+            t_frag_unswizzle(bw[True][i], mpad, c, pi16=True),
+        ), name
+
+
+@pytest.mark.gpu
+def test_pi16_fused_step_parity(dev):
+    """Whole fused step under RSDL_PI16: same loss, grads within the
+    wgrad reduction tolerance of the default layout (the MFMA k-order
+    over M differs under the permutation, so dW is equal only to
+    rounding)."""
+    import copy
+
+    from ray_shuffling_data_loader_amd.models import fused_step as fs
+    from ray_shuffling_data_loader_amd.models.mlp import TabularMLP
+
+    torch.manual_seed(14)
+    M = 1 << 15
+    m0 = TabularMLP(100).to(dev)
+    m1 = copy.deepcopy(m0)
+    x = torch.randn(M, 100, device=dev).bfloat16()
+    t = torch.randn(M, 1, device=dev)
+    orig = fs._PI16
+    try:
+        fs._PI16 = False
+        l0 = fs.fused_step(m0, x, t)
+        fs._PI16 = True
+        l1 = fs.fused_step(m1, x, t)
+    finally:
+        fs._PI16 = orig
+    assert torch.allclose(l0, l1, rtol=1e-5)
+    for (n, p), (_, q) in zip(
+        m0.named_parameters(), m1.named_parameters()
+    ):
+        tol = 0.02 * q.grad.abs().max().clamp(min=1e-5) + 1e-2
+        assert (p.grad - q.grad).abs().max() <= tol, n

@@ -175,3 +175,36 @@ def test_relu_mask_words_bits():
         # pad bits zero
         rest = ((bits >> sh) & 1).reshape(mt * 32, n)[m:]
         assert torch.all(rest == 0)
+
+
+def test_t_frag_swizzle_pi16_roundtrip_and_invariance():
+    """pi16 layout: swizzle/unswizzle round-trips, and the wgrad
+    contraction is invariant when BOTH operands share the permutation
+    (M is the contraction dim)."""
+    import torch
+
+    from ray_shuffling_data_loader_amd.ops.shuffle_ops import (
+        _PI16_IDX,
+        t_frag_swizzle,
+        t_frag_unswizzle,
+    )
+
+    # involution with bits 2<->3 swapped
+    assert [_PI16_IDX[p] for p in _PI16_IDX] == list(range(16))
+    x = torch.randn(50, 64)
+    for pi in (False, True):
+        r = t_frag_unswizzle(t_frag_swizzle(x, pi), 50, 64, pi)
+        assert torch.equal(r, x), pi
+    # mixing flags must scramble rows (the layouts really differ)
+    r = t_frag_unswizzle(t_frag_swizzle(x, True), 50, 64, False)
+    assert not torch.equal(r, x)
+
+    a, b = torch.randn(50, 32).bfloat16(), torch.randn(50, 64).bfloat16()
+    d = {}
+    for pi in (False, True):
+        sa, sb = t_frag_swizzle(a, pi), t_frag_swizzle(b, pi)
+        d[pi] = (
+            t_frag_unswizzle(sa, 64, 32).float().t()
+            @ t_frag_unswizzle(sb, 64, 64).float()
+        )
+    assert torch.allclose(d[False], d[True], atol=1e-4)
