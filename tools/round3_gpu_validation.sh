@@ -27,10 +27,16 @@ for S in 0 1; do
     | tail -6 | tee gpurun_out/r3_step_sched$S.txt
 done
 
-# 5. Driver-flag bench A/B (one run each; box-to-box band is ~±2%).
-for S in 0 1; do
-  RSDL_WGRAD_SCHED=$S timeout 180 python bench.py --gpus 1 --steps 20 --warmup 5 \
-    > gpurun_out/r3_bench_sched$S.json 2>gpurun_out/r3_bench_sched$S.err
+# 5. pi16 A/B: exchange-free chain epilogues (fwd -71 VALU/wave, bwd
+#    -101; profiles/r02/wgrad_sched_asm.md + commit message).
+RSDL_PI16=1 timeout 240 python tools/profile_fused_step.py 2>&1 \
+  | tail -6 | tee gpurun_out/r3_step_pi16.txt
+
+# 6. Driver-flag bench A/B over the three knob settings.
+for CFG in "" "RSDL_WGRAD_SCHED=1" "RSDL_WGRAD_SCHED=1 RSDL_PI16=1"; do
+  TAG=$(echo "$CFG" | tr -cd '01' | head -c 8)
+  env $CFG timeout 180 python bench.py --gpus 1 --steps 20 --warmup 5 \
+    > "gpurun_out/r3_bench_${TAG:-base}.json" 2>"gpurun_out/r3_bench_${TAG:-base}.err"
 done
-tail -1 gpurun_out/r3_bench_sched0.json
-tail -1 gpurun_out/r3_bench_sched1.json
+tail -1 gpurun_out/r3_bench_base.json
+cat gpurun_out/r3_bench_*.json | tail -3
