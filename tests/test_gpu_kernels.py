@@ -986,3 +986,48 @@ def test_pi16_fused_step_parity(dev):
     ):
         tol = 0.02 * q.grad.abs().max().clamp(min=1e-5) + 1e-2
         assert (p.grad - q.grad).abs().max() <= tol, n
+
+
+@pytest.mark.gpu
+def test_wgrad_sched_variant_numerics(dev):
+    """RSDL_WGRAD_SCHED=1 selects the pinned-schedule wgrad_frag +
+    slab_reduce instantiations (restructured steady loops — see
+    profiles/r02/wgrad_sched_asm.md). The env is latched at extension
+    load, so validate in a subprocess: fragment wgrad vs the torch
+    oracle at all three flagship shapes."""
+    import os
+    import subprocess
+    import sys
+
+    script = r"""
+import torch, sys
+from ray_shuffling_data_loader_amd.ops.shuffle_ops import (
+    t_frag_swizzle, wgrad_frag)
+torch.manual_seed(21)
+M = 8192 + 23
+mchunks = 2 * ((M + 31) // 32)
+mp = mchunks * 16
+for n, k in [(512, 128), (256, 512), (128, 256)]:
+    dz = torch.randn(M, n, device="cuda").bfloat16()
+    src = torch.randn(M, k, device="cuda").bfloat16()
+    pad = lambda t: torch.nn.functional.pad(t, (0, 0, 0, mp - M))
+    got = wgrad_frag(t_frag_swizzle(pad(dz)), t_frag_swizzle(pad(src)),
+                     n, k, mchunks)
+    ref = dz.float().t() @ src.float()
+    err = (got - ref).abs().max().item()
+    tol = 0.05 * ref.abs().max().item() + 2.0
+    assert err <= tol, (n, k, err, tol)
+print("SCHED_OK")
+"""
+    env = dict(os.environ, RSDL_WGRAD_SCHED="1")
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    proc = subprocess.run(
+        [sys.executable, "-c", script],
+        capture_output=True,
+        text=True,
+        timeout=300,
+        env=env,
+        cwd=repo,
+    )
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    assert "SCHED_OK" in proc.stdout
