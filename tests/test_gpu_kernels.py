@@ -918,10 +918,10 @@ def test_pi16_layout_and_chain_consistency(dev):
         xt = torch.empty(
             4 * mchunks * 512, dtype=torch.bfloat16, device=dev
         )
-        outs[pi] = hip.fwd_chain_bf16(
+        outs[pi] = tuple(hip.fwd_chain_bf16(
             x, Ws[0], bs[0], Ws[1], bs[1], Ws[2], bs[2],
             Ws[3].flatten(), bs[3], target=tgt, xt_out=xt, pi16=pi,
-        ) + (xt,)
+        )) + (xt,)
     a1t0, m10, a2t0, m20, a30, out0, dyb0, lp0, xt0 = outs[False]
     a1t1, m11, a2t1, m21, a31, out1, dyb1, lp1, xt1 = outs[True]
     # masks / row-major outputs: identical values, identical layout
