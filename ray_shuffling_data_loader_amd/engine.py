@@ -413,13 +413,18 @@ class ShuffleEngine:
             except BaseException as e:  # surfaced via join() AND the queue
                 self._error = e
                 failure = ShuffleEngineFailure(e)
-                try:
-                    for epoch in range(self.start_epoch, self.num_epochs):
-                        for t in self.owned_trainers:
+                # Per-(epoch, trainer) scope: delivering into an epoch that
+                # was already consumed/evicted (or a shut-down queue) may
+                # raise, and that must not stop the LATER epochs — the ones
+                # consumers are actually blocked on — from getting their
+                # failure marker.
+                for epoch in range(self.start_epoch, self.num_epochs):
+                    for t in self.owned_trainers:
+                        try:
                             self.consumer.consume(t, epoch, [failure])
                             self.consumer.producer_done(t, epoch)
-                except BaseException:
-                    pass  # queue may already be shut down
+                        except BaseException:
+                            pass
 
         self._thread = threading.Thread(
             target=_target, name=f"rsdl-shuffle-r{self.rank}", daemon=True

@@ -350,3 +350,52 @@ def test_torch_dataset_feature_matrix_with_key(tmp_path):
         assert data[0].abs().max() <= 1.0
         total += len(target)
     assert total == 4000
+
+
+def test_engine_failure_reaches_later_epochs(small_data):
+    """A mid-run engine death must deliver failure markers to EVERY
+    remaining epoch — a raise while delivering into an already-evicted
+    epoch must not truncate the distribution (consumers blocked on later
+    epochs would hang forever)."""
+    from ray_shuffling_data_loader_amd.engine import (
+        ShuffleEngine,
+        ShuffleEngineFailure,
+    )
+
+    filenames, _n = small_data
+
+    class Sink:
+        def __init__(self):
+            self.got = {}
+
+        def consume(self, rank, epoch, batches):
+            if epoch == 0:
+                # emulate the first epoch's queue being gone already
+                raise RuntimeError("epoch 0 evicted")
+            self.got.setdefault(epoch, []).extend(batches)
+
+        def producer_done(self, rank, epoch):
+            pass
+
+        def wait_until_ready(self, epoch):
+            if epoch == 0:
+                raise ValueError("boom: engine dies at the gate")
+
+        def wait_until_all_epochs_done(self):
+            pass
+
+    sink = Sink()
+    eng = ShuffleEngine(
+        list(filenames), sink, num_epochs=3, num_reducers=2,
+        num_trainers=1,
+    )
+    eng.start()
+    eng._thread.join(timeout=60)
+    assert eng._error is not None
+    # epochs 1 and 2 both received the failure marker despite epoch 0's
+    # consume raising
+    for epoch in (1, 2):
+        assert epoch in sink.got, sink.got.keys()
+        assert any(
+            isinstance(b, ShuffleEngineFailure) for b in sink.got[epoch]
+        ), epoch
