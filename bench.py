@@ -364,6 +364,21 @@ def main():
 
     if use_graph:
         one_step = one_step_graphed
+    elif is_cuda and os.environ.get("RSDL_HIPRI_STEP", "0") == "1":
+        # A/B knob: run the train step on a HIGH-priority stream so its
+        # kernels win CU arbitration against the side-stream reshuffle on
+        # rollover-adjacent steps (the ~15% gap between the 20-step driver
+        # window and steady state — profiles/PERF.md). Dequeued batches
+        # record_stream() against the current stream at dequeue time, so
+        # dequeuing inside the context keeps allocator safety.
+        _hipri = torch.cuda.Stream(device=device, priority=-1)
+        _base_step = one_step
+
+        def one_step_hipri():
+            with torch.cuda.stream(_hipri):
+                return _base_step()
+
+        one_step = one_step_hipri
 
     # Hang watchdog: if the whole job wedges (e.g. an RCCL collective
     # deadlock at N>1), dump every thread's stack to stderr so the failure
@@ -551,6 +566,17 @@ def main():
                 ),
             },
         }
+        # Self-describing A/B records: any non-default RSDL_* knob that
+        # was set is named in the JSON (empty when absent = stock run).
+        knobs = {
+            k: v
+            for k, v in sorted(os.environ.items())
+            if k.startswith("RSDL_")
+            and k
+            not in ("RSDL_NCCL_DEBUG", "RSDL_WATCHDOG_S", "RSDL_TUNABLEOP")
+        }
+        if knobs:
+            result["config"]["knobs"] = knobs
         print(json.dumps(result), flush=True)
 
     if world > 1:
