@@ -33,7 +33,7 @@ RSDL_PI16=1 timeout 240 python tools/profile_fused_step.py 2>&1 \
   | tail -6 | tee gpurun_out/r3_step_pi16.txt
 
 # 6. Driver-flag bench A/B over the three knob settings.
-for CFG in "" "RSDL_WGRAD_SCHED=1" "RSDL_WGRAD_SCHED=1 RSDL_PI16=1"; do
+for CFG in "" "RSDL_WGRAD_SCHED=1" "RSDL_WGRAD_SCHED=1 RSDL_PI16=1" "RSDL_HIPRI_STEP=1"; do
   TAG=$(echo "$CFG" | tr -cd '01' | head -c 8)
   env $CFG timeout 180 python bench.py --gpus 1 --steps 20 --warmup 5 \
     > "gpurun_out/r3_bench_${TAG:-base}.json" 2>"gpurun_out/r3_bench_${TAG:-base}.err"
