@@ -415,6 +415,17 @@ void launch_wgrad_frag(const void* AT, const void* BT, float* dW, int32_t N,
                        reinterpret_cast<const short*>(AT),
                        reinterpret_cast<const short*>(BT), dW, N, K, mchunks,
                        nb_n, nblk_k, nslabs, chunks_per_slab);
+  } else if (nt_w == 1 && kt_w == 4) {
+    // Small-tile config (RSDL_WGRAD_SMALL_TILES): 64 acc regs -> 3
+    // waves/SIMD, and NT_W+KT_W<=6 takes the depth-3 fenced loop when
+    // RSDL_WGRAD_SCHED is also set.
+    auto kern = sched ? wgrad_frag_kernel<1, 4, 3, 1>
+                      : wgrad_frag_kernel<1, 4, 3>;
+    hipLaunchKernelGGL(kern, dim3((uint32_t)grid),
+                       dim3(256), 0, stream,
+                       reinterpret_cast<const short*>(AT),
+                       reinterpret_cast<const short*>(BT), dW, N, K, mchunks,
+                       nb_n, nblk_k, nslabs, chunks_per_slab);
   }
 }
 

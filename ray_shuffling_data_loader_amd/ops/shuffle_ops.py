@@ -289,12 +289,25 @@ _WGRAD_FRAG_CFG = {
     (256, 512): (1, 8),
     (128, 256): (1, 8),
 }
+# RSDL_WGRAD_SMALL_TILES=1: (1,4) tiles for the (1,8) shapes — 64 acc
+# regs (3 waves/SIMD), and the depth-3 fenced loop when RSDL_WGRAD_SCHED
+# is also on (csrc/wgrad_frag.hip launch configs).
+_WGRAD_FRAG_SMALL = {
+    (512, 128): (2, 4),
+    (256, 512): (1, 4),
+    (128, 256): (1, 4),
+}
 
 
 def wgrad_frag(at_frag: torch.Tensor, bt_frag: torch.Tensor, n: int,
                k: int, mchunks: int) -> torch.Tensor:
     """dW = dz^T @ src from pre-swizzled fragment inputs (fp32 [N,K])."""
-    nt_w, kt_w = _WGRAD_FRAG_CFG[(n, k)]
+    cfg = (
+        _WGRAD_FRAG_SMALL
+        if os.environ.get("RSDL_WGRAD_SMALL_TILES", "0") == "1"
+        else _WGRAD_FRAG_CFG
+    )
+    nt_w, kt_w = cfg[(n, k)]
     hip = _load_hip()
     return hip.wgrad_frag_bf16(at_frag, bt_frag, n, k, mchunks, nt_w, kt_w)
 

@@ -1019,15 +1019,16 @@ for n, k in [(512, 128), (256, 512), (128, 256)]:
     assert err <= tol, (n, k, err, tol)
 print("SCHED_OK")
 """
-    env = dict(os.environ, RSDL_WGRAD_SCHED="1")
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    proc = subprocess.run(
-        [sys.executable, "-c", script],
-        capture_output=True,
-        text=True,
-        timeout=300,
-        env=env,
-        cwd=repo,
-    )
-    assert proc.returncode == 0, proc.stderr[-2000:]
-    assert "SCHED_OK" in proc.stdout
+    for extra in ({}, {"RSDL_WGRAD_SMALL_TILES": "1"}):
+        env = dict(os.environ, RSDL_WGRAD_SCHED="1", **extra)
+        proc = subprocess.run(
+            [sys.executable, "-c", script],
+            capture_output=True,
+            text=True,
+            timeout=300,
+            env=env,
+            cwd=repo,
+        )
+        assert proc.returncode == 0, (extra, proc.stderr[-2000:])
+        assert "SCHED_OK" in proc.stdout, extra
