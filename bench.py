@@ -227,19 +227,40 @@ def main():
         and args.num_cols == 100
     )
     flat_grad = None
+    grad_regions = {}
     if use_fused:
         if world > 1:
             params = list(model.parameters())
             # DDP would broadcast at wrap time; do it explicitly here.
             for p in params:
                 torch.distributed.broadcast(p.data, src=0)
+            # Flat layout [W1, W2, W3, W4, b1..b4]: weights first so each
+            # of fused_step's grad_hook stages ("w1"/"w2"/"w3" after the
+            # corresponding wgrad kernel, "bias" = W4 + every bias right
+            # after the backward chain) maps to ONE contiguous slice —
+            # RSDL_OVERLAP_ALLREDUCE=1 all-reduces each slice async while
+            # the remaining wgrad kernels run. Order is irrelevant to the
+            # default single-collective path.
+            weights = [p for n, p in model.named_parameters()
+                       if n.endswith("weight")]
+            biases = [p for n, p in model.named_parameters()
+                      if n.endswith("bias")]
+            ordered = weights + biases
             flat_grad = torch.zeros(
-                sum(p.numel() for p in params), device=device
+                sum(p.numel() for p in ordered), device=device
             )
             off = 0
-            for p in params:
+            offs = []
+            for p in ordered:
                 p.grad = flat_grad[off : off + p.numel()].view_as(p)
+                offs.append(off)
                 off += p.numel()
+            grad_regions = {
+                "w1": (offs[0], offs[1]),
+                "w2": (offs[1], offs[2]),
+                "w3": (offs[2], offs[3]),
+                "bias": (offs[3], off),  # W4 + b1..b4
+            }
             model._rsdl_flat_grads = True
     elif world > 1:
         model = torch.nn.parallel.DistributedDataParallel(model)
@@ -285,6 +306,23 @@ def main():
             fused_step,
         )
 
+    # RSDL_OVERLAP_ALLREDUCE=1 (flat-grad DP only): all-reduce each grad
+    # region async as fused_step reports it ready — the bias+head slice
+    # overlaps all three wgrad kernels, dW1's overlaps dW2+dW3's.
+    overlap_ar = (
+        flat_grad is not None
+        and os.environ.get("RSDL_OVERLAP_ALLREDUCE", "0") == "1"
+    )
+    ar_works = []
+
+    def _overlap_hook(stage):
+        s, e = grad_regions[stage]
+        sl = flat_grad[s:e]
+        sl.div_(world)
+        ar_works.append(
+            torch.distributed.all_reduce(sl, async_op=True)
+        )
+
     def one_step():
         t_wait0 = time.perf_counter()
         cur_epoch[0], (data, target) = next(it)
@@ -294,10 +332,16 @@ def main():
             x = x.to(device, non_blocking=True)
             target = target.to(device, non_blocking=True)
         if use_fused:
-            fused_step(model, x, target)
-            if flat_grad is not None:
-                flat_grad.div_(world)
-                torch.distributed.all_reduce(flat_grad)
+            if overlap_ar:
+                ar_works.clear()
+                fused_step(model, x, target, grad_hook=_overlap_hook)
+                for w in ar_works:
+                    w.wait()
+            else:
+                fused_step(model, x, target)
+                if flat_grad is not None:
+                    flat_grad.div_(world)
+                    torch.distributed.all_reduce(flat_grad)
             opt.step()
             return wait
         opt.zero_grad(set_to_none=True)

@@ -73,12 +73,20 @@ def _weight_buffers(model: TabularMLP, lin):
 
 
 def fused_step(
-    model: TabularMLP, x: torch.Tensor, target: torch.Tensor
+    model: TabularMLP, x: torch.Tensor, target: torch.Tensor,
+    grad_hook=None,
 ) -> Tuple[torch.Tensor, None]:
     """One manual fwd+bwd: computes the MSE loss and POPULATES .grad on
     every parameter of ``model`` (fp32, ready for an optimizer step).
     ``x`` is the bf16 [M,100] feature batch; ``target`` is [M,1].
-    Returns the (scalar fp32) loss."""
+    Returns the (scalar fp32) loss.
+
+    ``grad_hook`` (flat-grad DP mode): called as grads become ready —
+    ``grad_hook("bias")`` once every bias grad AND the head weight grad
+    are in their views (right after the backward chain, BEFORE the
+    wgrad kernels), then ``grad_hook("w1"|"w2"|"w3")`` after each weight
+    grad copy. Lets the caller overlap per-slice gradient collectives
+    with the remaining wgrad kernels (bench RSDL_OVERLAP_ALLREDUCE)."""
     from ray_shuffling_data_loader_amd.ops.shuffle_ops import _load_hip
 
     hip = _load_hip()
@@ -105,22 +113,39 @@ def fused_step(
     dz1t, dz2t, dz3t, db1, db2, db3, db4, dw4 = hip.bwd_chain_bf16(
         dyb, a3, mask1, mask2, buf["w4"], buf["W3"], buf["W2"], pi16=_PI16
     )
+    # The DP bench pre-creates .grad as views of one flat buffer (so the
+    # gradient all-reduce is a single collective) and marks the model;
+    # only then do we COPY into them. Otherwise assign the fresh tensors
+    # (copying would add 8 small kernels per step at N=1).
+    flat_mode = getattr(model, "_rsdl_flat_grads", False)
+    if flat_mode and grad_hook is not None:
+        # Bias + head grads are ready NOW — copy and signal so their
+        # collective overlaps the wgrad kernels below.
+        for m, gb in zip(lin, (db1, db2, db3, db4)):
+            m.bias.grad.copy_(gb.reshape(m.bias.shape))
+        lin[3].weight.grad.copy_(dw4)
+        grad_hook("bias")
     # Weight grads: fragment-major MFMA wgrad kernel (csrc/wgrad_frag.hip)
     # reading the transposed fragments the producers emitted.
     dw1 = wgrad_frag(dz1t, xt, 512, 128, mchunks)[:, :100].contiguous()
+    if flat_mode and grad_hook is not None:
+        lin[0].weight.grad.copy_(dw1)
+        grad_hook("w1")
     dw2 = wgrad_frag(dz2t, a1t, 256, 512, mchunks)
+    if flat_mode and grad_hook is not None:
+        lin[1].weight.grad.copy_(dw2)
+        grad_hook("w2")
     dw3 = wgrad_frag(dz3t, a2t, 128, 256, mchunks)
+    if flat_mode and grad_hook is not None:
+        lin[2].weight.grad.copy_(dw3)
+        grad_hook("w3")
+        return loss
     grads = [
         (dw1, db1),
         (dw2, db2),
         (dw3, db3),
         (dw4, db4),
     ]
-    # The DP bench pre-creates .grad as views of one flat buffer (so the
-    # gradient all-reduce is a single collective) and marks the model;
-    # only then do we COPY into them. Otherwise assign the fresh tensors
-    # (copying would add 8 small kernels per step at N=1).
-    flat_mode = getattr(model, "_rsdl_flat_grads", False)
     for m, (gw, gb) in zip(lin, grads):
         gb = gb.reshape(m.bias.shape)
         if flat_mode:
