@@ -366,3 +366,48 @@ def test_fused_step_pi16_consistency():
             assert torch.allclose(p.grad, q.grad, atol=1e-4), n
     finally:
         so._load_hip, fs._PI16 = orig_load, orig_pi
+
+
+def test_fused_step_pi16_with_grad_hook():
+    """Knob composition: pi16 emission + the overlap grad_hook must
+    together produce the same grads as the plain step (FakeHip)."""
+    import copy
+
+    import torch
+
+    from _fake_hip import FakeHip
+    from ray_shuffling_data_loader_amd.models import fused_step as fs
+    from ray_shuffling_data_loader_amd.models.mlp import TabularMLP
+    import ray_shuffling_data_loader_amd.ops.shuffle_ops as so
+
+    orig_load, orig_pi = so._load_hip, fs._PI16
+    so._load_hip = lambda: FakeHip
+    try:
+        torch.manual_seed(17)
+        m0 = TabularMLP(100)
+        m1 = copy.deepcopy(m0)
+        x = torch.randn(200, 100).bfloat16()
+        t = torch.randn(200, 1)
+
+        def flatten(model):
+            params = list(model.parameters())
+            flat = torch.zeros(sum(p.numel() for p in params))
+            off = 0
+            for p in params:
+                p.grad = flat[off : off + p.numel()].view_as(p)
+                off += p.numel()
+            model._rsdl_flat_grads = True
+            return flat
+
+        f0 = flatten(m0)
+        f1 = flatten(m1)
+        fs._PI16 = False
+        l0 = fs.fused_step(m0, x, t)
+        stages = []
+        fs._PI16 = True
+        l1 = fs.fused_step(m1, x, t, grad_hook=stages.append)
+        assert stages == ["bias", "w1", "w2", "w3"]
+        assert torch.allclose(l0, l1, rtol=1e-5)
+        assert torch.allclose(f0, f1, atol=1e-4)
+    finally:
+        so._load_hip, fs._PI16 = orig_load, orig_pi
