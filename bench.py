@@ -69,6 +69,34 @@ from ray_shuffling_data_loader_amd.models.mlp import TabularMLP
 from ray_shuffling_data_loader_amd.torch_dataset import TorchShufflingDataset
 
 
+def flat_grad_views(model, device):
+    """Pre-create every param's .grad as a view of ONE flat buffer, laid
+    out [W1, W2, W3, W4, b1..b4] so each of fused_step's grad_hook stages
+    ("w1"/"w2"/"w3" after the corresponding wgrad kernel, "bias" = W4 +
+    every bias right after the backward chain) maps to ONE contiguous
+    slice. Returns (flat, regions). Order is irrelevant to the default
+    single-collective path."""
+    weights = [p for n, p in model.named_parameters()
+               if n.endswith("weight")]
+    biases = [p for n, p in model.named_parameters()
+              if n.endswith("bias")]
+    ordered = weights + biases
+    flat = torch.zeros(sum(p.numel() for p in ordered), device=device)
+    off = 0
+    offs = []
+    for p in ordered:
+        p.grad = flat[off : off + p.numel()].view_as(p)
+        offs.append(off)
+        off += p.numel()
+    regions = {
+        "w1": (offs[0], offs[1]),
+        "w2": (offs[1], offs[2]),
+        "w3": (offs[2], offs[3]),
+        "bias": (offs[3], off),  # W4 + b1..b4
+    }
+    return flat, regions
+
+
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
@@ -234,33 +262,7 @@ def main():
             # DDP would broadcast at wrap time; do it explicitly here.
             for p in params:
                 torch.distributed.broadcast(p.data, src=0)
-            # Flat layout [W1, W2, W3, W4, b1..b4]: weights first so each
-            # of fused_step's grad_hook stages ("w1"/"w2"/"w3" after the
-            # corresponding wgrad kernel, "bias" = W4 + every bias right
-            # after the backward chain) maps to ONE contiguous slice —
-            # RSDL_OVERLAP_ALLREDUCE=1 all-reduces each slice async while
-            # the remaining wgrad kernels run. Order is irrelevant to the
-            # default single-collective path.
-            weights = [p for n, p in model.named_parameters()
-                       if n.endswith("weight")]
-            biases = [p for n, p in model.named_parameters()
-                      if n.endswith("bias")]
-            ordered = weights + biases
-            flat_grad = torch.zeros(
-                sum(p.numel() for p in ordered), device=device
-            )
-            off = 0
-            offs = []
-            for p in ordered:
-                p.grad = flat_grad[off : off + p.numel()].view_as(p)
-                offs.append(off)
-                off += p.numel()
-            grad_regions = {
-                "w1": (offs[0], offs[1]),
-                "w2": (offs[1], offs[2]),
-                "w3": (offs[2], offs[3]),
-                "bias": (offs[3], off),  # W4 + b1..b4
-            }
+            flat_grad, grad_regions = flat_grad_views(model, device)
             model._rsdl_flat_grads = True
     elif world > 1:
         model = torch.nn.parallel.DistributedDataParallel(model)

@@ -110,3 +110,43 @@ def test_bench_world4_gloo(tmp_path):
     assert d["n_gpus"] == 4
     assert d["config"]["parallelism"] == "dp4"
     assert d["value"] > 0
+
+
+def test_flat_grad_views_regions():
+    """bench.flat_grad_views: every param's .grad is a view of the flat
+    buffer, the hook regions are a disjoint cover, and each region holds
+    exactly the params fused_step populates at that stage."""
+    import sys as _sys
+
+    import torch
+
+    _sys.path.insert(0, REPO)
+    import bench
+    from ray_shuffling_data_loader_amd.models.mlp import TabularMLP
+
+    model = TabularMLP(100)
+    flat, regions = bench.flat_grad_views(model, torch.device("cpu"))
+    assert flat.numel() == sum(p.numel() for p in model.parameters())
+    # disjoint cover in order w1,w2,w3,bias
+    spans = [regions[k] for k in ("w1", "w2", "w3", "bias")]
+    assert spans[0][0] == 0 and spans[-1][1] == flat.numel()
+    for (a, b), (c, d) in zip(spans, spans[1:]):
+        assert b == c
+    # writing a param grad shows up in its region only
+    named = dict(model.named_parameters())
+    stage_of = {
+        "net.0.weight": "w1",
+        "net.1.weight": "w2",
+        "net.2.weight": "w3",
+        "net.3.weight": "bias",  # W4 rides with the biases
+        "net.0.bias": "bias",
+        "net.1.bias": "bias",
+        "net.2.bias": "bias",
+        "net.3.bias": "bias",
+    }
+    for name, p in named.items():
+        flat.zero_()
+        p.grad.fill_(1.0)
+        s, e = regions[stage_of[name]]
+        assert flat[s:e].abs().sum() == p.numel(), name
+        assert flat.abs().sum() == p.numel(), name
