@@ -946,7 +946,12 @@ def test_pi16_layout_and_chain_consistency(dev):
         )
     for i, name in [(3, "db1"), (4, "db2"), (5, "db3"), (6, "db4"),
                     (7, "dw4")]:
-        assert torch.equal(bw[False][i], bw[True][i]), name
+        # db/dw4 go through slab_reduce, whose split partials combine
+        # with fp32 atomics — bitwise nondeterministic ORDER between
+        # runs, so compare at ulp scale rather than torch.equal.
+        a, b = bw[False][i], bw[True][i]
+        tol = 1e-5 * b.abs().max().clamp(min=1.0) + 1e-5
+        assert (a - b).abs().max() <= tol, name
     for i, c, name in [(0, 512, "dz1t"), (1, 256, "dz2t"),
                        (2, 128, "dz3t")]:
         assert torch.equal(
